@@ -1,0 +1,87 @@
+"""Message subscribers: rich console, JSONL file, dummy
+(capability parity with reference src/modalities/logging_broker/
+subscriber_impl/results_subscriber.py and progress_subscriber.py; a WandB
+sink is intentionally not bundled — offline environment)."""
+
+import json
+from pathlib import Path
+from typing import Optional
+
+import torch
+
+from modalities_amd.batch import EvaluationResultBatch
+from modalities_amd.logging_broker.broker import Message, MessageSubscriberIF
+
+
+class DummyResultSubscriber(MessageSubscriberIF):
+    def consume_message(self, message: Message) -> None:
+        pass
+
+    def consume_dict(self, message_dict: dict) -> None:
+        pass
+
+
+class DummyProgressSubscriber(MessageSubscriberIF):
+    def consume_message(self, message: Message) -> None:
+        pass
+
+    def consume_dict(self, message_dict: dict) -> None:
+        pass
+
+
+def _jsonify(value):
+    if isinstance(value, torch.Tensor):
+        return value.item() if value.numel() == 1 else value.tolist()
+    return value
+
+
+class RichResultSubscriber(MessageSubscriberIF):
+    """Console sink for evaluation results (rank 0)."""
+
+    def __init__(self, num_ranks: int = 1, global_rank: int = 0):
+        self.enabled = global_rank == 0
+
+    def consume_message(self, message: Message) -> None:
+        if not self.enabled:
+            return
+        payload = message.payload
+        if isinstance(payload, EvaluationResultBatch):
+            print(str(payload), flush=True)
+
+    def consume_dict(self, message_dict: dict) -> None:
+        if self.enabled:
+            print(json.dumps(message_dict, default=str), flush=True)
+
+
+class ResultsToDiscSubscriber(MessageSubscriberIF):
+    """Rank-0 JSONL sink — the machine-readable benchmark/evaluation output
+    consumed by sweep tooling (reference: results_subscriber.py:120-168)."""
+
+    def __init__(self, output_file_path: Path, global_rank: int = 0):
+        self.enabled = global_rank == 0
+        self.path = Path(output_file_path)
+        if self.enabled:
+            self.path.parent.mkdir(parents=True, exist_ok=True)
+
+    def _write(self, record: dict) -> None:
+        with self.path.open("a") as f:
+            f.write(json.dumps(record, default=str) + "\n")
+
+    def consume_message(self, message: Message) -> None:
+        if not self.enabled:
+            return
+        payload = message.payload
+        if isinstance(payload, EvaluationResultBatch):
+            record = {
+                "dataloader_tag": payload.dataloader_tag,
+                "num_train_steps_done": payload.num_train_steps_done,
+                "losses": {k: _jsonify(v.value) for k, v in payload.losses.items()},
+                "metrics": {k: _jsonify(v.value) for k, v in payload.metrics.items()},
+                "throughput_metrics": {k: _jsonify(v.value)
+                                       for k, v in payload.throughput_metrics.items()},
+            }
+            self._write(record)
+
+    def consume_dict(self, message_dict: dict) -> None:
+        if self.enabled:
+            self._write(message_dict)

@@ -1,0 +1,87 @@
+"""Optimizers for the sharded engine + plain models.
+
+ShardedAdamW steps directly on the engine's flat fp32 master shards with the
+fused HIP AdamW kernel (K9) and a per-element weight-decay mask — the flat
+equivalent of the reference's regex weight-decay groups (reference:
+src/modalities/optimizers/optimizer_factory.py:22-215)."""
+
+import re
+from typing import Iterable, Optional
+
+import torch
+
+from modalities_amd.ops.adamw import fused_adamw_step
+from modalities_amd.parallel.fsdp import XGMIShardedModel
+
+
+class ShardedAdamW(torch.optim.Optimizer):
+    """AdamW over XGMIShardedModel master shards.
+
+    - exp_avg / exp_avg_sq are fp32 shards (persisted in state for DCP).
+    - weight decay is applied through each unit's wd_mask_shard so norm/bias
+      elements get 0 decay inside the flat shard.
+    """
+
+    def __init__(self, sharded_model: XGMIShardedModel, lr: float = 3e-4,
+                 betas: tuple[float, float] = (0.9, 0.95), eps: float = 1e-8,
+                 weight_decay: float = 0.1):
+        self.sharded_model = sharded_model
+        params = [u.master_shard for u in sharded_model.units]
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        for u in sharded_model.units:
+            st = self.state[u.master_shard]
+            st["step"] = 0
+            st["exp_avg"] = torch.zeros_like(u.master_shard)
+            st["exp_avg_sq"] = torch.zeros_like(u.master_shard)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = closure() if closure is not None else None
+        group = self.param_groups[0]
+        lr, (beta1, beta2) = group["lr"], group["betas"]
+        eps, wd = group["eps"], group["weight_decay"]
+        for u in self.sharded_model.units:
+            st = self.state[u.master_shard]
+            st["step"] += 1
+            step = st["step"]
+            m, v = st["exp_avg"], st["exp_avg_sq"]
+            g = u.grad_shard
+            p = u.master_shard
+            bc1 = 1.0 - beta1 ** step
+            bc2 = 1.0 - beta2 ** step
+            from modalities_amd.ops.backend import hip_available, hip_ext
+            if hip_available() and p.is_cuda:
+                hip_ext().fused_adamw_masked(p, g, m, v, u.wd_mask_shard,
+                                             lr, beta1, beta2, eps, wd, bc1, bc2)
+            else:
+                p.mul_(1.0 - lr * wd * u.wd_mask_shard)
+                m.mul_(beta1).add_(g, alpha=1 - beta1)
+                v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+                denom = (v / bc2).sqrt().add_(eps)
+                p.add_(-lr / bc1 * m / denom)
+        self.sharded_model.publish_master()
+        return loss
+
+    def zero_grad(self, set_to_none: bool = True):
+        self.sharded_model.zero_grad_shards()
+
+
+def get_adam_w(wrapped_model, lr: float = 3e-4, betas=(0.9, 0.95), eps: float = 1e-8,
+               weight_decay: float = 0.1, weight_decay_groups_excluded=None,
+               foreach: Optional[bool] = None, fused: Optional[bool] = None):
+    """Factory: sharded models get ShardedAdamW; plain modules get
+    torch.optim.AdamW with decay/no-decay param groups."""
+    if isinstance(wrapped_model, XGMIShardedModel):
+        return ShardedAdamW(wrapped_model, lr=lr, betas=betas, eps=eps,
+                            weight_decay=weight_decay)
+    decay, no_decay = [], []
+    for name, p in wrapped_model.named_parameters():
+        if not p.requires_grad:
+            continue
+        (no_decay if p.ndim < 2 or "norm" in name.lower() or name.endswith(".bias")
+         else decay).append(p)
+    groups = [{"params": decay, "weight_decay": weight_decay},
+              {"params": no_decay, "weight_decay": 0.0}]
+    return torch.optim.AdamW(groups, lr=lr, betas=betas, eps=eps,
+                             foreach=foreach, fused=False)
