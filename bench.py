@@ -1,0 +1,173 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark for modalities_amd on MI355X.
+
+Measures the BASELINE.json metric: whole-node training tokens/s (+MFU) for a
+2.7B-class GPT (h=2560, L=32, seq 4096, bf16, FSDP full-shard equivalent,
+micro-batch 2 per GPU — the reference's headline scaling config,
+/root/reference/README.md:321: 18.63 samples/s on 8xA100 = 76.3k tok/s).
+MI355X-first deviation from the reference 2.7B: head_dim 128 (20 heads)
+instead of 80 (32 heads) — identical h/L/param count/FLOPs per token, but
+MFMA-tile-aligned. Weak scaling: per-GPU batch fixed as N grows.
+
+Usage:
+  python bench.py --gpus 1 --steps 20 --warmup 5          # single GPU
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Synthetic data (random tokens, fixed seed per step), random-init weights.
+Rank 0 prints ONE JSON line with the whole-job aggregate.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def build_model_cfg(model_name: str):
+    from modalities_amd.models.gpt2 import GPT2LLMConfig
+    if model_name == "gpt2-2.7b":
+        return GPT2LLMConfig(
+            vocab_size=50304, n_layer=32, n_head_q=20, n_head_kv=20,
+            n_embd=2560, ffn_hidden=10240, sequence_length=4096,
+            activation_type="swiglu", use_weight_tying=False)
+    if model_name == "gpt2-tiny":  # smoke/debug
+        return GPT2LLMConfig(
+            vocab_size=50304, n_layer=4, n_head_q=4, n_head_kv=4,
+            n_embd=512, ffn_hidden=2048, sequence_length=1024)
+    raise ValueError(model_name)
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--micro-batch", type=int, default=2)
+    p.add_argument("--seq-len", type=int, default=4096)
+    p.add_argument("--model", type=str, default="gpt2-2.7b")
+    p.add_argument("--blocks-per-unit", type=int, default=4)
+    p.add_argument("--reshard", action="store_true")
+    args = p.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    assert world == args.gpus or world == 1, \
+        f"WORLD_SIZE={world} but --gpus={args.gpus}"
+
+    on_gpu = torch.cuda.is_available()
+    if world > 1:
+        dist.init_process_group("nccl" if on_gpu else "gloo")
+    if on_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    from modalities_amd.models.gpt2 import GPT2LLM
+    from modalities_amd.optimizers.optimizer_factory import get_adam_w
+    from modalities_amd.ops import fused_cross_entropy
+    from modalities_amd.parallel.fsdp import XGMIShardedModel
+    from modalities_amd.utils.mfu import GPT2MFUCalculator, detect_device_peak_flops
+
+    cfg = build_model_cfg(args.model)
+    cfg.sequence_length = args.seq_len
+    torch.manual_seed(1234)  # identical init on all ranks
+    model = GPT2LLM(cfg)
+    num_params = sum(p_.numel() for p_ in model.parameters())
+
+    sharded = XGMIShardedModel.from_transformer(
+        model, device, blocks_per_unit=args.blocks_per_unit,
+        param_dtype=torch.bfloat16 if on_gpu else torch.float32,
+        reshard_after_forward=args.reshard)
+    opt = get_adam_w(sharded, lr=3e-4, weight_decay=0.1)
+
+    B, T, V = args.micro_batch, args.seq_len, cfg.vocab_size
+
+    def make_batch(step: int):
+        g = torch.Generator().manual_seed(10_000 + step * world + rank)
+        ids = torch.randint(0, V, (B, T + 1), generator=g)
+        return ids[:, :-1].to(device, non_blocking=True), \
+            ids[:, 1:].to(device, non_blocking=True)
+
+    def one_step(step: int):
+        x, y = make_batch(step)
+        out = sharded({"input_ids": x})
+        loss = fused_cross_entropy(out["logits"], y)
+        loss.backward()
+        sharded.backward_epilogue()
+        sharded.clip_grad_norm_(1.0)
+        opt.step()
+        opt.zero_grad()
+        return loss
+
+    for s in range(args.warmup):
+        one_step(s)
+
+    if on_gpu:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    t0 = time.perf_counter()
+    for s in range(args.steps):
+        one_step(args.warmup + s)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks = slowest rank's wall time
+    t = torch.tensor([elapsed], device=device if on_gpu else "cpu")
+    if world > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = t.item()
+
+    ms_per_step = elapsed / args.steps * 1000
+    tokens_per_step_global = B * T * world
+    tokens_per_s = tokens_per_step_global * args.steps / elapsed
+
+    # Reference baseline: 2.7B 8xA100 FULL_SHARD mbs=2 -> 18.63 samples/s
+    #  = 76,308 tok/s on 8 GPUs => 9538.6 tok/s per GPU.
+    baseline_per_gpu = 18.63 * 4096 / 8
+    vs_baseline = tokens_per_s / (baseline_per_gpu * world) \
+        if args.model == "gpt2-2.7b" else None
+
+    mfu = None
+    if on_gpu:
+        calc = GPT2MFUCalculator(cfg.n_layer, T, cfg.n_embd, world, num_params,
+                                 detect_device_peak_flops())
+        m = calc.compute(torch.tensor(tokens_per_s / T))
+        mfu = round(m.item(), 4) if m.item() > 0 else None
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "train_tokens_per_s",
+            "value": round(tokens_per_s, 1),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(vs_baseline, 3) if vs_baseline else None,
+            "dtype": "bf16" if on_gpu else "fp32",
+            "data": "synthetic",
+            "mfu": mfu,
+            "config": {"model": args.model, "global_batch": B * world,
+                       "micro_batch": B, "seq_len": T,
+                       "parallelism": f"dp{world}_fullshard",
+                       "num_params": num_params},
+        }), flush=True)
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

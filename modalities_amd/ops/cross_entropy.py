@@ -32,8 +32,8 @@ class _FusedCEHip(torch.autograd.Function):
 def fused_cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
                         ignore_index: int = -100) -> torch.Tensor:
     """Mean CE over non-ignored targets. logits: [N, V]; targets: [N]."""
-    logits = logits.view(-1, logits.shape[-1])
-    targets = targets.view(-1)
+    logits = logits.reshape(-1, logits.shape[-1])
+    targets = targets.reshape(-1)
     if use_hip(logits):
         return _FusedCEHip.apply(logits.contiguous(), targets.contiguous(), ignore_index)
     return torch.nn.functional.cross_entropy(logits.float(), targets,

@@ -1,0 +1,127 @@
+// K9 fused AdamW on flat fp32 shards (+ per-element weight-decay mask for
+// the flat-shard equivalent of weight-decay groups) and K10 multi-tensor
+// L2-norm / scale. Replaces torch.optim.AdamW(fused=True)
+// (reference: optimizers/optimizer_factory.py:38-50) and
+// clip_grads_with_norm_ (fsdp_gradient_clipper.py:144-229).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+// p, g, m, v, wd_mask: flat fp32; vectorized f32x4.
+__global__ void adamw_masked_kernel(float* __restrict__ p,
+                                    const float* __restrict__ g,
+                                    float* __restrict__ m,
+                                    float* __restrict__ v,
+                                    const float* __restrict__ wd_mask,
+                                    long n4, float lr, float beta1, float beta2,
+                                    float eps, float wd, float bc1, float bc2) {
+  const float step_size = lr / bc1;
+  const float inv_bc2 = 1.0f / bc2;
+  floatx4* p4 = reinterpret_cast<floatx4*>(p);
+  const floatx4* g4 = reinterpret_cast<const floatx4*>(g);
+  floatx4* m4 = reinterpret_cast<floatx4*>(m);
+  floatx4* v4 = reinterpret_cast<floatx4*>(v);
+  const floatx4* w4 = reinterpret_cast<const floatx4*>(wd_mask);
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    floatx4 pp = p4[i], gg = g4[i], mm = m4[i], vv = v4[i], ww = w4[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      pp[j] *= (1.0f - lr * wd * ww[j]);
+      mm[j] = beta1 * mm[j] + (1.0f - beta1) * gg[j];
+      vv[j] = beta2 * vv[j] + (1.0f - beta2) * gg[j] * gg[j];
+      pp[j] -= step_size * mm[j] / (sqrtf(vv[j] * inv_bc2) + eps);
+    }
+    p4[i] = pp; m4[i] = mm; v4[i] = vv;
+  }
+}
+
+__global__ void sqsum_kernel(const float* __restrict__ x, long n4,
+                             float* __restrict__ out) {
+  __shared__ float scratch[256 / WAVE_SIZE];
+  const floatx4* x4 = reinterpret_cast<const floatx4*>(x);
+  float acc = 0.f;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    floatx4 v = x4[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc += v[j] * v[j];
+  }
+  acc = block_reduce_sum(acc, scratch);
+  if (threadIdx.x == 0) atomicAdd(out, acc);
+}
+
+__global__ void scale_kernel(float* __restrict__ x, long n4,
+                             const float* __restrict__ scale) {
+  const float s = *scale;
+  floatx4* x4 = reinterpret_cast<floatx4*>(x);
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    floatx4 v = x4[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) v[j] *= s;
+    x4[i] = v;
+  }
+}
+
+int grid_for(long work, int block) {
+  long g = (work + block - 1) / block;
+  return (int)min(g, (long)(256 * 8));
+}
+
+}  // namespace
+
+void fused_adamw_masked(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                        torch::Tensor v, torch::Tensor wd_mask, double lr,
+                        double beta1, double beta2, double eps, double wd,
+                        double bc1, double bc2) {
+  TORCH_CHECK(p.is_cuda() && p.dtype() == torch::kFloat32 && p.is_contiguous());
+  TORCH_CHECK(p.numel() % 4 == 0, "flat shard must be divisible by 4");
+  long n4 = p.numel() / 4;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(adamw_masked_kernel, dim3(grid_for(n4, 256)), dim3(256), 0,
+                     stream, p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     wd_mask.data_ptr<float>(), n4, (float)lr, (float)beta1,
+                     (float)beta2, (float)eps, (float)wd, (float)bc1, (float)bc2);
+  HIP_CHECK_KERNEL();
+}
+
+void fused_adamw(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
+                 std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                 double lr, double beta1, double beta2, double eps, double wd,
+                 double bc1, double bc2) {
+  for (size_t i = 0; i < ps.size(); ++i) {
+    auto ones = torch::ones_like(ps[i]);
+    fused_adamw_masked(ps[i], gs[i], ms[i], vs[i], ones, lr, beta1, beta2, eps,
+                       wd, bc1, bc2);
+  }
+}
+
+torch::Tensor multi_tensor_sqsum(std::vector<torch::Tensor> tensors) {
+  TORCH_CHECK(!tensors.empty());
+  auto out = torch::zeros({}, tensors[0].options().dtype(torch::kFloat32));
+  auto stream = at::cuda::getCurrentHIPStream();
+  for (auto& t : tensors) {
+    TORCH_CHECK(t.dtype() == torch::kFloat32 && t.is_contiguous()
+                && t.numel() % 4 == 0);
+    long n4 = t.numel() / 4;
+    hipLaunchKernelGGL(sqsum_kernel, dim3(grid_for(n4, 256)), dim3(256), 0,
+                       stream, t.data_ptr<float>(), n4, out.data_ptr<float>());
+    HIP_CHECK_KERNEL();
+  }
+  return out;
+}
+
+void multi_tensor_scale(std::vector<torch::Tensor> tensors, torch::Tensor scale) {
+  auto stream = at::cuda::getCurrentHIPStream();
+  auto s = scale.to(torch::kFloat32);
+  for (auto& t : tensors) {
+    long n4 = t.numel() / 4;
+    hipLaunchKernelGGL(scale_kernel, dim3(grid_for(n4, 256)), dim3(256), 0,
+                       stream, t.data_ptr<float>(), n4, s.data_ptr<float>());
+    HIP_CHECK_KERNEL();
+  }
+}

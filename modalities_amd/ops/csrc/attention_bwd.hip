@@ -1,0 +1,499 @@
+// K1 (backward): flash attention backward for CDNA4/gfx950, GQA-aware.
+// FA2-style split (no atomics):
+//   1. delta[b,h,q]  = rowsum(dO * O)                       (preprocess)
+//   2. dQ kernel  : per q-block, iterate kv tiles            (recomputes S)
+//   3. dKdV kernel: per kv-block, iterate q heads x q tiles  (recomputes S)
+// Math (S' = Q K^T, S = scale*S', P = exp(S - lse)):
+//   dV = P^T dO ; dP = dO V^T ; dS' = scale * P .* (dP - delta)
+//   dQ = dS' K  ; dK = dS'^T Q
+// All matmuls are 32x32x16 bf16 MFMAs using the same fragment maps as the
+// forward (A[i=lane&31][k=hi*8+m], B[k=hi*8+m][j=lane&31], D col=lane&31 /
+// row=crow(reg,hi)); in-register layout transposes via cvt_pk +
+// permlane32_swap (guide T12).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+namespace attnbwd {
+
+constexpr int KVBLK = 64;  // dQ kernel kv tile
+
+__device__ __forceinline__ int swz(int row, int col) {   // 64/128-wide rows
+  return col ^ ((row & 7) << 3);
+}
+__device__ __forceinline__ int swz32(int row, int col) {  // 32-wide rows
+  return col ^ ((row & 3) << 3);
+}
+__device__ __forceinline__ int crow(int reg, int hi) {
+  return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
+}
+
+// Re-layout f32 values from the MFMA C layout (col = lane&31 = j fixed,
+// rows i = crow(reg,hi)) into bf16 fragments with per-lane element
+// k = hi*8+m (+16 per frag). Used as operand A the result represents
+// X^T[j=lane&31][k=i]: exactly the transpose-in-place the bwd needs.
+// nfrag = source_rows/16 (regs p[8s..8s+7] cover rows 16s..16s+15).
+__device__ __forceinline__ void c_layout_to_frags(const float* p, bf16x8* frag,
+                                                  int nfrag) {
+  for (int s = 0; s < nfrag; ++s) {
+    unsigned int w[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      w[i] = ((unsigned int)f32_to_bf16(p[8 * s + 2 * i])) |
+             ((unsigned int)f32_to_bf16(p[8 * s + 2 * i + 1]) << 16);
+    }
+    auto r02 = __builtin_amdgcn_permlane32_swap(w[0], w[2], false, false);
+    auto r13 = __builtin_amdgcn_permlane32_swap(w[1], w[3], false, false);
+    unsigned int fw[4] = {(unsigned int)r02[0], (unsigned int)r13[0],
+                          (unsigned int)r02[1], (unsigned int)r13[1]};
+    frag[s] = *reinterpret_cast<bf16x8*>(fw);
+  }
+}
+
+// ---------------- delta preprocess ----------------------------------------
+__global__ void delta_kernel(const unsigned short* __restrict__ dout,
+                             const unsigned short* __restrict__ o,
+                             float* __restrict__ delta,
+                             long total_rows, int T, int H, int D) {
+  const long row = (blockIdx.x * (long)blockDim.x + threadIdx.x) / WAVE_SIZE;
+  const int lane = threadIdx.x & 63;
+  if (row >= total_rows) return;  // row = ((b*T)+t)*H + h
+  const long base = row * D;
+  float acc = 0.f;
+  for (int d = lane * 2; d < D; d += WAVE_SIZE * 2) {
+    unsigned int du = *reinterpret_cast<const unsigned int*>(dout + base + d);
+    unsigned int ou = *reinterpret_cast<const unsigned int*>(o + base + d);
+    acc += bf16_to_f32((unsigned short)(du & 0xffff)) *
+           bf16_to_f32((unsigned short)(ou & 0xffff));
+    acc += bf16_to_f32((unsigned short)(du >> 16)) *
+           bf16_to_f32((unsigned short)(ou >> 16));
+  }
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) {
+    const int h = (int)(row % H);
+    const long bt = row / H;
+    const int t = (int)(bt % T);
+    const long b = bt / T;
+    delta[((b * H) + h) * T + t] = acc;
+  }
+}
+
+// ---------------- dQ kernel -----------------------------------------------
+// 4 waves x 32 q rows = 128 q rows per WG; kv tiles of 64 in LDS (K row +
+// V row as S^T/dP^T A-operands; Kt as the dQ^T A-operand). S^T/P^T/dS'^T
+// all live in the col = q = lane&31 layout (like the forward), so lse and
+// delta are lane-local.
+template <int D>
+struct DQKernel {
+  static constexpr int NDSTEP = D / 16;
+  static constexpr int NDBLK = D / 32;
+  struct Smem {
+    unsigned short k[KVBLK * D];
+    unsigned short v[KVBLK * D];
+    unsigned short kt[D * KVBLK];
+  };
+
+  static __device__ void run(const unsigned short* q, const unsigned short* k,
+                             const unsigned short* v, const unsigned short* dout,
+                             const float* lse, const float* delta,
+                             unsigned short* dq, int B, int T, int Hq, int Hkv,
+                             float scale, char* smem_raw) {
+    Smem* sm = reinterpret_cast<Smem*>(smem_raw);
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int hi = lane >> 5, ln31 = lane & 31;
+    const int h = blockIdx.y, b = blockIdx.z;
+    const int hkv = h / (Hq / Hkv);
+    const int qblk0 = blockIdx.x * 128;
+    const int qg = qblk0 + wid * 32 + ln31;
+
+    const long q_base = (((long)b * T) * Hq + h) * D;
+    const long kv_base = (((long)b * T) * Hkv + hkv) * D;
+
+    bf16x8 qfrag[NDSTEP], dofrag[NDSTEP];
+    const unsigned short* qr = q + q_base + (long)qg * Hq * D;
+    const unsigned short* dor = dout + q_base + (long)qg * Hq * D;
+#pragma unroll
+    for (int s = 0; s < NDSTEP; ++s) {
+      if (qg < T) {
+        qfrag[s] = *reinterpret_cast<const bf16x8*>(qr + hi * 8 + 16 * s);
+        dofrag[s] = *reinterpret_cast<const bf16x8*>(dor + hi * 8 + 16 * s);
+      } else {
+#pragma unroll
+        for (int m = 0; m < 8; ++m) { qfrag[s][m] = (__bf16)0.f;
+                                      dofrag[s][m] = (__bf16)0.f; }
+      }
+    }
+    const float my_lse = (qg < T) ? lse[(((long)b * Hq) + h) * T + qg] : 0.f;
+    const float my_delta = (qg < T) ? delta[(((long)b * Hq) + h) * T + qg] : 0.f;
+
+    floatx16 acc_dq[NDBLK];
+#pragma unroll
+    for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc_dq[dblk][r] = 0.f;
+
+    const int q_hi_wg = min(qblk0 + 127, T - 1);
+    const int n_tiles = q_hi_wg / KVBLK + 1;
+
+    for (int tile = 0; tile < n_tiles; ++tile) {
+      const int kv0 = tile * KVBLK;
+      {  // stage K, V (row-major) and Kt (transposed)
+        const int tid = threadIdx.x, nthreads = 256;
+        const int total8 = KVBLK * D / 8;
+        for (int i = tid; i < total8; i += nthreads) {
+          const int r = (i * 8) / D, c = (i * 8) % D;
+          shortx8 kk, vv;
+          if (kv0 + r < T) {
+            kk = *reinterpret_cast<const shortx8*>(
+                k + kv_base + (long)(kv0 + r) * Hkv * D + c);
+            vv = *reinterpret_cast<const shortx8*>(
+                v + kv_base + (long)(kv0 + r) * Hkv * D + c);
+          } else {
+#pragma unroll
+            for (int m = 0; m < 8; ++m) { kk[m] = 0; vv[m] = 0; }
+          }
+          *reinterpret_cast<shortx8*>(&sm->k[r * D + swz(r, c)]) = kk;
+          *reinterpret_cast<shortx8*>(&sm->v[r * D + swz(r, c)]) = vv;
+#pragma unroll
+          for (int m = 0; m < 8; ++m) {
+            const int d = c + m;
+            sm->kt[d * KVBLK + swz(d, r)] = (unsigned short)kk[m];
+          }
+        }
+      }
+      __syncthreads();
+
+      // S^T = K Q^T ; dP^T = V dO^T   (both: col = q = ln31, rows = kv)
+      floatx16 s0, s1, dp0, dp1;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { s0[r] = s1[r] = dp0[r] = dp1[r] = 0.f; }
+#pragma unroll
+      for (int s = 0; s < NDSTEP; ++s) {
+        const int col = hi * 8 + 16 * s;
+        const int r0 = ln31, r1 = ln31 + 32;
+        bf16x8 ka = *reinterpret_cast<const bf16x8*>(&sm->k[r0 * D + swz(r0, col)]);
+        bf16x8 kb = *reinterpret_cast<const bf16x8*>(&sm->k[r1 * D + swz(r1, col)]);
+        bf16x8 va = *reinterpret_cast<const bf16x8*>(&sm->v[r0 * D + swz(r0, col)]);
+        bf16x8 vb = *reinterpret_cast<const bf16x8*>(&sm->v[r1 * D + swz(r1, col)]);
+        s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[s], s0, 0, 0, 0);
+        s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb, qfrag[s], s1, 0, 0, 0);
+        dp0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[s], dp0, 0, 0, 0);
+        dp1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vb, dofrag[s], dp1, 0, 0, 0);
+      }
+
+      // dS'^T = scale * P .* (dP - delta), P = exp(scale*S' - lse)
+      float ds[32];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kg0 = kv0 + crow(r, hi);
+        const int kg1 = kv0 + 32 + crow(r, hi);
+        const float p0 = (kg0 <= qg && kg0 < T)
+                             ? __expf(s0[r] * scale - my_lse) : 0.f;
+        const float p1 = (kg1 <= qg && kg1 < T)
+                             ? __expf(s1[r] * scale - my_lse) : 0.f;
+        ds[r] = scale * p0 * (dp0[r] - my_delta);
+        ds[16 + r] = scale * p1 * (dp1[r] - my_delta);
+      }
+      bf16x8 dsfrag[KVBLK / 16];
+      c_layout_to_frags(ds, dsfrag, KVBLK / 16);
+
+      // dQ^T += Kt dS'^T-frags : D[m=d][n=q], col = q = ln31
+#pragma unroll
+      for (int dblk = 0; dblk < NDBLK; ++dblk) {
+#pragma unroll
+        for (int s = 0; s < KVBLK / 16; ++s) {
+          const int d = dblk * 32 + ln31;
+          const int col = hi * 8 + 16 * s;
+          bf16x8 kta = *reinterpret_cast<const bf16x8*>(
+              &sm->kt[d * KVBLK + swz(d, col)]);
+          acc_dq[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              kta, dsfrag[s], acc_dq[dblk], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+
+    if (qg < T) {
+      unsigned short* dqr = dq + q_base + (long)qg * Hq * D;
+#pragma unroll
+      for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int d = dblk * 32 + crow(r, hi);
+          dqr[d] = f32_to_bf16(acc_dq[dblk][r]);
+        }
+    }
+  }
+};
+
+// ---------------- dKdV kernel ---------------------------------------------
+// 4 waves x 32 kv rows = 128 kv rows per WG; loop (q head in GQA group) x
+// (q tiles of 32). K,V live in REGISTERS (lane owns kv row
+// wavebase+ln31); LDS holds the q tile 4 ways: Q row-major, dO row-major
+// (A-operands of S/dP), Qt, dOt transposed (B-operands of dK/dV).
+//   S [q][kv]  = mfma(A=Q_lds,  B=K_regs)    col = kv = ln31, rows q = crow
+//   dP[q][kv]  = mfma(A=dO_lds, B=V_regs)    same layout
+//   dS' = scale * P .* (dP - delta[q])       (lse/delta loaded per q row)
+//   dV[kv][d] += mfma(A=frags(P),   B=dOt)   col = d = ln31, rows kv = crow
+//   dK[kv][d] += mfma(A=frags(dS'), B=Qt)
+template <int D>
+struct DKDVKernel {
+  static constexpr int NDSTEP = D / 16;
+  static constexpr int NDBLK = D / 32;
+  struct Smem {
+    unsigned short qrow[32 * D];
+    unsigned short dorow[32 * D];
+    unsigned short qt[D * 32];
+    unsigned short dot[D * 32];
+  };
+
+  static __device__ void run(const unsigned short* q, const unsigned short* k,
+                             const unsigned short* v, const unsigned short* dout,
+                             const float* lse, const float* delta,
+                             unsigned short* dk, unsigned short* dv,
+                             int B, int T, int Hq, int Hkv, float scale,
+                             char* smem_raw) {
+    Smem* sm = reinterpret_cast<Smem*>(smem_raw);
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int hi = lane >> 5, ln31 = lane & 31;
+    const int hkv = blockIdx.y, b = blockIdx.z;
+    const int rep = Hq / Hkv;
+    const int kvblk0 = blockIdx.x * 128;
+    const int kvg = kvblk0 + wid * 32 + ln31;  // this lane's kv row
+
+    const long kv_base = (((long)b * T) * Hkv + hkv) * D;
+
+    bf16x8 kfrag[NDSTEP], vfrag[NDSTEP];
+    {
+      const unsigned short* kr = k + kv_base + (long)kvg * Hkv * D;
+      const unsigned short* vr = v + kv_base + (long)kvg * Hkv * D;
+#pragma unroll
+      for (int s = 0; s < NDSTEP; ++s) {
+        if (kvg < T) {
+          kfrag[s] = *reinterpret_cast<const bf16x8*>(kr + hi * 8 + 16 * s);
+          vfrag[s] = *reinterpret_cast<const bf16x8*>(vr + hi * 8 + 16 * s);
+        } else {
+#pragma unroll
+          for (int m = 0; m < 8; ++m) { kfrag[s][m] = (__bf16)0.f;
+                                        vfrag[s][m] = (__bf16)0.f; }
+        }
+      }
+    }
+
+    floatx16 acc_dk[NDBLK], acc_dv[NDBLK];
+#pragma unroll
+    for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { acc_dk[dblk][r] = 0.f; acc_dv[dblk][r] = 0.f; }
+
+    const int first_qtile = kvblk0 / 32;  // causal: q >= kv
+    const int n_qtiles = (T + 31) / 32;
+
+    for (int hq = hkv * rep; hq < (hkv + 1) * rep; ++hq) {
+      const long q_base = (((long)b * T) * Hq + hq) * D;
+      const float* lse_h = lse + (((long)b * Hq) + hq) * T;
+      const float* delta_h = delta + (((long)b * Hq) + hq) * T;
+
+      for (int qt = first_qtile; qt < n_qtiles; ++qt) {
+        const int q0 = qt * 32;
+        {  // stage the q tile 4 ways
+          const int tid = threadIdx.x, nthreads = 256;
+          const int total8 = 32 * D / 8;
+          for (int i = tid; i < total8; i += nthreads) {
+            const int r = (i * 8) / D, c = (i * 8) % D;
+            shortx8 qq, dd;
+            if (q0 + r < T) {
+              qq = *reinterpret_cast<const shortx8*>(
+                  q + q_base + (long)(q0 + r) * Hq * D + c);
+              dd = *reinterpret_cast<const shortx8*>(
+                  dout + q_base + (long)(q0 + r) * Hq * D + c);
+            } else {
+#pragma unroll
+              for (int m = 0; m < 8; ++m) { qq[m] = 0; dd[m] = 0; }
+            }
+            *reinterpret_cast<shortx8*>(&sm->qrow[r * D + swz(r, c)]) = qq;
+            *reinterpret_cast<shortx8*>(&sm->dorow[r * D + swz(r, c)]) = dd;
+#pragma unroll
+            for (int m = 0; m < 8; ++m) {
+              const int d = c + m;
+              sm->qt[d * 32 + swz32(d, r)] = (unsigned short)qq[m];
+              sm->dot[d * 32 + swz32(d, r)] = (unsigned short)dd[m];
+            }
+          }
+        }
+        __syncthreads();
+
+        // S[q][kv] and dP[q][kv] (col = kv = ln31, rows q = crow)
+        floatx16 s_acc, dp_acc;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) { s_acc[r] = 0.f; dp_acc[r] = 0.f; }
+#pragma unroll
+        for (int s = 0; s < NDSTEP; ++s) {
+          const int col = hi * 8 + 16 * s;
+          bf16x8 qa = *reinterpret_cast<const bf16x8*>(
+              &sm->qrow[ln31 * D + swz(ln31, col)]);
+          bf16x8 da = *reinterpret_cast<const bf16x8*>(
+              &sm->dorow[ln31 * D + swz(ln31, col)]);
+          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[s], s_acc,
+                                                          0, 0, 0);
+          dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vfrag[s], dp_acc,
+                                                           0, 0, 0);
+        }
+
+        float p_c[16], ds_c[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qg = q0 + crow(r, hi);
+          const bool ok = (qg < T) && (qg >= kvg) && (kvg < T);
+          const float lq = (qg < T) ? lse_h[qg] : 0.f;
+          const float dl = (qg < T) ? delta_h[qg] : 0.f;
+          const float pv = ok ? __expf(s_acc[r] * scale - lq) : 0.f;
+          p_c[r] = pv;
+          ds_c[r] = scale * pv * (dp_acc[r] - dl);
+        }
+        bf16x8 pfrag[2], dsfrag[2];
+        c_layout_to_frags(p_c, pfrag, 2);
+        c_layout_to_frags(ds_c, dsfrag, 2);
+
+        // dV += P^T dO ; dK += dS'^T Q
+#pragma unroll
+        for (int dblk = 0; dblk < NDBLK; ++dblk) {
+#pragma unroll
+          for (int s = 0; s < 2; ++s) {
+            const int d = dblk * 32 + ln31;
+            const int col = hi * 8 + 16 * s;
+            bf16x8 dob = *reinterpret_cast<const bf16x8*>(
+                &sm->dot[d * 32 + swz32(d, col)]);
+            bf16x8 qb = *reinterpret_cast<const bf16x8*>(
+                &sm->qt[d * 32 + swz32(d, col)]);
+            acc_dv[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                pfrag[s], dob, acc_dv[dblk], 0, 0, 0);
+            acc_dk[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                dsfrag[s], qb, acc_dk[dblk], 0, 0, 0);
+          }
+        }
+        __syncthreads();
+      }
+    }
+
+    // dK/dV output layout: col = ln31 = d within block, rows = crow = kv
+    // local row (NOT this lane's kvg) -> scatter by crow.
+#pragma unroll
+    for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kvrow = kvblk0 + wid * 32 + crow(r, hi);
+        if (kvrow < T) {
+          const long off = kv_base + (long)kvrow * Hkv * D + dblk * 32 + ln31;
+          dk[off] = f32_to_bf16(acc_dk[dblk][r]);
+          dv[off] = f32_to_bf16(acc_dv[dblk][r]);
+        }
+      }
+  }
+};
+
+template <int D>
+__global__ __launch_bounds__(256) void dq_kernel(
+    const unsigned short* q, const unsigned short* k, const unsigned short* v,
+    const unsigned short* dout, const float* lse, const float* delta,
+    unsigned short* dq, int B, int T, int Hq, int Hkv, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  DQKernel<D>::run(q, k, v, dout, lse, delta, dq, B, T, Hq, Hkv, scale,
+                   smem_raw);
+}
+
+template <int D>
+__global__ __launch_bounds__(256) void dkdv_kernel(
+    const unsigned short* q, const unsigned short* k, const unsigned short* v,
+    const unsigned short* dout, const float* lse, const float* delta,
+    unsigned short* dk, unsigned short* dv, int B, int T, int Hq, int Hkv,
+    float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  DKDVKernel<D>::run(q, k, v, dout, lse, delta, dk, dv, B, T, Hq, Hkv, scale,
+                     smem_raw);
+}
+
+}  // namespace attnbwd
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    bool causal) {
+  TORCH_CHECK(causal, "attn_bwd: only causal attention is implemented");
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+  const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  const float scale = 1.0f / sqrtf((float)D);
+  auto stream = at::cuda::getCurrentHIPStream();
+
+  auto delta = torch::empty({B, Hq, T}, q.options().dtype(torch::kFloat32));
+  {
+    long total_rows = (long)B * T * Hq;
+    long waves = total_rows;
+    long threads = waves * WAVE_SIZE;
+    int block = 256;
+    long grid = (threads + block - 1) / block;
+    hipLaunchKernelGGL(attnbwd::delta_kernel, dim3((unsigned)grid), dim3(block),
+                       0, stream, (const unsigned short*)dout.data_ptr(),
+                       (const unsigned short*)o.data_ptr(),
+                       delta.data_ptr<float>(), total_rows, T, Hq, D);
+    HIP_CHECK_KERNEL();
+  }
+
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+
+  const int n_qblk = (T + 127) / 128;
+  const int n_kvblk = (T + 127) / 128;
+  if (D == 128) {
+    size_t smem_dq = sizeof(typename attnbwd::DQKernel<128>::Smem);
+    hipLaunchKernelGGL(attnbwd::dq_kernel<128>, dim3(n_qblk, Hq, B), dim3(256),
+                       smem_dq, stream, (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (unsigned short*)dq.data_ptr(), B, T, Hq, Hkv, scale);
+    HIP_CHECK_KERNEL();
+    size_t smem_kv = sizeof(typename attnbwd::DKDVKernel<128>::Smem);
+    hipLaunchKernelGGL(attnbwd::dkdv_kernel<128>, dim3(n_kvblk, Hkv, B),
+                       dim3(256), smem_kv, stream,
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (unsigned short*)dk.data_ptr(),
+                       (unsigned short*)dv.data_ptr(), B, T, Hq, Hkv, scale);
+    HIP_CHECK_KERNEL();
+  } else if (D == 64) {
+    size_t smem_dq = sizeof(typename attnbwd::DQKernel<64>::Smem);
+    hipLaunchKernelGGL(attnbwd::dq_kernel<64>, dim3(n_qblk, Hq, B), dim3(256),
+                       smem_dq, stream, (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (unsigned short*)dq.data_ptr(), B, T, Hq, Hkv, scale);
+    HIP_CHECK_KERNEL();
+    size_t smem_kv = sizeof(typename attnbwd::DKDVKernel<64>::Smem);
+    hipLaunchKernelGGL(attnbwd::dkdv_kernel<64>, dim3(n_kvblk, Hkv, B),
+                       dim3(256), smem_kv, stream,
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (unsigned short*)dk.data_ptr(),
+                       (unsigned short*)dv.data_ptr(), B, T, Hq, Hkv, scale);
+    HIP_CHECK_KERNEL();
+  } else {
+    TORCH_CHECK(false, "attn_bwd: head_dim must be 64 or 128");
+  }
+  return {dq, dk, dv};
+}

@@ -1,0 +1,335 @@
+// K1 (forward): flash-style causal attention for CDNA4/gfx950, GQA-aware.
+//
+// Replaces the reference's flash-attn wheel / torch SDPA paths (reference:
+// src/modalities/models/gpt2/gpt2_model.py:595-658) with a hand-written
+// MFMA kernel. Structure follows the CDNA4 guide's attention ladder:
+//
+// - workgroup = 4 waves; each wave owns a 32-row Q block (WG covers 128 rows)
+// - KV tiles of 64 staged in LDS by the whole WG:
+//     K   [64][D]  row-major, XOR-swizzled (guide T2: col ^= (row&7)<<3)
+//     V^T [D][64]  transposed+swizzled, so PV reads rows (consecutive kv)
+// - SWAPPED QK^T: S^T[kv][q] = mfma(A=K, B=Q^T) puts a full softmax row in
+//   one lane pair (q = lane&31) -> row reduce is in-register + 1 shfl_xor(32)
+// - SWAPPED PV:   O^T[d][q]  = mfma(A=V^T, B=P^T): the online-softmax O
+//   rescale stays lane-local (same q = lane&31 layout as the stats)
+// - P (f32, S-layout) -> PV A-fragment (bf16) via cvt_pk + permlane32_swap
+//   (guide T12) — no LDS round-trip for P.
+//
+// D in {64, 128}; causal only. bf16 in/out, f32 softmax state; lse saved
+// for the backward.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+namespace attn {
+
+constexpr int QBLK = 32;     // q rows per wave
+constexpr int NWAVES = 4;    // waves per workgroup
+constexpr int WG_Q = QBLK * NWAVES;  // 128 q rows per workgroup
+constexpr int KVBLK = 64;    // kv rows per LDS tile
+
+__device__ __forceinline__ int swz(int row, int col) {
+  // element-index XOR swizzle for bf16 tiles read with ds_read_b128
+  return col ^ ((row & 7) << 3);
+}
+
+// C/D register map of v_mfma_f32_32x32x16_bf16 (guide §3):
+//   col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+__device__ __forceinline__ int crow(int reg, int hi) {
+  return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
+}
+
+// A/B fragment k-index for 32x32x16: lane provides elements k = hi*8 + m.
+template <int D_>
+struct AttnFwdKernel {
+  static constexpr int D = D_;
+  static constexpr int NDSTEP = D / 16;   // QK^T contraction steps
+  static constexpr int NDBLK = D / 32;    // PV output column blocks
+
+  // LDS: K tile + V^T tile (both swizzled, bf16)
+  struct Smem {
+    unsigned short k[KVBLK * D];
+    unsigned short vt[D * KVBLK];
+  };
+
+  static __device__ void run(const unsigned short* __restrict__ q,
+                             const unsigned short* __restrict__ k,
+                             const unsigned short* __restrict__ v,
+                             unsigned short* __restrict__ o,
+                             float* __restrict__ lse,
+                             int B, int T, int Hq, int Hkv, float scale,
+                             char* smem_raw) {
+    Smem* sm = reinterpret_cast<Smem*>(smem_raw);
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int hi = lane >> 5;
+    const int ln31 = lane & 31;
+
+    const int qblk0 = blockIdx.x * WG_Q;       // first q row of workgroup
+    const int h = blockIdx.y;
+    const int b = blockIdx.z;
+    const int hkv = h / (Hq / Hkv);
+
+    const long q_base = (((long)b * T) * Hq + h) * D;        // + t*Hq*D
+    const long kv_base = (((long)b * T) * Hkv + hkv) * D;
+    const int qg = qblk0 + wid * QBLK + ln31;  // this lane's q row
+
+    // ---- Q preload: lane holds Q[qg][hi*8 + m + 16*s], s=0..NDSTEP-1 ----
+    bf16x8 qfrag[NDSTEP];
+    {
+      const unsigned short* qr = q + q_base + (long)qg * Hq * D;
+#pragma unroll
+      for (int s = 0; s < NDSTEP; ++s) {
+        if (qg < T) {
+          qfrag[s] = *reinterpret_cast<const bf16x8*>(qr + hi * 8 + 16 * s);
+        } else {
+          for (int m = 0; m < 8; ++m) qfrag[s][m] = (__bf16)0.f;
+        }
+      }
+    }
+
+    // ---- softmax state (per lane: q row = ln31; both hi halves track) ----
+    float m_run = -INFINITY, l_run = 0.f;
+    floatx16 acc_o[NDBLK];
+#pragma unroll
+    for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc_o[dblk][r] = 0.f;
+
+    const int q_hi_wg = min(qblk0 + WG_Q - 1, T - 1);  // causal upper bound
+    const int n_tiles = (min(q_hi_wg, T - 1)) / KVBLK + 1;
+
+    for (int tile = 0; tile < n_tiles; ++tile) {
+      const int kv0 = tile * KVBLK;
+      // ---- stage K and V^T into LDS (whole workgroup, coalesced reads) ----
+      {
+        const int tid = threadIdx.x;
+        const int nthreads = NWAVES * 64;
+        // each thread copies 8 bf16 per iteration
+        const int total8 = KVBLK * D / 8;
+        for (int i = tid; i < total8; i += nthreads) {
+          const int r = (i * 8) / D;        // kv row in tile
+          const int c = (i * 8) % D;        // d col
+          shortx8 val;
+          if (kv0 + r < T) {
+            val = *reinterpret_cast<const shortx8*>(
+                k + kv_base + (long)(kv0 + r) * Hkv * D + c);
+          } else {
+            for (int m = 0; m < 8; ++m) val[m] = 0;
+          }
+          *reinterpret_cast<shortx8*>(&sm->k[r * D + swz(r, c)]) = val;
+          // V: read the same global layout, write transposed (scalar LDS
+          // stores — v1; upgrade path: ds_read_b64_tr_b16 per guide T10)
+          shortx8 vv;
+          if (kv0 + r < T) {
+            vv = *reinterpret_cast<const shortx8*>(
+                v + kv_base + (long)(kv0 + r) * Hkv * D + c);
+          } else {
+            for (int m = 0; m < 8; ++m) vv[m] = 0;
+          }
+#pragma unroll
+          for (int m = 0; m < 8; ++m) {
+            const int d = c + m;
+            sm->vt[d * KVBLK + swz(d, r)] = (unsigned short)vv[m];
+          }
+        }
+      }
+      __syncthreads();
+
+      // ---- S^T = K Q^T : two 32x32 accs (kv 0-31, 32-63) ----
+      floatx16 s0, s1;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { s0[r] = 0.f; s1[r] = 0.f; }
+#pragma unroll
+      for (int s = 0; s < NDSTEP; ++s) {
+        const int col = hi * 8 + 16 * s;
+        const int r0 = ln31, r1 = ln31 + 32;
+        bf16x8 ka = *reinterpret_cast<const bf16x8*>(&sm->k[r0 * D + swz(r0, col)]);
+        bf16x8 kb = *reinterpret_cast<const bf16x8*>(&sm->k[r1 * D + swz(r1, col)]);
+        s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[s], s0, 0, 0, 0);
+        s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb, qfrag[s], s1, 0, 0, 0);
+      }
+
+      // ---- scale + causal mask; per-lane P rows (q = ln31) --------------
+      float p[32];
+      float tmax = -INFINITY;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kg0 = kv0 + crow(r, hi);
+        const int kg1 = kv0 + 32 + crow(r, hi);
+        float v0 = (kg0 <= qg && kg0 < T) ? s0[r] * scale : -INFINITY;
+        float v1 = (kg1 <= qg && kg1 < T) ? s1[r] * scale : -INFINITY;
+        p[r] = v0;
+        p[16 + r] = v1;
+        tmax = fmaxf(tmax, fmaxf(v0, v1));
+      }
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));  // other kv half, same q
+
+      // ---- online softmax update ----------------------------------------
+      const float m_new = fmaxf(m_run, tmax);
+      const float m_safe = (m_new == -INFINITY) ? 0.f : m_new;
+      // -inf guards: fully-masked tiles keep O/l at 0 without NaNs
+      const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_safe);
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 32; ++r) {
+        p[r] = __expf(p[r] - m_safe);
+        psum += p[r];
+      }
+      psum += __shfl_xor(psum, 32, 64);
+      l_run = l_run * alpha + psum;
+      m_run = m_new;
+#pragma unroll
+      for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc_o[dblk][r] *= alpha;
+
+      // ---- P (f32, S-layout) -> PV A/B fragment (bf16) -------------------
+      // Lane holds P[q=ln31][kv in crow set]. The PV B-operand needs
+      // P^T[k=kv = hi*8+m+16*s][n=q=ln31]. Word pairs plus one
+      // permlane32_swap per pair produce the fragment (guide T12 analysis):
+      //   after swap(w0=cvtpk(p0,p1), w2=cvtpk(p4,p5)):
+      //     lanes hi=0: w0 = kv(0,1), w2 = kv(4,5)
+      //     lanes hi=1: w0 = kv(8,9), w2 = kv(12,13)
+      bf16x8 pfrag[KVBLK / 16];
+#pragma unroll
+      for (int s = 0; s < KVBLK / 16; ++s) {
+        // registers covering kv block s*16..s*16+15 in this lane: p[8s..8s+7]
+        unsigned int w[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const float lo = p[8 * s + 2 * i];
+          const float hi_v = p[8 * s + 2 * i + 1];
+          w[i] = ((unsigned int)f32_to_bf16(lo)) |
+                 ((unsigned int)f32_to_bf16(hi_v) << 16);
+        }
+        auto r02 = __builtin_amdgcn_permlane32_swap(w[0], w[2], false, false);
+        auto r13 = __builtin_amdgcn_permlane32_swap(w[1], w[3], false, false);
+        unsigned int frag_words[4] = {(unsigned int)r02[0], (unsigned int)r13[0],
+                                      (unsigned int)r02[1], (unsigned int)r13[1]};
+        pfrag[s] = *reinterpret_cast<bf16x8*>(frag_words);
+      }
+
+      // ---- O^T += V^T P^T : A = V^T rows (consecutive kv), B = P^T -------
+#pragma unroll
+      for (int dblk = 0; dblk < NDBLK; ++dblk) {
+#pragma unroll
+        for (int s = 0; s < KVBLK / 16; ++s) {
+          const int d = dblk * 32 + ln31;
+          const int col = hi * 8 + 16 * s;  // kv index within tile
+          bf16x8 va = *reinterpret_cast<const bf16x8*>(
+              &sm->vt[d * KVBLK + swz(d, col)]);
+          acc_o[dblk] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, pfrag[s], acc_o[dblk],
+                                                      0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+
+    // ---- epilogue: normalize, write O (scatter by q row) + lse ----------
+    const float l_safe = (l_run > 0.f) ? l_run : 1.f;
+    const float inv_l = 1.f / l_safe;
+    if (qg < T) {
+      unsigned short* orow = o + q_base + (long)qg * Hq * D;
+      // lane holds O^T[d = dblk*32 + crow(r,hi)][q = ln31]?? NO:
+      // PV output D[m=d_row][n=q]: col = ln31 = q; row = crow(r,hi) = d idx
+      // within the 32-col block -> this lane's values are for ITS q row.(ok)
+#pragma unroll
+      for (int dblk = 0; dblk < NDBLK; ++dblk) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int d = dblk * 32 + crow(r, hi);
+          orow[d] = f32_to_bf16(acc_o[dblk][r] * inv_l);
+        }
+      }
+      if (hi == 0) {
+        lse[(((long)b * Hq) + h) * T + qg] =
+            (m_run == -INFINITY) ? 0.f : m_run + __logf(l_safe);
+      }
+    }
+  }
+};
+
+template <int D>
+__global__ __launch_bounds__(256) void attn_fwd_kernel(
+    const unsigned short* q, const unsigned short* k, const unsigned short* v,
+    unsigned short* o, float* lse, int B, int T, int Hq, int Hkv, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  AttnFwdKernel<D>::run(q, k, v, o, lse, B, T, Hq, Hkv, scale, smem_raw);
+}
+
+}  // namespace attn
+
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, bool causal) {
+  TORCH_CHECK(causal, "attn_fwd: only causal attention is implemented");
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 4
+              && q.is_contiguous(), "q must be contiguous bf16 [B,T,Hq,D]");
+  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  TORCH_CHECK(D == 64 || D == 128, "attn_fwd: head_dim must be 64 or 128");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, Hq, T}, q.options().dtype(torch::kFloat32));
+  const float scale = 1.0f / sqrtf((float)D);
+  const int n_qblk = (T + attn::WG_Q - 1) / attn::WG_Q;
+  dim3 grid(n_qblk, Hq, B);
+  auto stream = at::cuda::getCurrentHIPStream();
+  const size_t smem = sizeof(unsigned short) * (attn::KVBLK * D + D * attn::KVBLK);
+  if (D == 128) {
+    hipLaunchKernelGGL(attn::attn_fwd_kernel<128>, grid, dim3(256), smem, stream,
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
+                       B, T, Hq, Hkv, scale);
+  } else {
+    hipLaunchKernelGGL(attn::attn_fwd_kernel<64>, grid, dim3(256), smem, stream,
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
+                       B, T, Hq, Hkv, scale);
+  }
+  HIP_CHECK_KERNEL();
+  return {o, lse};
+}
+
+// ---------------------------------------------------------------------------
+// MFMA layout probes: empirical verification of the fragment maps used above
+// (guide §3: always check with ASYMMETRIC operands).
+// a: [32,16] f32, b: [16,32] f32 -> d: [32,32] = a @ b
+__global__ void mfma_probe_32x32x16_kernel(const float* a, const float* b,
+                                           float* d) {
+  const int lane = threadIdx.x & 63;
+  const int hi = lane >> 5, ln31 = lane & 31;
+  bf16x8 af, bf;
+#pragma unroll
+  for (int m = 0; m < 8; ++m) {
+    af[m] = (__bf16)a[ln31 * 16 + hi * 8 + m];       // A[row=ln31][k=hi*8+m]
+    bf[m] = (__bf16)b[(hi * 8 + m) * 32 + ln31];     // B[k][col=ln31]
+  }
+  floatx16 acc;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    d[row * 32 + ln31] = acc[r];
+  }
+}
+
+torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b) {
+  auto d = torch::zeros({32, 32}, a.options());
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(mfma_probe_32x32x16_kernel, dim3(1), dim3(64), 0, stream,
+                     a.data_ptr<float>(), b.data_ptr<float>(),
+                     d.data_ptr<float>());
+  HIP_CHECK_KERNEL();
+  return d;
+}

@@ -1,0 +1,53 @@
+// Python bindings for the modalities_amd HIP op library (gfx950-only).
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor w, torch::Tensor invrms);
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t,
+                       bool neg);
+torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u);
+std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dout, torch::Tensor g,
+                                        torch::Tensor u);
+std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
+                                             torch::Tensor targets,
+                                             long ignore_index);
+torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
+                                torch::Tensor lse, torch::Tensor scale,
+                                long ignore_index);
+void fused_adamw_masked(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                        torch::Tensor v, torch::Tensor wd_mask, double lr,
+                        double beta1, double beta2, double eps, double wd,
+                        double bc1, double bc2);
+void fused_adamw(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
+                 std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                 double lr, double beta1, double beta2, double eps, double wd,
+                 double bc1, double bc2);
+torch::Tensor multi_tensor_sqsum(std::vector<torch::Tensor> tensors);
+void multi_tensor_scale(std::vector<torch::Tensor> tensors, torch::Tensor scale);
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, bool causal);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    bool causal);
+torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (K5)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (K5)");
+  m.def("rope_fwd", &rope_fwd, "RoPE rotate-half (K4); neg=true => inverse");
+  m.def("silu_mul_fwd", &silu_mul_fwd, "silu(g)*u forward (K6)");
+  m.def("silu_mul_bwd", &silu_mul_bwd, "silu(g)*u backward (K6)");
+  m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE forward (K8)");
+  m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward (K8)");
+  m.def("fused_adamw_masked", &fused_adamw_masked,
+        "AdamW on a flat fp32 shard with per-element wd mask (K9)");
+  m.def("fused_adamw", &fused_adamw, "multi-tensor AdamW (K9)");
+  m.def("multi_tensor_sqsum", &multi_tensor_sqsum, "sum of squares (K10)");
+  m.def("multi_tensor_scale", &multi_tensor_scale, "in-place scale (K10)");
+  m.def("attn_fwd", &attn_fwd, "flash attention forward (K1)");
+  m.def("attn_bwd", &attn_bwd, "flash attention backward (K1)");
+  m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16,
+        "MFMA fragment-layout probe (verification)");
+}

@@ -1,0 +1,165 @@
+// K4 RoPE + K6 SiLU-mul: memory-bound elementwise kernels, bf16 vectorized.
+// RoPE uses HOST-precomputed cos/sin tables (guide Appendix B: on-device
+// trig turns these VALU-bound). Replaces reference RotaryTransform
+// (gpt2_model.py:114-229) and SwiGLU's eager silu*mul (model.py:141-157).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+// ---------------- RoPE (rotate-half) --------------------------------------
+// x: [B, T, H, D] bf16 contiguous; cos/sin: [T, D/2] f32.
+// out_j       = x_j * c_j - x_{j+D/2} * s_j          (j < D/2)
+// out_{j+D/2} = x_{j+D/2} * c_j + x_j * s_j
+// backward = rotation by -theta (neg=true flips the sin sign).
+// One thread handles 2 adjacent (lo, hi) pairs -> 2x ushort2 loads per half.
+__global__ void rope_kernel(const unsigned short* __restrict__ x,
+                            const float* __restrict__ cos_t,
+                            const float* __restrict__ sin_t,
+                            unsigned short* __restrict__ out,
+                            long total_pairs2,  // B*T*H*(D/2)/2
+                            int T, int Hn, int D, int neg) {
+  const int halfD = D / 2;
+  const int pairs_per_head2 = halfD / 2;
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x;
+       idx < total_pairs2; idx += (long)gridDim.x * blockDim.x) {
+    const long head_idx = idx / pairs_per_head2;       // b*T*H + t*H + h
+    const int j2 = (int)(idx % pairs_per_head2) * 2;   // pair start in [0, D/2)
+    const int t = (int)((head_idx / Hn) % T);
+    const long base = head_idx * (long)D;
+
+    const unsigned int* lo_p = reinterpret_cast<const unsigned int*>(x + base + j2);
+    const unsigned int* hi_p =
+        reinterpret_cast<const unsigned int*>(x + base + halfD + j2);
+    unsigned int lo_u = *lo_p, hi_u = *hi_p;
+    float lo0 = bf16_to_f32((unsigned short)(lo_u & 0xffff));
+    float lo1 = bf16_to_f32((unsigned short)(lo_u >> 16));
+    float hi0 = bf16_to_f32((unsigned short)(hi_u & 0xffff));
+    float hi1 = bf16_to_f32((unsigned short)(hi_u >> 16));
+
+    const float2 cs0 = make_float2(cos_t[t * halfD + j2], sin_t[t * halfD + j2]);
+    const float2 cs1 = make_float2(cos_t[t * halfD + j2 + 1],
+                                   sin_t[t * halfD + j2 + 1]);
+    const float s0 = neg ? -cs0.y : cs0.y;
+    const float s1 = neg ? -cs1.y : cs1.y;
+
+    float o_lo0 = lo0 * cs0.x - hi0 * s0;
+    float o_lo1 = lo1 * cs1.x - hi1 * s1;
+    float o_hi0 = hi0 * cs0.x + lo0 * s0;
+    float o_hi1 = hi1 * cs1.x + lo1 * s1;
+
+    unsigned int* out_lo = reinterpret_cast<unsigned int*>(out + base + j2);
+    unsigned int* out_hi = reinterpret_cast<unsigned int*>(out + base + halfD + j2);
+    *out_lo = (unsigned int)f32_to_bf16(o_lo0) |
+              ((unsigned int)f32_to_bf16(o_lo1) << 16);
+    *out_hi = (unsigned int)f32_to_bf16(o_hi0) |
+              ((unsigned int)f32_to_bf16(o_hi1) << 16);
+  }
+}
+
+// ---------------- SiLU * mul ----------------------------------------------
+__global__ void silu_mul_fwd_kernel(const unsigned short* __restrict__ g,
+                                    const unsigned short* __restrict__ u,
+                                    unsigned short* __restrict__ out, long n8) {
+  const shortx8* gv = reinterpret_cast<const shortx8*>(g);
+  const shortx8* uv = reinterpret_cast<const shortx8*>(u);
+  shortx8* ov = reinterpret_cast<shortx8*>(out);
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    shortx8 gg = gv[i], uu = uv[i], oo;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf16_to_f32((unsigned short)gg[j]);
+      float uf = bf16_to_f32((unsigned short)uu[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      oo[j] = (short)f32_to_bf16(gf * sig * uf);
+    }
+    ov[i] = oo;
+  }
+}
+
+__global__ void silu_mul_bwd_kernel(const unsigned short* __restrict__ dout,
+                                    const unsigned short* __restrict__ g,
+                                    const unsigned short* __restrict__ u,
+                                    unsigned short* __restrict__ dg,
+                                    unsigned short* __restrict__ du, long n8) {
+  const shortx8* dv = reinterpret_cast<const shortx8*>(dout);
+  const shortx8* gv = reinterpret_cast<const shortx8*>(g);
+  const shortx8* uv = reinterpret_cast<const shortx8*>(u);
+  shortx8* dgv = reinterpret_cast<shortx8*>(dg);
+  shortx8* duv = reinterpret_cast<shortx8*>(du);
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    shortx8 dd = dv[i], gg = gv[i], uu = uv[i], og, ou;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float df = bf16_to_f32((unsigned short)dd[j]);
+      float gf = bf16_to_f32((unsigned short)gg[j]);
+      float uf = bf16_to_f32((unsigned short)uu[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      float silu = gf * sig;
+      float dsilu = sig * (1.f + gf * (1.f - sig));
+      og[j] = (short)f32_to_bf16(df * uf * dsilu);
+      ou[j] = (short)f32_to_bf16(df * silu);
+    }
+    dgv[i] = og;
+    duv[i] = ou;
+  }
+}
+
+int grid_for(long work, int block) {
+  long g = (work + block - 1) / block;
+  return (int)min(g, (long)(256 * 8));  // cap + grid-stride (guide G11)
+}
+
+}  // namespace
+
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t,
+                       bool neg) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 4
+              && x.is_contiguous(), "rope: x must be contiguous bf16 [B,T,H,D]");
+  const int B = x.size(0), T = x.size(1), Hn = x.size(2), D = x.size(3);
+  TORCH_CHECK(D % 4 == 0, "rope: head_dim must be divisible by 4");
+  TORCH_CHECK(cos_t.size(0) >= T && cos_t.size(1) == D / 2);
+  TORCH_CHECK(cos_t.dtype() == torch::kFloat32 && cos_t.is_contiguous());
+  auto out = torch::empty_like(x);
+  long total2 = (long)B * T * Hn * (D / 2) / 2;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(rope_kernel, dim3(grid_for(total2, 256)), dim3(256), 0, stream,
+                     (const unsigned short*)x.data_ptr(),
+                     cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+                     (unsigned short*)out.data_ptr(), total2, T, Hn, D, neg ? 1 : 0);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
+torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u) {
+  TORCH_CHECK(g.is_cuda() && g.dtype() == torch::kBFloat16 && g.is_contiguous());
+  TORCH_CHECK(g.numel() % 8 == 0, "silu_mul: numel must be divisible by 8");
+  auto out = torch::empty_like(g);
+  long n8 = g.numel() / 8;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(silu_mul_fwd_kernel, dim3(grid_for(n8, 256)), dim3(256), 0,
+                     stream, (const unsigned short*)g.data_ptr(),
+                     (const unsigned short*)u.data_ptr(),
+                     (unsigned short*)out.data_ptr(), n8);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
+std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dout, torch::Tensor g,
+                                        torch::Tensor u) {
+  auto dg = torch::empty_like(g);
+  auto du = torch::empty_like(u);
+  long n8 = g.numel() / 8;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(silu_mul_bwd_kernel, dim3(grid_for(n8, 256)), dim3(256), 0,
+                     stream, (const unsigned short*)dout.data_ptr(),
+                     (const unsigned short*)g.data_ptr(),
+                     (const unsigned short*)u.data_ptr(),
+                     (unsigned short*)dg.data_ptr(),
+                     (unsigned short*)du.data_ptr(), n8);
+  HIP_CHECK_KERNEL();
+  return {dg, du};
+}
