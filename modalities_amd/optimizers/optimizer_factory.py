@@ -42,6 +42,8 @@ class ShardedAdamW(torch.optim.Optimizer):
         lr, (beta1, beta2) = group["lr"], group["betas"]
         eps, wd = group["eps"], group["weight_decay"]
         for u in self.sharded_model.units:
+            if not u.grad_fresh:
+                continue  # no grads reduced for this unit this step
             st = self.state[u.master_shard]
             st["step"] += 1
             step = st["step"]

@@ -105,7 +105,10 @@ class FlatParamUnit:
         self.wd_mask_shard = flat_mask[s:s + self.shard_numel].to(device)
 
         self.master_shard = torch.zeros(self.shard_numel, dtype=torch.float32, device=device)
-        self.grad_shard = torch.zeros(self.shard_numel, dtype=torch.float32, device=device)
+        # grad_shard carries no valid data until the first reduce of a step
+        # OVERWRITES it (grad_fresh flag) — avoids a full fp32 fill per step.
+        self.grad_shard = torch.empty(self.shard_numel, dtype=torch.float32, device=device)
+        self.grad_fresh = False  # True once this step's first reduce landed
         self.bf16_shard = torch.zeros(self.shard_numel, dtype=param_dtype, device=device)
         self.full_buf: Optional[torch.Tensor] = None
         self.grad_full: Optional[torch.Tensor] = None
@@ -142,15 +145,17 @@ class FlatParamUnit:
     def gather(self, streams: _CommStreams):
         if self.is_gathered:
             return
+        if self.world == 1:
+            # shard IS the full tensor: alias, no copy, no event
+            self.full_buf = self.bf16_shard
+            self.is_gathered = True
+            return
         with streams.on(streams.gather):
             if streams.on_gpu:
                 streams.gather.wait_stream(torch.cuda.current_stream(self.device))
             self.full_buf = torch.empty(self.total_numel, dtype=self.param_dtype,
                                         device=self.device)
-            if self.world > 1:
-                dist.all_gather_into_tensor(self.full_buf, self.bf16_shard, group=self.group)
-            else:
-                self.full_buf.copy_(self.bf16_shard)
+            dist.all_gather_into_tensor(self.full_buf, self.bf16_shard, group=self.group)
             if streams.on_gpu:
                 self._gather_event = torch.cuda.Event()
                 self._gather_event.record(streams.gather)
@@ -199,9 +204,16 @@ class FlatParamUnit:
                     dist.all_reduce(gf, group=self.group)
                     s = self.rank * self.shard_numel
                     out = gf[s:s + self.shard_numel]
-                self.grad_shard.add_(out.float(), alpha=1.0 / self.world)
             else:
-                self.grad_shard.add_(gf.float())
+                out = gf
+            scale = 1.0 / self.world
+            if self.grad_fresh:
+                self.grad_shard.add_(out, alpha=scale)
+            else:  # first reduce of the step overwrites (grad_shard is stale)
+                self.grad_shard.copy_(out)
+                if scale != 1.0:
+                    self.grad_shard.mul_(scale)
+                self.grad_fresh = True
             if streams.on_gpu:
                 self._reduce_event = torch.cuda.Event()
                 self._reduce_event.record(streams.reduce)
@@ -217,13 +229,9 @@ class FlatParamUnit:
 
     @torch.no_grad()
     def publish_master(self):
-        self.bf16_shard.copy_(self.master_shard.to(self.param_dtype))
-        if self.full_buf is not None:
-            # keep resident copy coherent without a fresh gather at world==1
-            if self.world == 1:
-                self.full_buf.copy_(self.bf16_shard)
-            else:
-                self.free_full()
+        self.bf16_shard.copy_(self.master_shard)  # single cast kernel
+        if self.world > 1 and self.full_buf is not None:
+            self.free_full()  # stale after the update; regathered next fwd
 
 
 class _UnshardBackwardAnchor(torch.autograd.Function):
@@ -437,8 +445,9 @@ class XGMIShardedModel(nn.Module):
 
     @torch.no_grad()
     def zero_grad_shards(self):
+        # no fill: the next step's first reduce overwrites (grad_fresh)
         for u in self.units:
-            u.grad_shard.zero_()
+            u.grad_fresh = False
 
     @torch.no_grad()
     def publish_master(self):
@@ -449,13 +458,16 @@ class XGMIShardedModel(nn.Module):
     def clip_grad_norm_(self, max_norm: Optional[float],
                         norm_type: float = 2.0) -> torch.Tensor:
         from modalities_amd.ops.adamw import multi_tensor_l2norm, multi_tensor_scale_
-        local = multi_tensor_l2norm([u.grad_shard for u in self.units]) ** 2
+        fresh = [u.grad_shard for u in self.units if u.grad_fresh]
+        local = multi_tensor_l2norm(fresh) ** 2
+        if local.device != torch.device(self.device):
+            local = local.to(self.device)
         if self.world > 1:
             dist.all_reduce(local, group=self.group)
         total = local.sqrt()
-        if max_norm is not None and max_norm > 0:
+        if max_norm is not None and max_norm > 0 and fresh:
             clip = (max_norm / (total + 1e-6)).clamp(max=1.0)
-            multi_tensor_scale_([u.grad_shard for u in self.units], clip)
+            multi_tensor_scale_(fresh, clip)
         return total
 
     # -- state (sharded checkpoints) --------------------------------------
