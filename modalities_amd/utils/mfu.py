@@ -31,6 +31,9 @@ def detect_device_peak_flops() -> Optional[float]:
             return peak
     if "GFX950" in name or "355" in name:
         return PEAK_FLOPS_16BIT["MI355X"]
+    if torch.version.hip is not None:
+        # unknown ROCm device string: this framework targets MI355X
+        return PEAK_FLOPS_16BIT["MI355X"]
     return None
 
 

@@ -6,7 +6,11 @@ Reports TFLOP/s against causal-attention flop counts (0.5 * 4*B*T^2*H*D per
 matmul pair)."""
 
 import argparse
+import sys
 import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
 
