@@ -20,11 +20,15 @@ namespace attnbwd {
 
 constexpr int KVBLK = 64;  // dQ kernel kv tile
 
-__device__ __forceinline__ int swz(int row, int col) {   // 64/128-wide rows
-  return col ^ ((row & 7) << 3);
+// Swizzles (see attention_fwd.hip for the derivation):
+__device__ __forceinline__ int swz(int row, int col) {   // row-major tiles
+  return col ^ (((row & 3) | (((row >> 4) & 1) << 2)) << 3);
 }
-__device__ __forceinline__ int swz32(int row, int col) {  // 32-wide rows
-  return col ^ ((row & 3) << 3);
+__device__ __forceinline__ int swz_t(int d, int r) {     // transposed, 64-wide
+  return r ^ ((((d >> 3) & 3) << 1 | ((d >> 1) & 1)) << 3);
+}
+__device__ __forceinline__ int swz32(int d, int r) {     // transposed, 32-wide
+  return r ^ (((((d >> 4) & 1) << 1) | ((d >> 1) & 1)) << 3);
 }
 __device__ __forceinline__ int crow(int reg, int hi) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
@@ -160,7 +164,7 @@ struct DQKernel {
 #pragma unroll
           for (int m = 0; m < 8; ++m) {
             const int d = c + m;
-            sm->kt[d * KVBLK + swz(d, r)] = (unsigned short)kk[m];
+            sm->kt[d * KVBLK + swz_t(d, r)] = (unsigned short)kk[m];
           }
         }
       }
@@ -208,7 +212,7 @@ struct DQKernel {
           const int d = dblk * 32 + ln31;
           const int col = hi * 8 + 16 * s;
           bf16x8 kta = *reinterpret_cast<const bf16x8*>(
-              &sm->kt[d * KVBLK + swz(d, col)]);
+              &sm->kt[d * KVBLK + swz_t(d, col)]);
           acc_dq[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               kta, dsfrag[s], acc_dq[dblk], 0, 0, 0);
         }

@@ -30,9 +30,18 @@ constexpr int NWAVES = 4;    // waves per workgroup
 constexpr int WG_Q = QBLK * NWAVES;  // 128 q rows per workgroup
 constexpr int KVBLK = 64;    // kv rows per LDS tile
 
+// Row-major tile swizzle (read rows differ within {0..3,16..19}+32k b128
+// lane groups): key distinct over that set -> conflict-free reads; staging
+// writes are 8-lane same-row groups with distinct 16B slots -> conflict-free.
 __device__ __forceinline__ int swz(int row, int col) {
-  // element-index XOR swizzle for bf16 tiles read with ds_read_b128
-  return col ^ ((row & 7) << 3);
+  return col ^ (((row & 3) | (((row >> 4) & 1) << 2)) << 3);
+}
+
+// Transposed-tile swizzle (scalar writes walk d = 8j+m at fixed kv row, so
+// the key needs (d>>3) entropy; reads need d1/d4): writes ~4-way (was 16),
+// reads ~1-way.
+__device__ __forceinline__ int swz_t(int d, int r) {
+  return r ^ ((((d >> 3) & 3) << 1 | ((d >> 1) & 1)) << 3);
 }
 
 // C/D register map of v_mfma_f32_32x32x16_bf16 (guide §3):
@@ -132,7 +141,7 @@ struct AttnFwdKernel {
 #pragma unroll
           for (int m = 0; m < 8; ++m) {
             const int d = c + m;
-            sm->vt[d * KVBLK + swz(d, r)] = (unsigned short)vv[m];
+            sm->vt[d * KVBLK + swz_t(d, r)] = (unsigned short)vv[m];
           }
         }
       }
@@ -220,7 +229,7 @@ struct AttnFwdKernel {
           const int d = dblk * 32 + ln31;
           const int col = hi * 8 + 16 * s;  // kv index within tile
           bf16x8 va = *reinterpret_cast<const bf16x8*>(
-              &sm->vt[d * KVBLK + swz(d, col)]);
+              &sm->vt[d * KVBLK + swz_t(d, col)]);
           acc_o[dblk] =
               __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, pfrag[s], acc_o[dblk],
                                                       0, 0, 0);
