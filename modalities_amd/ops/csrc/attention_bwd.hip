@@ -225,9 +225,16 @@ struct DQKernel {
 #pragma unroll
       for (int dblk = 0; dblk < NDBLK; ++dblk)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int d = dblk * 32 + crow(r, hi);
-          dqr[d] = f32_to_bf16(acc_dq[dblk][r]);
+        for (int r4 = 0; r4 < 4; ++r4) {  // pack 4 consecutive d per store
+          unsigned int w[2];
+#pragma unroll
+          for (int i = 0; i < 2; ++i) {
+            const int r = r4 * 4 + 2 * i;
+            w[i] = (unsigned int)f32_to_bf16(acc_dq[dblk][r]) |
+                   ((unsigned int)f32_to_bf16(acc_dq[dblk][r + 1]) << 16);
+          }
+          const int d = dblk * 32 + crow(r4 * 4, hi);
+          *reinterpret_cast<uint2*>(dqr + d) = make_uint2(w[0], w[1]);
         }
     }
   }

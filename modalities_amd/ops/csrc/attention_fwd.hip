@@ -51,7 +51,11 @@ __device__ __forceinline__ int crow(int reg, int hi) {
 }
 
 // A/B fragment k-index for 32x32x16: lane provides elements k = hi*8 + m.
-template <int D_>
+// ABL (ablation, guide mistake #8 "ablate before optimizing"):
+//   0 = full kernel; 1 = skip V-transpose staging; 2 = skip all staging;
+//   3 = skip softmax VALU (P := raw S). Modes 1-3 give WRONG results and
+//   exist only for cost attribution.
+template <int D_, int ABL = 0>
 struct AttnFwdKernel {
   static constexpr int D = D_;
   static constexpr int NDSTEP = D / 16;   // QK^T contraction steps
@@ -128,7 +132,8 @@ struct AttnFwdKernel {
           } else {
             for (int m = 0; m < 8; ++m) val[m] = 0;
           }
-          *reinterpret_cast<shortx8*>(&sm->k[r * D + swz(r, c)]) = val;
+          if constexpr (ABL < 2)
+            *reinterpret_cast<shortx8*>(&sm->k[r * D + swz(r, c)]) = val;
           // V: read the same global layout, write transposed (scalar LDS
           // stores — v1; upgrade path: ds_read_b64_tr_b16 per guide T10)
           shortx8 vv;
@@ -138,10 +143,14 @@ struct AttnFwdKernel {
           } else {
             for (int m = 0; m < 8; ++m) vv[m] = 0;
           }
+          if constexpr (ABL < 1) {
 #pragma unroll
-          for (int m = 0; m < 8; ++m) {
-            const int d = c + m;
-            sm->vt[d * KVBLK + swz_t(d, r)] = (unsigned short)vv[m];
+            for (int m = 0; m < 8; ++m) {
+              const int d = c + m;
+              sm->vt[d * KVBLK + swz_t(d, r)] = (unsigned short)vv[m];
+            }
+          } else {
+            asm volatile("" :: "v"(vv[0]));  // keep loads alive (rule 17)
           }
         }
       }
@@ -163,6 +172,11 @@ struct AttnFwdKernel {
 
       // ---- scale + causal mask; per-lane P rows (q = ln31) --------------
       float p[32];
+      if constexpr (ABL == 3) {  // softmax cost ablation: P := raw S
+#pragma unroll
+        for (int r = 0; r < 16; ++r) { p[r] = s0[r]; p[16 + r] = s1[r]; }
+        l_run += 1.f;
+      } else {
       float tmax = -INFINITY;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -194,6 +208,7 @@ struct AttnFwdKernel {
       for (int dblk = 0; dblk < NDBLK; ++dblk)
 #pragma unroll
         for (int r = 0; r < 16; ++r) acc_o[dblk][r] *= alpha;
+      }  // ABL != 3
 
       // ---- P (f32, S-layout) -> PV A/B fragment (bf16) -------------------
       // Lane holds P[q=ln31][kv in crow set]. The PV B-operand needs
@@ -243,15 +258,22 @@ struct AttnFwdKernel {
     const float inv_l = 1.f / l_safe;
     if (qg < T) {
       unsigned short* orow = o + q_base + (long)qg * Hq * D;
-      // lane holds O^T[d = dblk*32 + crow(r,hi)][q = ln31]?? NO:
-      // PV output D[m=d_row][n=q]: col = ln31 = q; row = crow(r,hi) = d idx
-      // within the 32-col block -> this lane's values are for ITS q row.(ok)
+      // PV output: col = ln31 = this lane's q row; rows crow(r,hi) = d.
+      // crow runs in 4-consecutive groups (r&3), so pack 4 bf16 -> one
+      // 8-byte store (16 stores/lane instead of 64 scalar).
 #pragma unroll
       for (int dblk = 0; dblk < NDBLK; ++dblk) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int d = dblk * 32 + crow(r, hi);
-          orow[d] = f32_to_bf16(acc_o[dblk][r] * inv_l);
+        for (int r4 = 0; r4 < 4; ++r4) {
+          unsigned int w[2];
+#pragma unroll
+          for (int i = 0; i < 2; ++i) {
+            const int r = r4 * 4 + 2 * i;
+            w[i] = (unsigned int)f32_to_bf16(acc_o[dblk][r] * inv_l) |
+                   ((unsigned int)f32_to_bf16(acc_o[dblk][r + 1] * inv_l) << 16);
+          }
+          const int d = dblk * 32 + crow(r4 * 4, hi);
+          *reinterpret_cast<uint2*>(orow + d) = make_uint2(w[0], w[1]);
         }
       }
       if (hi == 0) {
@@ -262,12 +284,12 @@ struct AttnFwdKernel {
   }
 };
 
-template <int D>
+template <int D, int ABL = 0>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
     unsigned short* o, float* lse, int B, int T, int Hq, int Hkv, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  AttnFwdKernel<D>::run(q, k, v, o, lse, B, T, Hq, Hkv, scale, smem_raw);
+  AttnFwdKernel<D, ABL>::run(q, k, v, o, lse, B, T, Hq, Hkv, scale, smem_raw);
 }
 
 }  // namespace attn
@@ -303,6 +325,38 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                        (const unsigned short*)v.data_ptr(),
                        (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
                        B, T, Hq, Hkv, scale);
+  }
+  HIP_CHECK_KERNEL();
+  return {o, lse};
+}
+
+// Ablation entry (D=128 only): mode 0 full / 1 no-Vt-staging / 2 no-staging
+// / 3 no-softmax. Modes>0 produce wrong outputs; timing only.
+std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
+                                           torch::Tensor v, long mode) {
+  const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 128);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, Hq, T}, q.options().dtype(torch::kFloat32));
+  const float scale = 1.0f / sqrtf((float)D);
+  dim3 grid((T + attn::WG_Q - 1) / attn::WG_Q, Hq, B);
+  auto stream = at::cuda::getCurrentHIPStream();
+  const size_t smem = sizeof(unsigned short) * (attn::KVBLK * D + D * attn::KVBLK);
+  auto launch = [&](auto kfn) {
+    hipLaunchKernelGGL(kfn, grid, dim3(256), smem, stream,
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
+                       B, T, Hq, Hkv, scale);
+  };
+  switch (mode) {
+    case 0: launch(attn::attn_fwd_kernel<128, 0>); break;
+    case 1: launch(attn::attn_fwd_kernel<128, 1>); break;
+    case 2: launch(attn::attn_fwd_kernel<128, 2>); break;
+    case 3: launch(attn::attn_fwd_kernel<128, 3>); break;
+    default: TORCH_CHECK(false, "mode 0-3");
   }
   HIP_CHECK_KERNEL();
   return {o, lse};

@@ -32,6 +32,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor o, torch::Tensor lse,
                                     bool causal);
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
+std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
+                                           torch::Tensor v, long mode);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (K5)");
@@ -50,4 +52,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd, "flash attention backward (K1)");
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16,
         "MFMA fragment-layout probe (verification)");
+  m.def("attn_fwd_ablate", &attn_fwd_ablate,
+        "attention fwd cost-attribution ablation (timing only)");
 }

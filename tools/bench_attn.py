@@ -23,7 +23,8 @@ def main():
     p.add_argument("--Hkv", type=int, default=20)
     p.add_argument("--D", type=int, default=128)
     p.add_argument("--iters", type=int, default=20)
-    p.add_argument("--what", choices=["fwd", "bwd", "both"], default="both")
+    p.add_argument("--what", choices=["fwd", "bwd", "both", "ablate"],
+                   default="both")
     args = p.parse_args()
 
     from modalities_amd.ops.backend import hip_ext
@@ -50,6 +51,19 @@ def main():
         torch.cuda.synchronize()
         dt = (time.perf_counter() - t0) / args.iters
         print(f"fwd: {dt*1e3:8.3f} ms  {flops_fwd/dt/1e12:8.1f} TF/s")
+
+    if args.what == "ablate":
+        for mode, name in [(0, "full"), (1, "no-Vt-stage"), (2, "no-stage"),
+                           (3, "no-softmax")]:
+            ext.attn_fwd_ablate(q, k, v, mode)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(args.iters):
+                ext.attn_fwd_ablate(q, k, v, mode)
+            torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / args.iters
+            print(f"ablate {name:12s}: {dt*1e3:8.3f} ms  "
+                  f"{flops_fwd/dt/1e12:8.1f} TF/s")
 
     if args.what in ("bwd", "both"):
         t0 = time.perf_counter()
