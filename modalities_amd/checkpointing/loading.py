@@ -1,0 +1,105 @@
+"""Sharded checkpoint loading with cross-world-size resharding.
+
+Capability parity with the reference's DCP load path (reference:
+src/modalities/checkpointing/fsdp/fsdp_checkpoint_loading.py:104-133), native
+to the flat-shard engine: because every unit's parameters live in ONE
+deterministic flat fp32 vector, resharding to a different world size is pure
+index arithmetic — each rank streams only the saved rank-files that overlap
+its new shard range (one saved file in memory at a time), no global gather.
+
+Padding note: a unit's flat length is padded to ``world*64``, so the padded
+total differs between world sizes; all slicing here happens in LOGICAL
+coordinates [0, logical_numel) recorded in ``meta.json``.
+"""
+
+import json
+from pathlib import Path
+
+import torch
+
+from modalities_amd.checkpointing.app_state import AppState
+
+
+def read_last_checkpoint_info(experiment_checkpoint_root: Path) -> Path:
+    with open(Path(experiment_checkpoint_root) / "last_checkpoint_info.json",
+              encoding="utf-8") as f:
+        return Path(json.load(f)["checkpoint_folder_path"])
+
+
+def read_checkpoint_meta(folder: Path) -> dict:
+    with open(Path(folder) / "meta.json", encoding="utf-8") as f:
+        return json.load(f)
+
+
+class ShardedCheckpointLoading:
+    def __init__(self, global_rank: int):
+        self.global_rank = global_rank
+
+    @torch.no_grad()
+    def load_checkpoint_(self, app_state: AppState, folder: Path) -> dict:
+        """Load (in-place) a checkpoint folder into ``app_state``; reshard if
+        the saved world size differs. Returns the saved training-progress
+        metadata dict."""
+        folder = Path(folder)
+        meta = read_checkpoint_meta(folder)
+        saved_world = meta["world_size"]
+        saved_layout = meta["shard_layout"]
+
+        model = app_state.model
+        new_world, new_rank = model.world, model.rank
+
+        # target CPU staging buffers for this rank's new shards
+        targets: dict[str, torch.Tensor] = {}
+        for u in model.units:
+            targets[f"model.{u.name}.master_shard"] = torch.zeros(
+                u.shard_numel, dtype=torch.float32)
+            targets[f"optim.{u.name}.exp_avg"] = torch.zeros(
+                u.shard_numel, dtype=torch.float32)
+            targets[f"optim.{u.name}.exp_avg_sq"] = torch.zeros(
+                u.shard_numel, dtype=torch.float32)
+
+        for saved_rank in range(saved_world):
+            if not self._rank_file_overlaps(model, saved_layout, saved_rank,
+                                            saved_world, new_rank):
+                continue
+            shards = torch.load(folder / f"shards_rank_{saved_rank}.pt",
+                                map_location="cpu", weights_only=True)
+            for u in model.units:
+                lay = saved_layout[u.name]
+                logical = lay["logical_numel"]
+                ss = lay["shard_numel"]
+                s_lo, s_hi = saved_rank * ss, min((saved_rank + 1) * ss, logical)
+                n_lo = new_rank * u.shard_numel
+                n_hi = min((new_rank + 1) * u.shard_numel, logical)
+                lo, hi = max(s_lo, n_lo), min(s_hi, n_hi)
+                if lo >= hi:
+                    continue
+                for prefix in ("model", "optim"):
+                    for suffix in (("master_shard",) if prefix == "model"
+                                   else ("exp_avg", "exp_avg_sq")):
+                        key = f"{prefix}.{u.name}.{suffix}"
+                        if key not in shards:
+                            continue
+                        src = shards[key][lo - s_lo:hi - s_lo]
+                        targets[key][lo - n_lo:hi - n_lo].copy_(src)
+            del shards
+
+        optim_steps = meta["replicated_state"].get("optim_steps", {})
+        app_state.load_shard_state(
+            {k: v for k, v in targets.items()}, optim_steps)
+        app_state.load_replicated_state(meta["replicated_state"])
+        return meta["training_progress"]
+
+    @staticmethod
+    def _rank_file_overlaps(model, saved_layout, saved_rank, saved_world,
+                            new_rank) -> bool:
+        for u in model.units:
+            lay = saved_layout[u.name]
+            ss = lay["shard_numel"]
+            logical = lay["logical_numel"]
+            s_lo, s_hi = saved_rank * ss, min((saved_rank + 1) * ss, logical)
+            n_lo = new_rank * u.shard_numel
+            n_hi = min((new_rank + 1) * u.shard_numel, logical)
+            if max(s_lo, n_lo) < min(s_hi, n_hi):
+                return True
+        return False
