@@ -28,7 +28,12 @@ __device__ __forceinline__ int swz_t(int d, int r) {     // transposed, 64-wide
   return r ^ ((((d >> 3) & 3) << 1 | ((d >> 1) & 1)) << 3);
 }
 __device__ __forceinline__ int swz32(int d, int r) {     // transposed, 32-wide
-  return r ^ (((((d >> 4) & 1) << 1) | ((d >> 1) & 1)) << 3);
+  // Read pattern (dV/dK B-operands): lanes d = dblk*32 + ln31 all read the
+  // same r-range; row stride is 64 B so bank(dword) = (d*16 + r/2) % 64 —
+  // only d&3 separates lanes without a swizzle. Injecting (d>>2)&3 into the
+  // 4 dword slots makes lanes d, d+4, d+8, d+12 land on distinct slots
+  // (<=2-way from d vs d+16 only).
+  return r ^ (((d >> 2) & 3) << 3);
 }
 __device__ __forceinline__ int crow(int reg, int hi) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
@@ -250,16 +255,91 @@ struct DQKernel {
 //   dS' = scale * P .* (dP - delta[q])       (lse/delta loaded per q row)
 //   dV[kv][d] += mfma(A=frags(P),   B=dOt)   col = d = ln31, rows kv = crow
 //   dK[kv][d] += mfma(A=frags(dS'), B=Qt)
-template <int D>
+// ABL (ablation; guide mistake #8): 0 full; 1 skip transposed qt/dot image
+// writes; 2 skip ALL LDS staging writes; 3 skip lse/delta scalar loads;
+// 4 skip dV/dK MFMAs; 5 skip S/dP MFMAs + softmax (staging only).
+// Modes >0 give WRONG results; timing-only.
+template <int D, int ABL = 0>
 struct DKDVKernel {
   static constexpr int NDSTEP = D / 16;
   static constexpr int NDBLK = D / 32;
-  struct Smem {
+  // Double-buffered q-tile: Q/dO row-major swizzled (S/dP A-operands),
+  // Q^T/dO^T transposed swizzled (dK/dV B-operands), and the per-row
+  // softmax stats staged once per tile (read back as LDS broadcasts —
+  // the v1 per-reg scalar GLOBAL loads of lse/delta were 26% of kernel
+  // time by ablation).
+  struct Tile {
     unsigned short qrow[32 * D];
     unsigned short dorow[32 * D];
     unsigned short qt[D * 32];
     unsigned short dot[D * 32];
+    float lse[32];
+    float delta[32];
   };
+  struct Smem {
+    Tile t[2];
+  };
+
+  // One thread's staged registers for a tile (NCH chunks x 8 bf16 x {q,do};
+  // 32*D/8 chunks over 256 threads: D=128 -> 2/thread, D=64 -> 1).
+  static constexpr int NCH = (32 * D / 8 + 255) / 256;
+  struct StageRegs {
+    shortx8 qq[NCH], dd[NCH];
+    float stat;  // thread's lse/delta element (tid < 64)
+  };
+
+  static __device__ void stage_load(StageRegs& r, const unsigned short* q,
+                                    const unsigned short* dout, long q_base,
+                                    const float* lse_h, const float* delta_h,
+                                    int q0, int T, int Hq) {
+    const int tid = threadIdx.x;
+#pragma unroll
+    for (int it = 0; it < NCH; ++it) {
+      const int i = tid + it * 256;
+      const int row = (i * 8) / D, c = (i * 8) % D;
+      if (row >= 32) continue;
+      if (q0 + row < T) {
+        r.qq[it] = *reinterpret_cast<const shortx8*>(
+            q + q_base + (long)(q0 + row) * Hq * D + c);
+        r.dd[it] = *reinterpret_cast<const shortx8*>(
+            dout + q_base + (long)(q0 + row) * Hq * D + c);
+      } else {
+#pragma unroll
+        for (int m = 0; m < 8; ++m) { r.qq[it][m] = 0; r.dd[it][m] = 0; }
+      }
+    }
+    if (tid < 32) {
+      r.stat = (q0 + tid < T) ? lse_h[q0 + tid] : 0.f;
+    } else if (tid < 64) {
+      r.stat = (q0 + tid - 32 < T) ? delta_h[q0 + tid - 32] : 0.f;
+    }
+  }
+
+  static __device__ void stage_write(const StageRegs& r, Tile* tl) {
+    const int tid = threadIdx.x;
+#pragma unroll
+    for (int it = 0; it < NCH; ++it) {
+      const int i = tid + it * 256;
+      const int row = (i * 8) / D, c = (i * 8) % D;
+      if (row >= 32) continue;
+      if constexpr (ABL < 2) {
+        *reinterpret_cast<shortx8*>(&tl->qrow[row * D + swz(row, c)]) = r.qq[it];
+        *reinterpret_cast<shortx8*>(&tl->dorow[row * D + swz(row, c)]) = r.dd[it];
+      } else {
+        asm volatile("" :: "v"(r.qq[it][0]), "v"(r.dd[it][0]));
+      }
+      if constexpr (ABL < 1) {
+#pragma unroll
+        for (int m = 0; m < 8; ++m) {
+          const int d = c + m;
+          tl->qt[d * 32 + swz32(d, row)] = (unsigned short)r.qq[it][m];
+          tl->dot[d * 32 + swz32(d, row)] = (unsigned short)r.dd[it][m];
+        }
+      }
+    }
+    if (tid < 32) tl->lse[tid] = r.stat;
+    else if (tid < 64) tl->delta[tid - 32] = r.stat;
+  }
 
   static __device__ void run(const unsigned short* q, const unsigned short* k,
                              const unsigned short* v, const unsigned short* dout,
@@ -301,76 +381,125 @@ struct DKDVKernel {
 #pragma unroll
       for (int r = 0; r < 16; ++r) { acc_dk[dblk][r] = 0.f; acc_dv[dblk][r] = 0.f; }
 
+    // Flattened (hq, qtile) iteration space, software-pipelined two deep:
+    // iter t computes tile t from LDS buf[t&1] while tile t+1's staged
+    // registers are written to buf[(t+1)&1] and tile t+2's global loads are
+    // issued (T14 issue-early/write-late; one barrier per iteration).
     const int first_qtile = kvblk0 / 32;  // causal: q >= kv
     const int n_qtiles = (T + 31) / 32;
+    const int tiles_per_head = n_qtiles - first_qtile;
+    const int n_iter = rep * tiles_per_head;
+    const int q0_first = first_qtile * 32;
 
-    for (int hq = hkv * rep; hq < (hkv + 1) * rep; ++hq) {
-      const long q_base = (((long)b * T) * Hq + hq) * D;
-      const float* lse_h = lse + (((long)b * Hq) + hq) * T;
-      const float* delta_h = delta + (((long)b * Hq) + hq) * T;
+    // Incremental cursors instead of per-iteration div/mod (an integer
+    // division expands to ~25 VALU ops; called 2x per tile it was ~6% of
+    // the issue budget). A cursor = {q0, q_base, lse/delta base} advanced
+    // by one tile per step, wrapping to the next GQA q-head.
+    struct Cursor {
+      int q0;
+      long q_base;
+      const float *lse_h, *delta_h;
+    };
+    auto cursor_init = [&](Cursor& c) {
+      const int hq = hkv * rep;
+      c.q0 = q0_first;
+      c.q_base = (((long)b * T) * Hq + hq) * D;
+      c.lse_h = lse + (((long)b * Hq) + hq) * T;
+      c.delta_h = delta + (((long)b * Hq) + hq) * T;
+    };
+    auto cursor_next = [&](Cursor& c) {
+      c.q0 += 32;
+      if (c.q0 >= n_qtiles * 32) {  // next q-head of the GQA group
+        c.q0 = q0_first;
+        c.q_base += (long)D;
+        c.lse_h += T;
+        c.delta_h += T;
+      }
+    };
 
-      for (int qt = first_qtile; qt < n_qtiles; ++qt) {
-        const int q0 = qt * 32;
-        {  // stage the q tile 4 ways
-          const int tid = threadIdx.x, nthreads = 256;
-          const int total8 = 32 * D / 8;
-          for (int i = tid; i < total8; i += nthreads) {
-            const int r = (i * 8) / D, c = (i * 8) % D;
-            shortx8 qq, dd;
-            if (q0 + r < T) {
-              qq = *reinterpret_cast<const shortx8*>(
-                  q + q_base + (long)(q0 + r) * Hq * D + c);
-              dd = *reinterpret_cast<const shortx8*>(
-                  dout + q_base + (long)(q0 + r) * Hq * D + c);
-            } else {
-#pragma unroll
-              for (int m = 0; m < 8; ++m) { qq[m] = 0; dd[m] = 0; }
-            }
-            *reinterpret_cast<shortx8*>(&sm->qrow[r * D + swz(r, c)]) = qq;
-            *reinterpret_cast<shortx8*>(&sm->dorow[r * D + swz(r, c)]) = dd;
-#pragma unroll
-            for (int m = 0; m < 8; ++m) {
-              const int d = c + m;
-              sm->qt[d * 32 + swz32(d, r)] = (unsigned short)qq[m];
-              sm->dot[d * 32 + swz32(d, r)] = (unsigned short)dd[m];
-            }
-          }
+    StageRegs rA, rB;  // ping-pong staging registers (rule #20: static names)
+    Cursor cc, cp;     // compute cursor (tile t), prefetch cursor (tile t+2)
+    cursor_init(cc);
+    cursor_init(cp);
+    {
+      stage_load(rA, q, dout, cc.q_base, cc.lse_h, cc.delta_h, cc.q0, T, Hq);
+      stage_write(rA, &sm->t[0]);
+      if (n_iter > 1) {
+        cursor_next(cp);
+        stage_load(rA, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0, T, Hq);
+      }
+      __syncthreads();
+    }
+
+    for (int t = 0; t < n_iter; ++t) {
+      Tile* cur = &sm->t[t & 1];
+      // write tile t+1 (in rA) into the other buffer; issue tile t+2 loads
+      if (t + 1 < n_iter) {
+        stage_write(rA, &sm->t[(t + 1) & 1]);
+        if (t + 2 < n_iter) {
+          cursor_next(cp);
+          stage_load(rB, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0, T, Hq);
         }
-        __syncthreads();
+      }
+      const int q0 = cc.q0;
+      cursor_next(cc);
 
-        // S[q][kv] and dP[q][kv] (col = kv = ln31, rows q = crow)
-        floatx16 s_acc, dp_acc;
+      // S[q][kv] and dP[q][kv] (col = kv = ln31, rows q = crow)
+      floatx16 s_acc, dp_acc;
 #pragma unroll
-        for (int r = 0; r < 16; ++r) { s_acc[r] = 0.f; dp_acc[r] = 0.f; }
+      for (int r = 0; r < 16; ++r) { s_acc[r] = 0.f; dp_acc[r] = 0.f; }
+      if constexpr (ABL != 5) {
 #pragma unroll
         for (int s = 0; s < NDSTEP; ++s) {
           const int col = hi * 8 + 16 * s;
           bf16x8 qa = *reinterpret_cast<const bf16x8*>(
-              &sm->qrow[ln31 * D + swz(ln31, col)]);
+              &cur->qrow[ln31 * D + swz(ln31, col)]);
           bf16x8 da = *reinterpret_cast<const bf16x8*>(
-              &sm->dorow[ln31 * D + swz(ln31, col)]);
+              &cur->dorow[ln31 * D + swz(ln31, col)]);
           s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[s], s_acc,
                                                           0, 0, 0);
           dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vfrag[s], dp_acc,
                                                            0, 0, 0);
         }
+      }
 
-        float p_c[16], ds_c[16];
+      float p_c[16], ds_c[16];
+      if constexpr (ABL != 5) {
+        if (q0 >= kvblk0 + 128 && q0 + 32 <= T) {
+          // interior tile: every q row of this tile is > every kv row of
+          // the workgroup and in-range -> no mask compares needed
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const float lq = (ABL != 3) ? cur->lse[crow(r, hi)] : 0.f;
+            const float dl = (ABL != 3) ? cur->delta[crow(r, hi)] : 0.f;
+            const float pv = __expf(s_acc[r] * scale - lq);
+            p_c[r] = pv;
+            ds_c[r] = scale * pv * (dp_acc[r] - dl);
+          }
+        } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int qg = q0 + crow(r, hi);
           const bool ok = (qg < T) && (qg >= kvg) && (kvg < T);
-          const float lq = (qg < T) ? lse_h[qg] : 0.f;
-          const float dl = (qg < T) ? delta_h[qg] : 0.f;
+          // LDS broadcast (same address across the 32 lanes of one half)
+          const float lq = (ABL != 3) ? cur->lse[crow(r, hi)] : 0.f;
+          const float dl = (ABL != 3) ? cur->delta[crow(r, hi)] : 0.f;
           const float pv = ok ? __expf(s_acc[r] * scale - lq) : 0.f;
           p_c[r] = pv;
           ds_c[r] = scale * pv * (dp_acc[r] - dl);
         }
-        bf16x8 pfrag[2], dsfrag[2];
-        c_layout_to_frags(p_c, pfrag, 2);
-        c_layout_to_frags(ds_c, dsfrag, 2);
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) { p_c[r] = 0.f; ds_c[r] = 0.f;
+          asm volatile("" :: "v"(p_c[r]), "v"(ds_c[r])); }
+      }
+      bf16x8 pfrag[2], dsfrag[2];
+      c_layout_to_frags(p_c, pfrag, 2);
+      c_layout_to_frags(ds_c, dsfrag, 2);
 
-        // dV += P^T dO ; dK += dS'^T Q
+      // dV += P^T dO ; dK += dS'^T Q
+      if constexpr (ABL != 4) {
 #pragma unroll
         for (int dblk = 0; dblk < NDBLK; ++dblk) {
 #pragma unroll
@@ -378,27 +507,41 @@ struct DKDVKernel {
             const int d = dblk * 32 + ln31;
             const int col = hi * 8 + 16 * s;
             bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-                &sm->dot[d * 32 + swz32(d, col)]);
+                &cur->dot[d * 32 + swz32(d, col)]);
             bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-                &sm->qt[d * 32 + swz32(d, col)]);
+                &cur->qt[d * 32 + swz32(d, col)]);
             acc_dv[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 pfrag[s], dob, acc_dv[dblk], 0, 0, 0);
             acc_dk[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 dsfrag[s], qb, acc_dk[dblk], 0, 0, 0);
           }
         }
-        __syncthreads();
+      } else {
+        asm volatile("" :: "v"(pfrag[0][0]), "v"(dsfrag[0][0]));
       }
+      // swap staging register sets: rB (loads for t+2) becomes next rA
+      if (t + 2 < n_iter) {
+        rA = rB;
+      }
+      __syncthreads();
     }
 
     // dK/dV output layout: col = ln31 = d within block, rows = crow = kv
-    // local row (NOT this lane's kvg) -> scatter by crow.
+    // local row. crow covers rows in 4-consecutive groups per (r>>2), but
+    // along the D axis the lane's element index is ln31 (stride 1 in d) —
+    // the 4-consecutive runs are in the KV direction (r&3), i.e. different
+    // rows, so pack via LDS-free transpose is not available; instead swap
+    // the roles: for each dblk the lane owns column d=dblk*32+ln31 of 16 kv
+    // rows. Pack pairs along r&3 into uint stores per row is not possible
+    // (different rows). Keep scalar stores but hoist the bounds check out
+    // of the loop (interior workgroups skip all 128 compares).
+    const bool all_in = (kvblk0 + 128 <= T);
 #pragma unroll
     for (int dblk = 0; dblk < NDBLK; ++dblk)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kvrow = kvblk0 + wid * 32 + crow(r, hi);
-        if (kvrow < T) {
+        if (all_in || kvrow < T) {
           const long off = kv_base + (long)kvrow * Hkv * D + dblk * 32 + ln31;
           dk[off] = f32_to_bf16(acc_dk[dblk][r]);
           dv[off] = f32_to_bf16(acc_dv[dblk][r]);
@@ -417,15 +560,15 @@ __global__ __launch_bounds__(256) void dq_kernel(
                    smem_raw);
 }
 
-template <int D>
+template <int D, int ABL = 0>
 __global__ __launch_bounds__(256) void dkdv_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
     const unsigned short* dout, const float* lse, const float* delta,
     unsigned short* dk, unsigned short* dv, int B, int T, int Hq, int Hkv,
     float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  DKDVKernel<D>::run(q, k, v, dout, lse, delta, dk, dv, B, T, Hq, Hkv, scale,
-                     smem_raw);
+  DKDVKernel<D, ABL>::run(q, k, v, dout, lse, delta, dk, dv, B, T, Hq, Hkv,
+                          scale, smem_raw);
 }
 
 }  // namespace attnbwd
@@ -507,4 +650,45 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
     TORCH_CHECK(false, "attn_bwd: head_dim must be 64 or 128");
   }
   return {dq, dk, dv};
+}
+
+// Ablation entry: times ONLY the dkdv kernel at the given mode (D=128).
+// Modes >0 produce wrong dk/dv; timing only.
+std::vector<torch::Tensor> attn_bwd_dkdv_ablate(torch::Tensor dout,
+                                                torch::Tensor q,
+                                                torch::Tensor k,
+                                                torch::Tensor v,
+                                                torch::Tensor lse,
+                                                torch::Tensor delta,
+                                                long mode) {
+  const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 128);
+  const float scale = 1.0f / sqrtf((float)D);
+  auto stream = at::cuda::getCurrentHIPStream();
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  const int n_kvblk = (T + 127) / 128;
+  size_t smem_kv = sizeof(typename attnbwd::DKDVKernel<128>::Smem);
+  auto launch = [&](auto kfn) {
+    hipLaunchKernelGGL(kfn, dim3(n_kvblk, Hkv, B), dim3(256), smem_kv, stream,
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (unsigned short*)dk.data_ptr(),
+                       (unsigned short*)dv.data_ptr(), B, T, Hq, Hkv, scale);
+  };
+  switch (mode) {
+    case 0: launch(attnbwd::dkdv_kernel<128, 0>); break;
+    case 1: launch(attnbwd::dkdv_kernel<128, 1>); break;
+    case 2: launch(attnbwd::dkdv_kernel<128, 2>); break;
+    case 3: launch(attnbwd::dkdv_kernel<128, 3>); break;
+    case 4: launch(attnbwd::dkdv_kernel<128, 4>); break;
+    case 5: launch(attnbwd::dkdv_kernel<128, 5>); break;
+    default: TORCH_CHECK(false, "mode 0-5");
+  }
+  HIP_CHECK_KERNEL();
+  return {dk, dv};
 }

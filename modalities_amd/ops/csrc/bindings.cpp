@@ -34,6 +34,12 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
 std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
                                            torch::Tensor v, long mode);
+std::vector<torch::Tensor> attn_bwd_dkdv_ablate(torch::Tensor dout,
+                                                torch::Tensor q,
+                                                torch::Tensor k,
+                                                torch::Tensor v,
+                                                torch::Tensor lse,
+                                                torch::Tensor delta, long mode);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (K5)");
@@ -54,4 +60,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA fragment-layout probe (verification)");
   m.def("attn_fwd_ablate", &attn_fwd_ablate,
         "attention fwd cost-attribution ablation (timing only)");
+  m.def("attn_bwd_dkdv_ablate", &attn_bwd_dkdv_ablate,
+        "attention bwd dkdv cost-attribution ablation (timing only)");
 }

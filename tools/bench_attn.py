@@ -23,8 +23,8 @@ def main():
     p.add_argument("--Hkv", type=int, default=20)
     p.add_argument("--D", type=int, default=128)
     p.add_argument("--iters", type=int, default=20)
-    p.add_argument("--what", choices=["fwd", "bwd", "both", "ablate"],
-                   default="both")
+    p.add_argument("--what", choices=["fwd", "bwd", "both", "ablate",
+                                      "ablate-bwd"], default="both")
     args = p.parse_args()
 
     from modalities_amd.ops.backend import hip_ext
@@ -64,6 +64,23 @@ def main():
             dt = (time.perf_counter() - t0) / args.iters
             print(f"ablate {name:12s}: {dt*1e3:8.3f} ms  "
                   f"{flops_fwd/dt/1e12:8.1f} TF/s")
+
+    if args.what == "ablate-bwd":
+        # dkdv-only flops: 3 of the 5 bwd matmuls (S recompute, dV, dK)
+        flops_dkdv = 0.5 * 6 * B * T * T * Hq * D
+        delta = (do.float() * o.float()).sum(-1).permute(0, 2, 1).contiguous()
+        for mode, name in [(0, "full"), (1, "no-qt/dot-stage"),
+                           (2, "no-stage"), (3, "no-lse/delta-loads"),
+                           (4, "no-dkdv-mfma"), (5, "stage-only")]:
+            ext.attn_bwd_dkdv_ablate(do, q, k, v, lse, delta, mode)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(args.iters):
+                ext.attn_bwd_dkdv_ablate(do, q, k, v, lse, delta, mode)
+            torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / args.iters
+            print(f"dkdv ablate {name:20s}: {dt*1e3:8.3f} ms  "
+                  f"{flops_dkdv/dt/1e12:8.1f} TF/s")
 
     if args.what in ("bwd", "both"):
         t0 = time.perf_counter()
