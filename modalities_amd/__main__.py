@@ -1,0 +1,251 @@
+"""CLI entrypoints (capability parity with reference
+src/modalities/__main__.py:44-749): ``run``, ``warmstart``, ``generate_text``,
+``convert_pytorch_to_hf_checkpoint``, the ``data ...`` preprocessing family,
+``benchmark ...`` sweep helpers and ``profile`` — plus per-rank structured
+JSON error logs on any exception."""
+
+import json
+import os
+import traceback
+from datetime import datetime
+from pathlib import Path
+
+import click
+
+from modalities_amd import api
+from modalities_amd.api import FileExistencePolicy
+
+
+def _exception_handling(run_fn, error_log_dir: Path = Path("logs")):
+    """Per-rank structured JSON error log (reference __main__.py:726-749)."""
+    try:
+        run_fn()
+    except Exception as e:
+        rank = os.environ.get("RANK", "0")
+        error_log_dir.mkdir(parents=True, exist_ok=True)
+        record = {
+            "rank": rank,
+            "hostname": os.uname().nodename,
+            "timestamp": datetime.now().isoformat(),
+            "exception_type": type(e).__name__,
+            "message": str(e),
+            "traceback": traceback.format_exc(),
+        }
+        with open(error_log_dir / f"error_rank_{rank}.json", "w") as f:
+            json.dump(record, f, indent=1)
+        raise
+
+
+@click.group(name="modalities-amd")
+def main():
+    pass
+
+
+@main.command(name="run")
+@click.option("--config_file_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+@click.option("--test_comm", is_flag=True, default=False)
+def CMD_entry_point_run_modalities(config_file_path: Path, test_comm: bool):
+    from modalities_amd.main import Main
+    from modalities_amd.running_env import DistEnv
+    from modalities_amd.utils.communication_test import run_communication_test
+
+    def run():
+        with DistEnv():
+            if test_comm:
+                run_communication_test()
+            main_obj = Main(config_file_path)
+            components = main_obj.build_components()
+            main_obj.run(components)
+
+    _exception_handling(run)
+
+
+@main.command(name="warmstart")
+@click.option("--config_file_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+@click.option("--last_checkpoint_info_file_path",
+              type=click.Path(exists=True, path_type=Path), required=True)
+def CMD_entry_point_warmstart_modalities(config_file_path: Path,
+                                         last_checkpoint_info_file_path: Path):
+    """Resume from the checkpoint pointed to by last_checkpoint_info.json
+    (reference __main__.py:137-163: registers the ${warmstart_env:...}
+    resolver with the checkpoint path)."""
+    from modalities_amd.main import Main
+    from modalities_amd.running_env import DistEnv
+
+    with open(last_checkpoint_info_file_path) as f:
+        checkpoint_paths = json.load(f)
+
+    def warmstart_env_resolver(key: str):
+        if key == "checkpoint_folder_path":
+            return checkpoint_paths["checkpoint_folder_path"]
+        return checkpoint_paths[key]
+
+    def run():
+        with DistEnv():
+            main_obj = Main(config_file_path,
+                            additional_resolver_funs={
+                                "warmstart_env": warmstart_env_resolver})
+            components = main_obj.build_components()
+            main_obj.run(components)
+
+    _exception_handling(run)
+
+
+@main.command(name="generate_text")
+@click.option("--config_file_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+def CMD_entry_point_generate_text(config_file_path: Path):
+    api.generate_text(config_file_path)
+
+
+@main.command(name="convert_pytorch_to_hf_checkpoint")
+@click.option("--config_file_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+@click.option("--output_hf_checkpoint_dir", type=click.Path(path_type=Path),
+              required=True)
+@click.option("--prediction_key", type=str, default="logits")
+def CMD_entry_point_convert_pytorch_to_hf_checkpoint(
+        config_file_path: Path, output_hf_checkpoint_dir: Path, prediction_key: str):
+    api.convert_pytorch_to_hf_checkpoint(config_file_path,
+                                         output_hf_checkpoint_dir, prediction_key)
+
+
+# ---- data subcommands -------------------------------------------------------
+
+@main.group(name="data")
+def data():
+    pass
+
+
+_POLICY = click.Choice([p.value for p in FileExistencePolicy])
+
+
+@data.command(name="create_raw_index")
+@click.argument("src_path", type=click.Path(exists=True, path_type=Path))
+@click.option("--index_path", type=click.Path(path_type=Path), default=None)
+@click.option("--file_existence_policy", type=_POLICY, default="error")
+def CMD_create_raw_index(src_path, index_path, file_existence_policy):
+    n = api.create_raw_data_index(src_path, index_path,
+                                  FileExistencePolicy(file_existence_policy))
+    click.echo(f"indexed {n} lines")
+
+
+@data.command(name="pack_encoded_data")
+@click.argument("config_path", type=click.Path(exists=True, path_type=Path))
+@click.option("--file_existence_policy", type=_POLICY, default="error")
+def CMD_pack_encoded_data(config_path, file_existence_policy):
+    from modalities_amd.config.yaml_loader import load_app_config_dict
+    config_dict = load_app_config_dict(config_path)
+    n = api.pack_encoded_data(config_dict, FileExistencePolicy(file_existence_policy))
+    click.echo(f"packed {n} documents")
+
+
+@data.command(name="merge_packed_data")
+@click.argument("src_paths", type=click.Path(exists=True, path_type=Path), nargs=-1)
+@click.argument("target_path", type=click.Path(path_type=Path))
+def CMD_merge_packed_data(src_paths, target_path):
+    api.merge_packed_data_files(list(src_paths), target_path)
+
+
+@data.command(name="shuffle_tokenized_data")
+@click.option("--input_data_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+@click.option("--output_data_path", type=click.Path(path_type=Path), required=True)
+@click.option("--batch_size", type=int, default=1024)
+@click.option("--seed", type=int, default=None)
+@click.option("--file_existence_policy", type=_POLICY, default="error")
+def CMD_shuffle_tokenized_data(input_data_path, output_data_path, batch_size, seed,
+                               file_existence_policy):
+    api.shuffle_tokenized_data(input_data_path, output_data_path, batch_size, seed,
+                               FileExistencePolicy(file_existence_policy))
+
+
+@data.command(name="shuffle_jsonl_data")
+@click.option("--input_data_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+@click.option("--output_data_path", type=click.Path(path_type=Path), required=True)
+@click.option("--seed", type=int, default=None)
+@click.option("--file_existence_policy", type=_POLICY, default="error")
+def CMD_shuffle_jsonl_data(input_data_path, output_data_path, seed,
+                           file_existence_policy):
+    api.shuffle_jsonl_data(input_data_path, output_data_path, seed,
+                           FileExistencePolicy(file_existence_policy))
+
+
+@data.command(name="create_shuffled_dataset_chunk")
+@click.option("--input_file_list", type=click.Path(exists=True, path_type=Path),
+              required=True, help="text file: one pbin path per line")
+@click.option("--output_chunk_file_path", type=click.Path(path_type=Path),
+              required=True)
+@click.option("--chunk_id", type=int, required=True)
+@click.option("--num_chunks", type=int, required=True)
+@click.option("--global_seed", type=int, default=None)
+@click.option("--file_existence_policy", type=_POLICY, default="error")
+def CMD_create_shuffled_dataset_chunk(input_file_list, output_chunk_file_path,
+                                      chunk_id, num_chunks, global_seed,
+                                      file_existence_policy):
+    with open(input_file_list) as f:
+        files = [Path(line.strip()) for line in f if line.strip()]
+    api.create_shuffled_dataset_chunk(files, output_chunk_file_path, chunk_id,
+                                      num_chunks, global_seed,
+                                      FileExistencePolicy(file_existence_policy))
+
+
+@data.command(name="create_shuffled_jsonl_chunk")
+@click.option("--input_file_list", type=click.Path(exists=True, path_type=Path),
+              required=True)
+@click.option("--output_chunk_file_path", type=click.Path(path_type=Path),
+              required=True)
+@click.option("--chunk_id", type=int, required=True)
+@click.option("--num_chunks", type=int, required=True)
+@click.option("--global_seed", type=int, default=None)
+@click.option("--file_existence_policy", type=_POLICY, default="error")
+def CMD_create_shuffled_jsonl_chunk(input_file_list, output_chunk_file_path,
+                                    chunk_id, num_chunks, global_seed,
+                                    file_existence_policy):
+    with open(input_file_list) as f:
+        files = [Path(line.strip()) for line in f if line.strip()]
+    api.create_shuffled_jsonl_dataset_chunk(files, output_chunk_file_path, chunk_id,
+                                            num_chunks, global_seed,
+                                            FileExistencePolicy(file_existence_policy))
+
+
+@data.command(name="prepare_instruction_tuning_data")
+@click.option("--config_file_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+def CMD_prepare_instruction_tuning_data(config_file_path):
+    from modalities_amd.dataloader.instruction_tuning import \
+        create_instruction_tuning_data
+    create_instruction_tuning_data(config_file_path)
+
+
+# ---- benchmark subcommands --------------------------------------------------
+
+@main.group(name="benchmark")
+def benchmark():
+    pass
+
+
+@benchmark.command(name="prepare_sweep_configs")
+@click.option("--sweep_config_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+@click.option("--output_dir", type=click.Path(path_type=Path), required=True)
+def CMD_prepare_sweep_configs(sweep_config_path, output_dir):
+    from modalities_amd.utils.benchmarking import prepare_sweep_configs
+    n = prepare_sweep_configs(sweep_config_path, output_dir)
+    click.echo(f"wrote {n} sweep configs")
+
+
+@benchmark.command(name="list_remaining_runs")
+@click.option("--sweep_dir", type=click.Path(exists=True, path_type=Path),
+              required=True)
+def CMD_list_remaining_runs(sweep_dir):
+    from modalities_amd.utils.benchmarking import list_remaining_runs
+    for p in list_remaining_runs(sweep_dir):
+        click.echo(str(p))
+
+
+if __name__ == "__main__":
+    main()

