@@ -1,0 +1,312 @@
+"""Tensor parallelism (TP) + sequence parallelism (SP) for MI355X.
+
+Capability parity with the reference's DTensor parallelize_module plan
+(reference: src/modalities/models/model_factory.py:657-766 — q/k/v colwise,
+attn out-proj + MLP down-proj rowwise, SP norms) but implemented as explicit
+sharded nn.Linears with hand-placed RCCL collectives over the TP process
+group — the MI355X-native design: TP groups stay inside one
+fully-connected xGMI node, the rowwise all-reduce is one activation-sized
+collective per block per direction.
+
+Autograd-correct conjugate pairs (Megatron-style):
+  _CopyToTP      fwd identity      bwd all-reduce   (entry of col-parallel)
+  _ReduceFromTP  fwd all-reduce    bwd identity     (exit of row-parallel)
+  _GatherSeq     fwd all-gather(seq)  bwd reduce-scatter(seq)   (SP entry)
+  _ScatterSeq    fwd reduce-scatter(seq)  bwd all-gather(seq)   (SP exit)
+"""
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+
+def _backend_is_gloo(group) -> bool:
+    try:
+        return dist.get_backend(group) == "gloo"
+    except Exception:
+        return dist.get_backend() == "gloo"
+
+
+def _all_gather_cat(x: torch.Tensor, group, dim: int) -> torch.Tensor:
+    world = dist.get_world_size(group)
+    outs = [torch.empty_like(x) for _ in range(world)]
+    dist.all_gather(outs, x.contiguous(), group=group)
+    return torch.cat(outs, dim=dim)
+
+
+def _reduce_scatter_seq(x: torch.Tensor, group, dim: int) -> torch.Tensor:
+    """Sum over ranks, return this rank's 1/world slice along dim."""
+    world = dist.get_world_size(group)
+    rank = dist.get_rank(group)
+    if _backend_is_gloo(group):
+        x = x.contiguous()
+        dist.all_reduce(x, group=group)
+        return x.chunk(world, dim=dim)[rank].contiguous()
+    chunks = [c.contiguous() for c in x.chunk(world, dim=dim)]
+    out = torch.empty_like(chunks[0])
+    dist.reduce_scatter(out, chunks, group=group)
+    return out
+
+
+class _CopyToTP(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return x
+
+    @staticmethod
+    def backward(ctx, grad):
+        grad = grad.contiguous()
+        dist.all_reduce(grad, group=ctx.group)
+        return grad, None
+
+
+class _ReduceFromTP(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, group):
+        x = x.contiguous()
+        dist.all_reduce(x, group=group)
+        return x
+
+    @staticmethod
+    def backward(ctx, grad):
+        return grad, None
+
+
+class _GatherSeq(torch.autograd.Function):
+    """SP -> TP boundary: all-gather the seq-sharded activation."""
+
+    @staticmethod
+    def forward(ctx, x, group, dim):
+        ctx.group, ctx.dim = group, dim
+        return _all_gather_cat(x, group, dim)
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _reduce_scatter_seq(grad, ctx.group, ctx.dim), None, None
+
+
+class _ScatterSeq(torch.autograd.Function):
+    """TP -> SP boundary: reduce-scatter partial sums along seq."""
+
+    @staticmethod
+    def forward(ctx, x, group, dim):
+        ctx.group, ctx.dim = group, dim
+        return _reduce_scatter_seq(x, group, dim)
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _all_gather_cat(grad, ctx.group, ctx.dim), None, None
+
+
+class _GatherSeqToReplicated(torch.autograd.Function):
+    """Seq-sharded -> replicated where the DOWNSTREAM computation is
+    replicated across the TP group (the lm-head path): every rank computes
+    the identical full gradient, so backward takes this rank's slice —
+    a reduce-scatter here would double-count by the group size."""
+
+    @staticmethod
+    def forward(ctx, x, group, dim):
+        ctx.group, ctx.dim = group, dim
+        return _all_gather_cat(x, group, dim)
+
+    @staticmethod
+    def backward(ctx, grad):
+        world = dist.get_world_size(ctx.group)
+        rank = dist.get_rank(ctx.group)
+        return grad.chunk(world, dim=ctx.dim)[rank].contiguous(), None, None
+
+
+class _SliceSeq(torch.autograd.Function):
+    """Replicated -> seq-sharded: fwd takes this rank's slice, bwd
+    all-gathers the grads (used once after the embedding under SP)."""
+
+    @staticmethod
+    def forward(ctx, x, group, dim):
+        ctx.group, ctx.dim = group, dim
+        world = dist.get_world_size(group)
+        rank = dist.get_rank(group)
+        return x.chunk(world, dim=dim)[rank].contiguous()
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _all_gather_cat(grad, ctx.group, ctx.dim), None, None
+
+
+def copy_to_tp(x, group):
+    return _CopyToTP.apply(x, group) if group is not None else x
+
+
+def reduce_from_tp(x, group):
+    return _ReduceFromTP.apply(x, group) if group is not None else x
+
+
+def gather_seq(x, group, dim=1):
+    return _GatherSeq.apply(x, group, dim) if group is not None else x
+
+
+def scatter_seq(x, group, dim=1):
+    return _ScatterSeq.apply(x, group, dim) if group is not None else x
+
+
+# ---------------------------------------------------------------------------
+
+@torch.no_grad()
+def _shard_linear_(lin: nn.Linear, tp_rank: int, tp_size: int, dim: int) -> None:
+    """Shard a Linear's weight (and bias for dim=0) in place.
+    dim=0: column-parallel (split out_features); dim=1: row-parallel."""
+    w = lin.weight
+    if dim == 0:
+        out_per = w.shape[0] // tp_size
+        lin.weight = nn.Parameter(
+            w[tp_rank * out_per:(tp_rank + 1) * out_per].clone())
+        if lin.bias is not None:
+            lin.bias = nn.Parameter(
+                lin.bias[tp_rank * out_per:(tp_rank + 1) * out_per].clone())
+        lin.out_features = out_per
+    else:
+        in_per = w.shape[1] // tp_size
+        lin.weight = nn.Parameter(
+            w[:, tp_rank * in_per:(tp_rank + 1) * in_per].clone())
+        # bias for row-parallel stays full but must be added once: scale it
+        # by 1/tp so the all-reduce sums to one bias application.
+        if lin.bias is not None:
+            lin.bias = nn.Parameter(lin.bias.clone() / tp_size)
+        lin.in_features = in_per
+
+
+class _TPAttentionForward:
+    """Replacement forward for CausalSelfAttention under TP(+SP)."""
+
+    def __init__(self, attn, group, sp: bool):
+        self.attn = attn
+        self.group = group
+        self.sp = sp
+
+    def __call__(self, x: torch.Tensor, rope_cos, rope_sin) -> torch.Tensor:
+        attn = self.attn
+        if self.sp:
+            x = gather_seq(x, self.group, dim=1)
+        else:
+            x = copy_to_tp(x, self.group)
+        B, T, C = x.shape
+        q = attn.q_attn(x).view(B, T, attn.n_head_q, attn.head_dim)
+        k = attn.k_attn(x).view(B, T, attn.n_head_kv, attn.head_dim)
+        v = attn.v_attn(x).view(B, T, attn.n_head_kv, attn.head_dim)
+        if attn.q_norm is not None:
+            q = attn.q_norm(q)
+            k = attn.k_norm(k)
+        if rope_cos is not None:
+            from modalities_amd.ops import rope_apply
+            q = rope_apply(q, rope_cos, rope_sin)
+            k = rope_apply(k, rope_cos, rope_sin)
+        y = attn._attend(q, k, v)
+        y = y.reshape(B, T, attn.n_head_q * attn.head_dim)
+        y = attn.c_proj(y)
+        if self.sp:
+            y = scatter_seq(y, self.group, dim=1)
+        else:
+            y = reduce_from_tp(y, self.group)
+        return attn.resid_dropout(y)
+
+
+class _TPSwiGLUForward:
+    def __init__(self, mlp, group, sp: bool):
+        self.mlp = mlp
+        self.group = group
+        self.sp = sp
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        from modalities_amd.ops import silu_mul
+        if self.sp:
+            x = gather_seq(x, self.group, dim=1)
+        else:
+            x = copy_to_tp(x, self.group)
+        y = self.mlp.W_2(silu_mul(self.mlp.W(x), self.mlp.V(x)))
+        if self.sp:
+            return scatter_seq(y, self.group, dim=1)
+        return reduce_from_tp(y, self.group)
+
+
+class _TPGeluMLPForward:
+    def __init__(self, mlp, group, sp: bool):
+        self.mlp = mlp
+        self.group = group
+        self.sp = sp
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        if self.sp:
+            x = gather_seq(x, self.group, dim=1)
+        else:
+            x = copy_to_tp(x, self.group)
+        y = self.mlp.c_proj(self.mlp.gelu(self.mlp.c_fc(x)))
+        if self.sp:
+            y = scatter_seq(y, self.group, dim=1)
+        else:
+            y = reduce_from_tp(y, self.group)
+        return self.mlp.dropout(y)
+
+
+def get_gpt2_tensor_parallelized_model(model, device_mesh=None, group=None,
+                                       tp_rank: Optional[int] = None,
+                                       tp_size: Optional[int] = None,
+                                       sequence_parallel: bool = False):
+    """Shard a GPT2LLM in place for TP (reference analog:
+    model_factory.py:657-766).
+
+    - q/k/v projections column-parallel (whole heads per rank)
+    - attention out-proj and MLP down-proj row-parallel
+    - embeddings / lm_head / norms stay replicated in their weights (they
+      are DP-sharded by the XGMI engine afterwards)
+    - with sequence_parallel=True the residual stream between blocks is
+      seq-sharded (reference SequenceParallel plan: norms run on 1/tp of
+      the tokens): one slice after the embedding dropout, all-gather at
+      attention/MLP entry, reduce-scatter at exit, and a final all-gather
+      before the lm-head norm.
+    """
+    if device_mesh is not None:
+        from modalities_amd.parallel.mesh import ParallelismDegrees
+        dim = device_mesh.dims[ParallelismDegrees.TP]
+        group, tp_rank, tp_size = dim.group, dim.rank, dim.size
+    if tp_size in (None, 1):
+        return model
+    cfg = model.config
+    if cfg.n_head_q % tp_size or cfg.n_head_kv % tp_size:
+        raise ValueError(f"n_head_q ({cfg.n_head_q}) and n_head_kv "
+                         f"({cfg.n_head_kv}) must be divisible by tp={tp_size}")
+
+    for block in model.blocks:
+        attn = block.attn
+        _shard_linear_(attn.q_attn, tp_rank, tp_size, dim=0)
+        _shard_linear_(attn.k_attn, tp_rank, tp_size, dim=0)
+        _shard_linear_(attn.v_attn, tp_rank, tp_size, dim=0)
+        _shard_linear_(attn.c_proj, tp_rank, tp_size, dim=1)
+        attn.n_head_q //= tp_size
+        attn.n_head_kv //= tp_size
+        block.attn.forward = _TPAttentionForward(attn, group,
+                                                 sequence_parallel)
+        mlp = block.mlp
+        if hasattr(mlp, "W_2"):  # SwiGLU
+            if mlp.hidden_dim % tp_size:
+                raise ValueError(f"SwiGLU hidden_dim {mlp.hidden_dim} not "
+                                 f"divisible by tp={tp_size}")
+            _shard_linear_(mlp.W, tp_rank, tp_size, dim=0)
+            _shard_linear_(mlp.V, tp_rank, tp_size, dim=0)
+            _shard_linear_(mlp.W_2, tp_rank, tp_size, dim=1)
+            block.mlp.forward = _TPSwiGLUForward(mlp, group, sequence_parallel)
+        else:  # GELU MLP
+            _shard_linear_(mlp.c_fc, tp_rank, tp_size, dim=0)
+            _shard_linear_(mlp.c_proj, tp_rank, tp_size, dim=1)
+            block.mlp.forward = _TPGeluMLPForward(mlp, group, sequence_parallel)
+
+    if sequence_parallel:
+        # slice the residual stream after the embedding dropout; re-gather
+        # before the lm-head norm (reference: SequenceParallel on norms +
+        # PrepareModuleInput, model_factory.py:672-727)
+        model.drop.register_forward_hook(
+            lambda mod, args, out: _SliceSeq.apply(out, group, 1))
+        model.lm_head_norm.register_forward_pre_hook(
+            lambda mod, args: (_GatherSeqToReplicated.apply(args[0], group, 1),))
+    return model
