@@ -21,8 +21,10 @@ import torch
 from modalities_amd.ops.backend import use_hip, hip_ext
 
 
-def _attention_ref(q, k, v, causal=True):
+def _attention_ref(q, k, v, causal=True, q_offset=None):
     # q: [B, T, Hq, D], k/v: [B, S, Hkv, D] -> [B, T, Hq, D]; fp32 compute.
+    # q_offset: global position of q row 0 relative to k row 0 (context
+    # parallelism); default aligns the q block to the END of the keys.
     B, T, Hq, D = q.shape
     S, Hkv = k.shape[1], k.shape[2]
     rep = Hq // Hkv
@@ -31,7 +33,8 @@ def _attention_ref(q, k, v, causal=True):
     vf = v.permute(0, 2, 1, 3).float().repeat_interleave(rep, dim=1)
     att = qf @ kf.transpose(-2, -1) / math.sqrt(D)          # [B,Hq,T,S]
     if causal:
-        mask = torch.ones(T, S, dtype=torch.bool, device=q.device).tril(S - T)
+        diag = (S - T) if q_offset is None else q_offset
+        mask = torch.ones(T, S, dtype=torch.bool, device=q.device).tril(diag)
         att = att.masked_fill(~mask, float("-inf"))
     att = att.softmax(-1)
     out = att @ vf                                          # [B,Hq,T,D]
@@ -53,9 +56,33 @@ class _FlashAttnHip(torch.autograd.Function):
         return dq, dk, dv, None
 
 
+from typing import Optional
+
+
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                    causal: bool = True) -> torch.Tensor:
-    """Causal flash attention. q: [B,T,Hq,D]; k,v: [B,S,Hkv,D]."""
-    if use_hip(q, k, v):
+                    causal: bool = True,
+                    q_offset: Optional[int] = None) -> torch.Tensor:
+    """Causal flash attention. q: [B,T,Hq,D]; k,v: [B,S,Hkv,D].
+
+    q_offset (context parallelism): global key position of q row 0; None =
+    classic alignment (q block ends at the last key, i.e. offset S-T).
+    Offset attention currently runs the composed-ops path on device;
+    teaching the HIP kernels the offset is a planned optimization."""
+    if use_hip(q, k, v) and q.shape[1] == k.shape[1] and q_offset in (None, 0):
         return _FlashAttnHip.apply(q.contiguous(), k.contiguous(), v.contiguous(), causal)
-    return _attention_ref(q, k, v, causal)
+    if q.is_cuda:
+        # SDPA with an explicit offset-causal mask (keeps CP functional on
+        # GPU; slower than K1 — kernel offset support is the follow-up)
+        B, T, Hq, D = q.shape
+        S, Hkv = k.shape[1], k.shape[2]
+        rep = Hq // Hkv
+        qt = q.transpose(1, 2)
+        kt = k.transpose(1, 2).repeat_interleave(rep, dim=1)
+        vt = v.transpose(1, 2).repeat_interleave(rep, dim=1)
+        diag = (S - T) if q_offset is None else q_offset
+        mask = torch.ones(T, S, dtype=torch.bool, device=q.device).tril(
+            diag if causal else S)
+        y = torch.nn.functional.scaled_dot_product_attention(
+            qt, kt, vt, attn_mask=mask)
+        return y.transpose(1, 2)
+    return _attention_ref(q, k, v, causal, q_offset=q_offset)

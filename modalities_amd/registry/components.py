@@ -40,6 +40,7 @@ from modalities_amd.optimizers.lr_schedulers import (DummyLRScheduler,
                                                      get_onecycle_lr, get_step_lr)
 from modalities_amd.optimizers.optimizer_factory import get_adam_w
 from modalities_amd.parallel.mesh import get_device_mesh
+from modalities_amd.parallel.cp import get_gpt2_context_parallel_model
 from modalities_amd.parallel.tp import get_gpt2_tensor_parallelized_model
 from modalities_amd.registry.registry import ComponentEntity, Registry
 from modalities_amd.tokenization.tokenizer_wrapper import (CharTokenizer,
@@ -191,6 +192,8 @@ COMPONENTS: list[ComponentEntity] = [
     ComponentEntity("device_mesh", "default", get_device_mesh, None),
     ComponentEntity("tensor_parallelized_model", "gpt2_tp",
                     get_gpt2_tensor_parallelized_model, None),
+    ComponentEntity("context_parallelized_model", "gpt2_cp",
+                    get_gpt2_context_parallel_model, None),
     # training aux
     ComponentEntity("gradient_clipper", "fsdp2", GradientClipper, None),
     ComponentEntity("gradient_clipper", "default", GradientClipper, None),

@@ -1,0 +1,101 @@
+"""CP correctness: offset-causal attention, cp=2 logits/grad equivalence vs
+the full-sequence single-process run. (The reference has NO CP compute —
+SURVEY.md §5 — so the correctness model is our own full-seq run.)"""
+
+import pytest
+import torch
+
+from modalities_amd.models.gpt2 import GPT2LLM, GPT2LLMConfig
+from modalities_amd.ops.attention import _attention_ref, flash_attention
+from tests.conftest import find_free_port
+from tests.utils_dist import run_distributed
+
+VOCAB = 128
+T = 32
+
+
+def tiny_cfg():
+    return GPT2LLMConfig(vocab_size=VOCAB, n_layer=2, n_head_q=4, n_head_kv=2,
+                         n_embd=64, ffn_hidden=256, sequence_length=T,
+                         seed=3, dropout=0.0)
+
+
+def make_batch(seed=5, batch=2):
+    g = torch.Generator().manual_seed(seed)
+    ids = torch.randint(0, VOCAB, (batch, T + 1), generator=g)
+    return ids[:, :-1], ids[:, 1:]
+
+
+def test_offset_causal_attention_chunks_compose():
+    """Full causal attention == concatenation of per-chunk offset-causal
+    attentions against the full K/V."""
+    torch.manual_seed(0)
+    B, Hq, Hkv, D = 2, 4, 2, 16
+    q = torch.randn(B, T, Hq, D)
+    k = torch.randn(B, T, Hkv, D)
+    v = torch.randn(B, T, Hkv, D)
+    full = _attention_ref(q, k, v, causal=True)
+    for n_chunks in (2, 4):
+        tl = T // n_chunks
+        parts = [flash_attention(q[:, r * tl:(r + 1) * tl], k, v, causal=True,
+                                 q_offset=r * tl)
+                 for r in range(n_chunks)]
+        torch.testing.assert_close(torch.cat(parts, dim=1), full,
+                                   rtol=1e-4, atol=1e-5)
+
+
+def reference_run():
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    x, y = make_batch()
+    out = model({"input_ids": x})["logits"]
+    loss = torch.nn.functional.cross_entropy(out.reshape(-1, VOCAB).float(),
+                                             y.reshape(-1))
+    loss.backward()
+    grads = {n: p.grad.clone().numpy() for n, p in model.named_parameters()}
+    return out.detach(), loss.detach(), grads
+
+
+def _cp_worker(rank, world):
+    import torch.distributed as dist
+
+    from modalities_amd.parallel.cp import (get_gpt2_context_parallel_model,
+                                            slice_targets_for_cp)
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    model = get_gpt2_context_parallel_model(model, group=dist.group.WORLD,
+                                            cp_rank=rank, cp_size=world)
+    x, y = make_batch()
+    out = model({"input_ids": x})["logits"]  # [B, T/world, V]
+    y_local = slice_targets_for_cp(y, rank, world)
+    # global mean loss: local sum / global count
+    local_sum = torch.nn.functional.cross_entropy(
+        out.reshape(-1, VOCAB).float(), y_local.reshape(-1), reduction="sum")
+    n_total = torch.tensor(float(y.numel()))
+    loss = local_sum / n_total
+    # all CP ranks contribute: total loss = sum of per-rank losses
+    loss_full = loss.clone().detach()
+    dist.all_reduce(loss_full)
+    loss.backward()
+    g_q = model.blocks[0].attn.q_attn.weight.grad.clone()
+    g_wte = model.wte.weight.grad.clone()
+    return (out.detach().numpy(), loss_full.item(), g_q.numpy(), g_wte.numpy())
+
+
+def test_cp2_matches_full_sequence():
+    ref_out, ref_loss, ref_grads = reference_run()
+    results = run_distributed(_cp_worker, world_size=2, port=find_free_port())
+    tl = T // 2
+    outs = [torch.from_numpy(results[r][0]) for r in range(2)]
+    torch.testing.assert_close(torch.cat(outs, dim=1), ref_out,
+                               rtol=1e-4, atol=1e-4)
+    assert results[0][1] == pytest.approx(ref_loss.item(), rel=1e-5)
+    # weight grads: each rank's grad covers its chunk's contribution; the DP
+    # engine (or an explicit all-reduce) sums them — verify the SUM matches.
+    g_sum = torch.from_numpy(results[0][2]) + torch.from_numpy(results[1][2])
+    torch.testing.assert_close(g_sum,
+                               torch.from_numpy(ref_grads["blocks.0.attn.q_attn.weight"]),
+                               rtol=1e-4, atol=1e-5)
+    wte_sum = torch.from_numpy(results[0][3]) + torch.from_numpy(results[1][3])
+    torch.testing.assert_close(wte_sum, torch.from_numpy(ref_grads["wte.weight"]),
+                               rtol=1e-4, atol=1e-5)
