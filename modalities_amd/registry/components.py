@@ -48,7 +48,9 @@ from modalities_amd.tokenization.tokenizer_wrapper import (CharTokenizer,
                                                            PreTrainedSPTokenizer)
 from modalities_amd.training.gradient_clipping import (DummyGradientClipper,
                                                        GradientClipper)
+from modalities_amd.utils.debug_components import get_debugging_enriched_model
 from modalities_amd.utils.mfu import GPT2MFUCalculator, detect_device_peak_flops
+from modalities_amd.utils.profilers import get_profiler
 from modalities_amd.utils.number_conversion import NumberConversion
 
 
@@ -199,7 +201,10 @@ COMPONENTS: list[ComponentEntity] = [
     ComponentEntity("gradient_clipper", "default", GradientClipper, None),
     ComponentEntity("gradient_clipper", "dummy", DummyGradientClipper, None),
     ComponentEntity("mfu_calculator", "gpt2", get_gpt2_mfu_calculator, None),
-    # observability
+    # observability / debugging / profiling
+    ComponentEntity("profiler", "default", get_profiler, None),
+    ComponentEntity("debugging_enriched_model", "default",
+                    get_debugging_enriched_model, None),
     ComponentEntity("progress_subscriber", "dummy", DummyProgressSubscriber, None),
     ComponentEntity("progress_subscriber", "rich", DummyProgressSubscriber, None),
     ComponentEntity("results_subscriber", "dummy", DummyResultSubscriber, None),
@@ -214,6 +219,8 @@ COMPONENTS: list[ComponentEntity] = [
                     NumberConversion.get_num_samples_from_num_tokens, None),
     ComponentEntity("number_conversion", "num_steps_from_num_samples",
                     NumberConversion.get_num_steps_from_num_samples, None),
+    ComponentEntity("number_conversion", "num_samples_from_num_steps",
+                    NumberConversion.get_num_samples_from_num_steps, None),
     ComponentEntity("number_conversion", "num_steps_from_num_tokens",
                     NumberConversion.get_num_steps_from_num_tokens, None),
     ComponentEntity("number_conversion", "num_tokens_from_num_steps",

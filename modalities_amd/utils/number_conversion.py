@@ -62,6 +62,13 @@ class NumberConversion:
         return (num_steps * dp_degree * local_micro_batch_size * sequence_length
                 * gradient_accumulation_steps)
 
+    @staticmethod
+    def get_num_samples_from_num_steps(num_steps: int, dp_degree: int,
+                                       local_micro_batch_size: int,
+                                       gradient_accumulation_steps: int) -> int:
+        return (num_steps * dp_degree * local_micro_batch_size
+                * gradient_accumulation_steps)
+
     # ---- checkpoint-path parsing ----------------------------------------
 
     @staticmethod
