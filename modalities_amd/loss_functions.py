@@ -40,3 +40,31 @@ class CLMCrossEntropyLoss(Loss):
             logits, labels = forward_batch
         return fused_cross_entropy(logits.view(-1, logits.shape[-1]),
                                    labels.reshape(-1), ignore_index=-100)
+
+
+class NCELoss(Loss):
+    """Symmetric InfoNCE contrastive loss over paired embeddings
+    (capability parity with reference src/modalities/loss_functions.py:125-167;
+    used by CoCa)."""
+
+    def __init__(self, prediction_key1: str, prediction_key2: str,
+                 is_asymmetric: bool = False, temperature: float = 1.0,
+                 tag: str = "NCELoss"):
+        super().__init__(tag)
+        self.prediction_key1 = prediction_key1
+        self.prediction_key2 = prediction_key2
+        self.is_asymmetric = is_asymmetric
+        self.temperature = temperature
+
+    def forward(self, forward_batch: InferenceResultBatch) -> torch.Tensor:
+        emb1 = forward_batch.get_predictions(self.prediction_key1).float()
+        emb2 = forward_batch.get_predictions(self.prediction_key2).float()
+        emb1 = torch.nn.functional.normalize(emb1, dim=-1)
+        emb2 = torch.nn.functional.normalize(emb2, dim=-1)
+        logits = emb1 @ emb2.t() / self.temperature
+        labels = torch.arange(emb1.shape[0], device=emb1.device)
+        loss = torch.nn.functional.cross_entropy(logits, labels)
+        if not self.is_asymmetric:
+            loss = 0.5 * (loss + torch.nn.functional.cross_entropy(
+                logits.t(), labels))
+        return loss

@@ -23,12 +23,14 @@ from modalities_amd.dataloader.dataset import (CombinedDataset, DummyDataset,
                                                PackedMemMapDatasetMegatron,
                                                SyntheticLMDataset)
 from modalities_amd.dataloader.samplers import ResumableDistributedSampler
-from modalities_amd.loss_functions import CLMCrossEntropyLoss
+from modalities_amd.loss_functions import CLMCrossEntropyLoss, NCELoss
 from modalities_amd.logging_broker.subscribers import (DummyProgressSubscriber,
                                                        DummyResultSubscriber,
                                                        ResultsToDiscSubscriber,
                                                        RichResultSubscriber)
+from modalities_amd.models.coca import CoCa
 from modalities_amd.models.gpt2 import GPT2LLM, GPT2LLMConfig
+from modalities_amd.models.vision_transformer import VisionTransformer
 from modalities_amd.models.model_factory import ModelFactory
 from modalities_amd.nn.model_initialization import (Llama3LikeInitialization,
                                                     get_composed_model_initializer)
@@ -134,6 +136,8 @@ COMPONENTS: list[ComponentEntity] = [
     # models
     ComponentEntity("model", "gpt2", get_gpt2_model, GPT2LLMConfig),
     ComponentEntity("model", "checkpointed", get_checkpointed_model, None),
+    ComponentEntity("model", "coca", CoCa, None),
+    ComponentEntity("model", "vision_transformer", VisionTransformer, None),
     ComponentEntity("sharded_model", "xgmi_fsdp", ModelFactory.get_sharded_model, None),
     ComponentEntity("wrapped_model", "fsdp2", ModelFactory.get_sharded_model, None),
     ComponentEntity("initialized_model", "default",
@@ -158,6 +162,7 @@ COMPONENTS: list[ComponentEntity] = [
     ComponentEntity("scheduler", "onecycle_lr", get_onecycle_lr, None),
     # loss
     ComponentEntity("loss", "clm_cross_entropy_loss", CLMCrossEntropyLoss, None),
+    ComponentEntity("loss", "nce_loss", NCELoss, None),
     # data
     ComponentEntity("dataset", "packed_mem_map_dataset_continuous",
                     PackedMemMapDatasetContinuous, None),
