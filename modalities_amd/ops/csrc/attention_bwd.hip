@@ -15,8 +15,27 @@
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(2))) unsigned int uintx2;
 
 namespace attnbwd {
+
+// ds_read_b64_tr_b16 (semantics verified by tr16_probe on gfx950): the 16
+// lanes of each contiguous lane group supply one 8-byte (4x bf16) chunk
+// each; with the [panel][row][16-col] image and per-lane address
+//   base + (panel*512 + (k0 + (lambda>>2))*16 + 4*(lambda&3)) * 2B,
+// lambda = lane&15, the read returns to lane l element j = X[k0+j][d0+(l&15)]
+// -- exactly the 32x32x16 MFMA A/B fragment k-run (4 of the 8 elements;
+// two reads at k0 and k0+4 complete it). Conflict-free: the 16 lanes hit 16
+// distinct even dwords of a 256-dword-aligned panel.
+__device__ __forceinline__ uintx2 tr_read_b64(unsigned addr_bytes) {
+  uintx2 r;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(r) : "v"(addr_bytes));
+  return r;
+}
+
+__device__ __forceinline__ unsigned lds_addr(const void* p) {
+  return (unsigned)(unsigned long long)p;  // LDS aperture: low 32 = offset
+}
 
 constexpr int KVBLK = 64;  // dQ kernel kv tile
 
@@ -101,7 +120,7 @@ struct DQKernel {
   struct Smem {
     unsigned short k[KVBLK * D];
     unsigned short v[KVBLK * D];
-    unsigned short kt[D * KVBLK];
+    unsigned short kt[D * KVBLK];  // panel image [D/16][KVBLK][16] for tr reads
   };
 
   static __device__ void run(const unsigned short* q, const unsigned short* k,
@@ -147,33 +166,50 @@ struct DQKernel {
     const int q_hi_wg = min(qblk0 + 127, T - 1);
     const int n_tiles = q_hi_wg / KVBLK + 1;
 
-    for (int tile = 0; tile < n_tiles; ++tile) {
-      const int kv0 = tile * KVBLK;
-      {  // stage K, V (row-major) and Kt (transposed)
-        const int tid = threadIdx.x, nthreads = 256;
-        const int total8 = KVBLK * D / 8;
-        for (int i = tid; i < total8; i += nthreads) {
-          const int r = (i * 8) / D, c = (i * 8) % D;
-          shortx8 kk, vv;
-          if (kv0 + r < T) {
-            kk = *reinterpret_cast<const shortx8*>(
-                k + kv_base + (long)(kv0 + r) * Hkv * D + c);
-            vv = *reinterpret_cast<const shortx8*>(
-                v + kv_base + (long)(kv0 + r) * Hkv * D + c);
-          } else {
+    // T14 issue-early/write-late with ONE register set and ONE LDS buffer:
+    // the next tile's global loads are issued before this tile's compute
+    // (HBM latency hides under the MFMAs); the LDS write happens after the
+    // barrier that ends this tile's reads. Keeps the kernel at 2 waves/SIMD
+    // (a second buffer or register set would cross the 256-VGPR cliff).
+    constexpr int NCH = KVBLK * D / 8 / 256;  // chunks per thread
+    shortx8 kreg[NCH], vreg[NCH];
+    const int tid = threadIdx.x;
+
+    auto stage_load = [&](int kv0) {
 #pragma unroll
-            for (int m = 0; m < 8; ++m) { kk[m] = 0; vv[m] = 0; }
-          }
-          *reinterpret_cast<shortx8*>(&sm->k[r * D + swz(r, c)]) = kk;
-          *reinterpret_cast<shortx8*>(&sm->v[r * D + swz(r, c)]) = vv;
+      for (int it = 0; it < NCH; ++it) {
+        const int i = tid + it * 256;
+        const int r = (i * 8) / D, c = (i * 8) % D;
+        if (kv0 + r < T) {
+          kreg[it] = *reinterpret_cast<const shortx8*>(
+              k + kv_base + (long)(kv0 + r) * Hkv * D + c);
+          vreg[it] = *reinterpret_cast<const shortx8*>(
+              v + kv_base + (long)(kv0 + r) * Hkv * D + c);
+        } else {
 #pragma unroll
-          for (int m = 0; m < 8; ++m) {
-            const int d = c + m;
-            sm->kt[d * KVBLK + swz_t(d, r)] = (unsigned short)kk[m];
-          }
+          for (int m = 0; m < 8; ++m) { kreg[it][m] = 0; vreg[it][m] = 0; }
         }
       }
-      __syncthreads();
+    };
+    auto stage_write = [&]() {
+#pragma unroll
+      for (int it = 0; it < NCH; ++it) {
+        const int i = tid + it * 256;
+        const int r = (i * 8) / D, c = (i * 8) % D;
+        *reinterpret_cast<shortx8*>(&sm->k[r * D + swz(r, c)]) = kreg[it];
+        *reinterpret_cast<shortx8*>(&sm->v[r * D + swz(r, c)]) = vreg[it];
+        *reinterpret_cast<shortx8*>(
+            &sm->kt[(c >> 4) * (KVBLK * 16) + r * 16 + (c & 15)]) = kreg[it];
+      }
+    };
+
+    stage_load(0);
+    stage_write();
+    if (n_tiles > 1) stage_load(KVBLK);
+    __syncthreads();
+
+    for (int tile = 0; tile < n_tiles; ++tile) {
+      const int kv0 = tile * KVBLK;
 
       // S^T = K Q^T ; dP^T = V dO^T   (both: col = q = ln31, rows = kv)
       floatx16 s0, s1, dp0, dp1;
@@ -209,20 +245,42 @@ struct DQKernel {
       bf16x8 dsfrag[KVBLK / 16];
       c_layout_to_frags(ds, dsfrag, KVBLK / 16);
 
-      // dQ^T += Kt dS'^T-frags : D[m=d][n=q], col = q = ln31
+      // dQ^T += Kt dS'^T-frags : D[m=d][n=q], col = q = ln31. The Kt
+      // A-operand (lane holds K[kv-run][own d]) comes from the panel image
+      // via hardware transpose reads (see tr_read_b64).
+      {
+        const unsigned kt_base = lds_addr(sm->kt);
+        const int lam = ln31 & 15;
+        const unsigned lane_off = ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 #pragma unroll
-      for (int dblk = 0; dblk < NDBLK; ++dblk) {
+        for (int dblk = 0; dblk < NDBLK; ++dblk) {
+          const unsigned pan = (unsigned)((dblk * 2 + (ln31 >> 4))
+                                          * (KVBLK * 16)) * 2;
+          uintx2 rk[2 * (KVBLK / 16)];
 #pragma unroll
-        for (int s = 0; s < KVBLK / 16; ++s) {
-          const int d = dblk * 32 + ln31;
-          const int col = hi * 8 + 16 * s;
-          bf16x8 kta = *reinterpret_cast<const bf16x8*>(
-              &sm->kt[d * KVBLK + swz_t(d, col)]);
-          acc_dq[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              kta, dsfrag[s], acc_dq[dblk], 0, 0, 0);
+          for (int s = 0; s < KVBLK / 16; ++s) {
+            const unsigned a = pan + (16 * s + 8 * hi) * 32 + lane_off;
+            rk[2 * s] = tr_read_b64(kt_base + a);
+            rk[2 * s + 1] = tr_read_b64(kt_base + a + 128);
+          }
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+          for (int s = 0; s < KVBLK / 16; ++s) {
+            unsigned w[4] = {rk[2 * s][0], rk[2 * s][1],
+                             rk[2 * s + 1][0], rk[2 * s + 1][1]};
+            bf16x8 kta = *reinterpret_cast<bf16x8*>(w);
+            acc_dq[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                kta, dsfrag[s], acc_dq[dblk], 0, 0, 0);
+          }
         }
       }
-      __syncthreads();
+      __syncthreads();  // all waves finished reading tile `tile`
+      if (tile + 1 < n_tiles) {
+        stage_write();                                   // tile+1 -> LDS
+        if (tile + 2 < n_tiles) stage_load((tile + 2) * KVBLK);
+        __syncthreads();                                 // tile+1 ready
+      }
     }
 
     if (qg < T) {
@@ -329,12 +387,12 @@ struct DKDVKernel {
         asm volatile("" :: "v"(r.qq[it][0]), "v"(r.dd[it][0]));
       }
       if constexpr (ABL < 1) {
-#pragma unroll
-        for (int m = 0; m < 8; ++m) {
-          const int d = c + m;
-          tl->qt[d * 32 + swz32(d, row)] = (unsigned short)r.qq[it][m];
-          tl->dot[d * 32 + swz32(d, row)] = (unsigned short)r.dd[it][m];
-        }
+        // panel image [D/16][32 q][16 d]: the thread's 8 consecutive d of
+        // one q row land contiguously -> one 16B write per tensor (was 8
+        // scalar b16 scatter stores with 8-way write-bank conflicts)
+        const int pan = (c >> 4) * 512 + row * 16 + (c & 15);
+        *reinterpret_cast<shortx8*>(&tl->qt[pan]) = r.qq[it];
+        *reinterpret_cast<shortx8*>(&tl->dot[pan]) = r.dd[it];
       }
     }
     if (tid < 32) tl->lse[tid] = r.stat;
@@ -498,18 +556,36 @@ struct DKDVKernel {
       c_layout_to_frags(p_c, pfrag, 2);
       c_layout_to_frags(ds_c, dsfrag, 2);
 
-      // dV += P^T dO ; dK += dS'^T Q
+      // dV += P^T dO ; dK += dS'^T Q  (B-operands via hardware transpose
+      // reads from the panel images; see tr_read_b64 note above)
       if constexpr (ABL != 4) {
+        const unsigned qt_base = lds_addr(cur->qt);
+        const unsigned dot_base = lds_addr(cur->dot);
+        const int lam = ln31 & 15;
+        const unsigned lane_off = ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 #pragma unroll
         for (int dblk = 0; dblk < NDBLK; ++dblk) {
+          const unsigned pan = (unsigned)((dblk * 2 + (ln31 >> 4)) * 512) * 2;
+          uintx2 rd[4], rq[4];
 #pragma unroll
           for (int s = 0; s < 2; ++s) {
-            const int d = dblk * 32 + ln31;
-            const int col = hi * 8 + 16 * s;
-            bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-                &cur->dot[d * 32 + swz32(d, col)]);
-            bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-                &cur->qt[d * 32 + swz32(d, col)]);
+            const unsigned k0 = (16 * s + 8 * hi) * 32;  // *16 elems *2 B
+            const unsigned a = pan + k0 + lane_off;
+            rd[2 * s] = tr_read_b64(dot_base + a);
+            rd[2 * s + 1] = tr_read_b64(dot_base + a + 128);  // k0+4 rows
+            rq[2 * s] = tr_read_b64(qt_base + a);
+            rq[2 * s + 1] = tr_read_b64(qt_base + a + 128);
+          }
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+          for (int s = 0; s < 2; ++s) {
+            unsigned wd[4] = {rd[2 * s][0], rd[2 * s][1],
+                              rd[2 * s + 1][0], rd[2 * s + 1][1]};
+            unsigned wq[4] = {rq[2 * s][0], rq[2 * s][1],
+                              rq[2 * s + 1][0], rq[2 * s + 1][1]};
+            bf16x8 dob = *reinterpret_cast<bf16x8*>(wd);
+            bf16x8 qb = *reinterpret_cast<bf16x8*>(wq);
             acc_dv[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 pfrag[s], dob, acc_dv[dblk], 0, 0, 0);
             acc_dk[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(

@@ -387,6 +387,43 @@ __global__ void mfma_probe_32x32x16_kernel(const float* a, const float* b,
   }
 }
 
+// ---------------------------------------------------------------------------
+// ds_read_b64_tr_b16 semantics probe: stage bf16 values v[i] = i for
+// i in [0,512) into LDS, have each lane of one wave issue the transpose read
+// with a configurable per-lane address, dump the 4 returned elements per
+// lane. mode selects the addressing convention under test:
+//   0: addr = lane*8                    (contiguous 8B rows per lane)
+//   1: addr = (lane&15)*2 + (lane>>4)*128   (strided column interpretation)
+//   2: addr = 0                         (uniform)
+typedef __attribute__((ext_vector_type(2))) unsigned int uintx2;
+
+__global__ void tr16_probe_kernel(unsigned short* out, int mode) {
+  __shared__ unsigned short lds[512];
+  const int tid = threadIdx.x;
+  for (int i = tid; i < 512; i += blockDim.x) lds[i] = (unsigned short)i;
+  __syncthreads();
+  if (tid >= 64) return;
+  unsigned base = (unsigned)(unsigned long long)(void*)&lds[0];
+  unsigned addr;
+  if (mode == 0) addr = base + tid * 8;
+  else if (mode == 1) addr = base + (tid & 15) * 2 + (tid >> 4) * 128;
+  else addr = base;
+  uintx2 r;
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=v"(r) : "v"(addr) : "memory");
+  const unsigned short* rr = reinterpret_cast<const unsigned short*>(&r);
+  for (int j = 0; j < 4; ++j) out[tid * 4 + j] = rr[j];
+}
+
+torch::Tensor tr16_probe(long mode) {
+  auto out = torch::zeros({64, 4}, torch::dtype(torch::kInt16).device(torch::kCUDA));
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (unsigned short*)out.data_ptr(), (int)mode);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b) {
   auto d = torch::zeros({32, 32}, a.options());
   auto stream = at::cuda::getCurrentHIPStream();

@@ -32,6 +32,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor o, torch::Tensor lse,
                                     bool causal);
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
+torch::Tensor tr16_probe(long mode);
 std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
                                            torch::Tensor v, long mode);
 std::vector<torch::Tensor> attn_bwd_dkdv_ablate(torch::Tensor dout,
@@ -56,6 +57,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_scale", &multi_tensor_scale, "in-place scale (K10)");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (K1)");
   m.def("attn_bwd", &attn_bwd, "flash attention backward (K1)");
+  m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16,
         "MFMA fragment-layout probe (verification)");
   m.def("attn_fwd_ablate", &attn_fwd_ablate,
