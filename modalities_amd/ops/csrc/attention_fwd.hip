@@ -22,8 +22,22 @@
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(2))) unsigned int uintx2v;
 
 namespace attn {
+
+// Hardware transpose read (semantics verified by tr16_probe; see the
+// derivation note in attention_bwd.hip): from a [panel][row][16-col] LDS
+// image, lane l receives X[k0+j][d0+(l&15)] — the MFMA A/B k-run.
+__device__ __forceinline__ uintx2v tr_read_b64(unsigned addr_bytes) {
+  uintx2v r;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(r) : "v"(addr_bytes));
+  return r;
+}
+
+__device__ __forceinline__ unsigned lds_addr(const void* p) {
+  return (unsigned)(unsigned long long)p;
+}
 
 constexpr int QBLK = 32;     // q rows per wave
 constexpr int NWAVES = 4;    // waves per workgroup
@@ -114,44 +128,37 @@ struct AttnFwdKernel {
     const int q_hi_wg = min(qblk0 + WG_Q - 1, T - 1);  // causal upper bound
     const int n_tiles = (min(q_hi_wg, T - 1)) / KVBLK + 1;
 
+    // V goes into a panel image [D/16][KVBLK][16] consumed via hardware
+    // transpose reads (vectorized writes; the v1 scalar scatter stores had
+    // 8-way write-bank conflicts). Staging stays single-buffered in-loop:
+    // the T14 issue-early variant costs 32 VGPRs and drops this kernel
+    // from 2 waves/SIMD to 1, which measured slower.
+    constexpr int NCH = KVBLK * D / 8 / (NWAVES * 64);
+    const int tid = threadIdx.x;
+
     for (int tile = 0; tile < n_tiles; ++tile) {
       const int kv0 = tile * KVBLK;
-      // ---- stage K and V^T into LDS (whole workgroup, coalesced reads) ----
-      {
-        const int tid = threadIdx.x;
-        const int nthreads = NWAVES * 64;
-        // each thread copies 8 bf16 per iteration
-        const int total8 = KVBLK * D / 8;
-        for (int i = tid; i < total8; i += nthreads) {
-          const int r = (i * 8) / D;        // kv row in tile
-          const int c = (i * 8) % D;        // d col
-          shortx8 val;
-          if (kv0 + r < T) {
-            val = *reinterpret_cast<const shortx8*>(
-                k + kv_base + (long)(kv0 + r) * Hkv * D + c);
-          } else {
-            for (int m = 0; m < 8; ++m) val[m] = 0;
-          }
-          if constexpr (ABL < 2)
-            *reinterpret_cast<shortx8*>(&sm->k[r * D + swz(r, c)]) = val;
-          // V: read the same global layout, write transposed (scalar LDS
-          // stores — v1; upgrade path: ds_read_b64_tr_b16 per guide T10)
-          shortx8 vv;
-          if (kv0 + r < T) {
-            vv = *reinterpret_cast<const shortx8*>(
-                v + kv_base + (long)(kv0 + r) * Hkv * D + c);
-          } else {
-            for (int m = 0; m < 8; ++m) vv[m] = 0;
-          }
-          if constexpr (ABL < 1) {
 #pragma unroll
-            for (int m = 0; m < 8; ++m) {
-              const int d = c + m;
-              sm->vt[d * KVBLK + swz_t(d, r)] = (unsigned short)vv[m];
-            }
-          } else {
-            asm volatile("" :: "v"(vv[0]));  // keep loads alive (rule 17)
-          }
+      for (int it = 0; it < NCH; ++it) {
+        const int i = tid + it * NWAVES * 64;
+        const int r = (i * 8) / D, c = (i * 8) % D;
+        shortx8 kk, vv;
+        if (kv0 + r < T) {
+          kk = *reinterpret_cast<const shortx8*>(
+              k + kv_base + (long)(kv0 + r) * Hkv * D + c);
+          vv = *reinterpret_cast<const shortx8*>(
+              v + kv_base + (long)(kv0 + r) * Hkv * D + c);
+        } else {
+#pragma unroll
+          for (int m = 0; m < 8; ++m) { kk[m] = 0; vv[m] = 0; }
+        }
+        if constexpr (ABL < 2)
+          *reinterpret_cast<shortx8*>(&sm->k[r * D + swz(r, c)]) = kk;
+        if constexpr (ABL < 1) {
+          *reinterpret_cast<shortx8*>(
+              &sm->vt[(c >> 4) * (KVBLK * 16) + r * 16 + (c & 15)]) = vv;
+        } else {
+          asm volatile("" :: "v"(vv[0]));  // keep loads alive (rule 17)
         }
       }
       __syncthreads();
@@ -236,18 +243,30 @@ struct AttnFwdKernel {
         pfrag[s] = *reinterpret_cast<bf16x8*>(frag_words);
       }
 
-      // ---- O^T += V^T P^T : A = V^T rows (consecutive kv), B = P^T -------
+      // ---- O^T += V^T P^T : A = V^T (lane holds V[kv-run][own d]) via
+      // hardware transpose reads from the panel image --------------------
+      {
+        const unsigned vt_base = lds_addr(sm->vt);
+        const int lam = ln31 & 15;
+        const unsigned lane_off = ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 #pragma unroll
-      for (int dblk = 0; dblk < NDBLK; ++dblk) {
+        for (int dblk = 0; dblk < NDBLK; ++dblk) {
+          const unsigned pan = (unsigned)((dblk * 2 + (ln31 >> 4))
+                                          * (KVBLK * 16)) * 2;
+          // small batches (one fragment per wait) keep this kernel under
+          // the 256-VGPR 2-waves/SIMD line
 #pragma unroll
-        for (int s = 0; s < KVBLK / 16; ++s) {
-          const int d = dblk * 32 + ln31;
-          const int col = hi * 8 + 16 * s;  // kv index within tile
-          bf16x8 va = *reinterpret_cast<const bf16x8*>(
-              &sm->vt[d * KVBLK + swz_t(d, col)]);
-          acc_o[dblk] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, pfrag[s], acc_o[dblk],
-                                                      0, 0, 0);
+          for (int s = 0; s < KVBLK / 16; ++s) {
+            const unsigned a = pan + (16 * s + 8 * hi) * 32 + lane_off;
+            uintx2v r0 = tr_read_b64(vt_base + a);
+            uintx2v r1 = tr_read_b64(vt_base + a + 128);
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_sched_barrier(0);
+            unsigned w[4] = {r0[0], r0[1], r1[0], r1[1]};
+            bf16x8 va = *reinterpret_cast<bf16x8*>(w);
+            acc_o[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                va, pfrag[s], acc_o[dblk], 0, 0, 0);
+          }
         }
       }
       __syncthreads();
@@ -285,7 +304,7 @@ struct AttnFwdKernel {
 };
 
 template <int D, int ABL = 0>
-__global__ __launch_bounds__(256) void attn_fwd_kernel(
+__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
     unsigned short* o, float* lse, int B, int T, int Hq, int Hkv, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
