@@ -1,0 +1,157 @@
+"""GPT2 model variant coverage (reference model:
+tests/models/test_causal_self_attention.py, test_rotary_qkv_transform.py,
+layer norm tests): positional modes, GELU vs SwiGLU, QK-norm, weight tying,
+GQA, activation checkpointing equivalence, debug hooks, profilers."""
+
+import pytest
+import torch
+
+from modalities_amd.models.gpt2 import (ActivationType, AttentionImplementation,
+                                        GPT2LLM, GPT2LLMConfig, PositionTypes)
+
+VOCAB = 128
+
+
+def cfg(**kw):
+    d = dict(vocab_size=VOCAB, n_layer=2, n_head_q=4, n_head_kv=2, n_embd=64,
+             ffn_hidden=256, sequence_length=32, seed=1, dropout=0.0)
+    d.update(kw)
+    return GPT2LLMConfig(**d)
+
+
+def run_fwd(model, seed=9, batch=2, seqlen=16):
+    g = torch.Generator().manual_seed(seed)
+    ids = torch.randint(0, VOCAB, (batch, seqlen), generator=g)
+    return model({"input_ids": ids})["logits"]
+
+
+@pytest.mark.parametrize("poe", [PositionTypes.ABSOLUTE, PositionTypes.NOPE])
+@pytest.mark.parametrize("act", [ActivationType.GELU, ActivationType.SWIGLU])
+def test_gpt2_variants_forward_backward(poe, act):
+    model = GPT2LLM(cfg(poe_type=poe, activation_type=act))
+    out = run_fwd(model)
+    assert out.shape == (2, 16, VOCAB)
+    out.float().sum().backward()
+    assert model.wte.weight.grad is not None
+    if poe == PositionTypes.ABSOLUTE:
+        assert model.wpe is not None and model.wpe.weight.grad is not None
+
+
+def test_gpt2_qk_norm_and_weight_tying():
+    model = GPT2LLM(cfg(use_qk_norm=True, use_weight_tying=True))
+    assert model.lm_head.weight is model.wte.weight
+    out = run_fwd(model)
+    assert torch.isfinite(out.float()).all()
+
+
+def test_attention_impls_agree():
+    """pytorch_flash (SDPA) and manual O(T^2) paths must agree; both serve
+    as CPU references for the HIP kernel (tested on GPU)."""
+    m1 = GPT2LLM(cfg(attention_implementation=AttentionImplementation.PYTORCH_FLASH))
+    m2 = GPT2LLM(cfg(attention_implementation=AttentionImplementation.MANUAL))
+    m2.load_state_dict(m1.state_dict())
+    m1.eval(), m2.eval()
+    o1, o2 = run_fwd(m1), run_fwd(m2)
+    torch.testing.assert_close(o1, o2, rtol=1e-4, atol=1e-5)
+
+
+def test_causality():
+    model = GPT2LLM(cfg())
+    model.eval()
+    g = torch.Generator().manual_seed(4)
+    ids = torch.randint(0, VOCAB, (1, 16), generator=g)
+    out1 = model({"input_ids": ids})["logits"]
+    ids2 = ids.clone()
+    ids2[0, -1] = (ids2[0, -1] + 1) % VOCAB
+    out2 = model({"input_ids": ids2})["logits"]
+    torch.testing.assert_close(out1[0, :-1], out2[0, :-1], rtol=1e-4, atol=1e-5)
+
+
+# ---- activation checkpointing ----------------------------------------------
+
+@pytest.mark.parametrize("variant", ["full_activation_checkpointing",
+                                     "selective_layer_activation_checkpointing",
+                                     "selective_op_activation_checkpointing"])
+def test_activation_checkpointing_grad_equivalence(variant):
+    from modalities_amd.training.activation_checkpointing import (
+        ActivationCheckpointingVariant, apply_activation_checkpointing_)
+    torch.manual_seed(0)
+    ref = GPT2LLM(cfg())
+    torch.manual_seed(0)
+    ckpt = GPT2LLM(cfg())
+    apply_activation_checkpointing_(ckpt, ActivationCheckpointingVariant(variant),
+                                    every_k_layers=2)
+    g = torch.Generator().manual_seed(3)
+    ids = torch.randint(0, VOCAB, (2, 17), generator=g)
+    for model in (ref, ckpt):
+        out = model({"input_ids": ids[:, :-1]})["logits"]
+        loss = torch.nn.functional.cross_entropy(out.reshape(-1, VOCAB).float(),
+                                                 ids[:, 1:].reshape(-1))
+        loss.backward()
+    for (n1, p1), (n2, p2) in zip(ref.named_parameters(), ckpt.named_parameters()):
+        torch.testing.assert_close(p1.grad, p2.grad, rtol=1e-5, atol=1e-6,
+                                   msg=lambda m: f"{n1}/{n2}: {m}")
+
+
+def test_activation_checkpointing_composes_with_sharding_engine():
+    from modalities_amd.parallel.fsdp import XGMIShardedModel
+    from modalities_amd.training.activation_checkpointing import (
+        ActivationCheckpointingVariant, apply_activation_checkpointing_)
+    torch.manual_seed(0)
+    model = GPT2LLM(cfg())
+    apply_activation_checkpointing_(
+        model, ActivationCheckpointingVariant.FULL_ACTIVATION_CHECKPOINTING)
+    sharded = XGMIShardedModel.from_transformer(model, torch.device("cpu"),
+                                                param_dtype=torch.float32)
+    g = torch.Generator().manual_seed(3)
+    ids = torch.randint(0, VOCAB, (2, 17), generator=g)
+    out = sharded({"input_ids": ids[:, :-1]})["logits"]
+    loss = torch.nn.functional.cross_entropy(out.reshape(-1, VOCAB).float(),
+                                             ids[:, 1:].reshape(-1))
+    loss.backward()
+    sharded.backward_epilogue()
+    assert all(u.grad_fresh for u in sharded.units)
+
+
+# ---- debug components -------------------------------------------------------
+
+def test_nan_hook_raises():
+    from modalities_amd.utils.debug_components import register_nan_hooks
+    model = GPT2LLM(cfg())
+    register_nan_hooks(model, module_name_filter="lm_head")
+    with torch.no_grad():
+        model.lm_head.weight[0, 0] = float("nan")
+    with pytest.raises(RuntimeError, match="Non-finite"):
+        run_fwd(model)
+
+
+def test_tensor_stats_hook(tmp_path):
+    import json
+
+    from modalities_amd.utils.debug_components import register_tensor_stats_hooks
+    model = GPT2LLM(cfg())
+    n = register_tensor_stats_hooks(model, tmp_path / "stats.jsonl",
+                                    module_name_filter="lm_head_norm")
+    assert n == 1
+    run_fwd(model)
+    recs = [json.loads(ln) for ln in (tmp_path / "stats.jsonl").read_text().splitlines()]
+    assert recs and recs[0]["module"] == "lm_head_norm"
+    assert "mean" in recs[0] and "absmax" in recs[0]
+
+
+# ---- profilers --------------------------------------------------------------
+
+def test_steppable_profilers(tmp_path):
+    from modalities_amd.utils.profilers import (SteppableKernelProfiler,
+                                                SteppableNoProfiler, get_profiler)
+    model = GPT2LLM(cfg())
+    prof = SteppableKernelProfiler(tmp_path, wait=0, warmup=1, active=1, repeat=1)
+    with prof:
+        for _ in range(len(prof)):
+            run_fwd(model)
+            prof.step()
+    assert list(tmp_path.glob("trace_step*.json"))
+    assert list(tmp_path.glob("key_averages_step*.txt"))
+    # rank filtering
+    assert isinstance(get_profiler("kernel", tmp_path, global_rank=1,
+                                   tracked_ranks=[0]), SteppableNoProfiler)
