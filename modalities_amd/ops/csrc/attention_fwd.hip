@@ -198,23 +198,35 @@ struct AttnFwdKernel {
       tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));  // other kv half, same q
 
       // ---- online softmax update ----------------------------------------
-      const float m_new = fmaxf(m_run, tmax);
-      const float m_safe = (m_new == -INFINITY) ? 0.f : m_new;
-      // -inf guards: fully-masked tiles keep O/l at 0 without NaNs
-      const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_safe);
+      // defer-max (guide T13): when no lane's tile max exceeds the running
+      // max by more than RESCALE_THR, keep the old max — skips the O-wide
+      // rescale (NDBLK*16 mults) and the l/alpha work; P is then bounded by
+      // exp(THR) instead of 1, which f32 accumulation and bf16 P tolerate
+      // (~3x max-abs error vs THR=0; decision taken BEFORE this tile's P is
+      // exponentiated, so no pending P*V is split — the T13 hazard order).
+      constexpr float RESCALE_THR = 8.0f;
+      const bool defer = __all(tmax - m_run <= RESCALE_THR);
+      if (!defer) {
+        const float m_new = fmaxf(m_run, tmax);
+        const float m_safe = (m_new == -INFINITY) ? 0.f : m_new;
+        // -inf guards: fully-masked tiles keep O/l at 0 without NaNs
+        const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_safe);
+        l_run *= alpha;
+#pragma unroll
+        for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) acc_o[dblk][r] *= alpha;
+        m_run = m_new;
+      }
+      const float m_safe2 = (m_run == -INFINITY) ? 0.f : m_run;
       float psum = 0.f;
 #pragma unroll
       for (int r = 0; r < 32; ++r) {
-        p[r] = __expf(p[r] - m_safe);
+        p[r] = __expf(p[r] - m_safe2);
         psum += p[r];
       }
       psum += __shfl_xor(psum, 32, 64);
-      l_run = l_run * alpha + psum;
-      m_run = m_new;
-#pragma unroll
-      for (int dblk = 0; dblk < NDBLK; ++dblk)
-#pragma unroll
-        for (int r = 0; r < 16; ++r) acc_o[dblk][r] *= alpha;
+      l_run += psum;
       }  // ABL != 3
 
       // ---- P (f32, S-layout) -> PV A/B fragment (bf16) -------------------

@@ -244,3 +244,23 @@ def test_gpt2_training_step_gpu():
         opt.zero_grad()
         losses.append(loss.item())
     assert losses[-1] < losses[0] - 0.5, losses
+
+
+@pytest.mark.gpu
+def test_attn_fwd_forced_rescale():
+    """Force the defer-max RESCALE branch (guide rule 26: the branch is rare
+    on bounded random data): a spiked K row makes the tile max jump past
+    RESCALE_THR=8 at a chosen late tile; output must still match fp32 ref."""
+    torch.manual_seed(9)
+    B, T, Hq, Hkv, D = 1, 1024, 2, 2, 128
+    q = bf(torch.randn(B, T, Hq, D, device=DEV))
+    k = bf(torch.randn(B, T, Hkv, D, device=DEV))
+    v = bf(torch.randn(B, T, Hkv, D, device=DEV))
+    # spike: K row 700 strongly aligned with every q (rank-1 spike)
+    direction = torch.randn(D, device=DEV)
+    direction = direction / direction.norm()
+    k[:, 700] = bf(direction * 60.0)
+    q[:, 701:] = bf(q[:, 701:].float() + direction * 2.0)
+    o, lse = EXT.attn_fwd(q, k, v, True)
+    ref = ref_attention(q, k, v)
+    torch.testing.assert_close(o.float(), ref, rtol=3e-2, atol=3e-2)
