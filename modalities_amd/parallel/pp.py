@@ -213,6 +213,21 @@ class PipelineSchedule:
              loss_fn: Optional[Callable], losses_out: Optional[list] = None):
         raise NotImplementedError
 
+    def broadcast_mean_loss(self, losses: list) -> torch.Tensor:
+        """Mean micro-batch loss, broadcast from the last stage so every PP
+        rank logs the true value (the trainer's loss reduction is then
+        PP-agnostic)."""
+        ranks = dist.get_process_group_ranks(self.group) if self.group is not None \
+            else list(range(dist.get_world_size()))
+        src = ranks[self.num_stages - 1]
+        if self.is_last and losses:
+            val = torch.stack([l.detach().float().cpu() for l in losses]).mean()
+        else:
+            val = torch.zeros((), dtype=torch.float32)
+        buf = val.to(self.device) if self.device.type == "cuda" else val
+        dist.broadcast(buf, src=src, group=self.group)
+        return buf.cpu()
+
 
 class ScheduleGPipe(PipelineSchedule):
     """All forwards, then all backwards (reference maps

@@ -46,7 +46,8 @@ class Trainer:
                  evaluation_interval_in_steps: int = 0,
                  checkpointing_interval_in_steps: int = 0,
                  training_log_interval_in_steps: int = 1,
-                 gc_freq: int = 1000, profiler=None, device=None):
+                 gc_freq: int = 1000, profiler=None, device=None,
+                 pp_schedule=None):
         self.global_rank = global_rank
         self.progress_publisher = progress_publisher
         self.evaluation_result_publisher = evaluation_result_publisher
@@ -60,6 +61,7 @@ class Trainer:
         self.gc = GarbageCollection(gc_freq)
         self.profiler = profiler
         self.device = device or torch.device("cpu")
+        self.pp_schedule = pp_schedule  # PipelineSchedule when PP is active
         self.training_progress = TrainingProgress(
             num_seen_steps_current_run=0, num_seen_tokens_current_run=0,
             num_target_steps=num_target_steps, num_target_tokens=num_target_tokens,
@@ -70,7 +72,27 @@ class Trainer:
     def _train_batch(self, batch: DatasetBatch, model, optimizer, scheduler,
                      loss_fun: Loss, micro_batch_id: int):
         """One micro-batch: forward, loss, backward; on accumulation boundary
-        clip + step + zero. Returns (step_performed, loss detached, grad_norm)."""
+        clip + step + zero. Returns (step_performed, loss detached, grad_norm).
+
+        With a pipeline schedule, the schedule's internal micro-batching IS
+        the accumulation: one call = one full fwd/bwd over its splits
+        (reference trainer.py:162-177 pp_schedule.step dispatch)."""
+        if self.pp_schedule is not None:
+            inputs = next(iter(batch.samples.values()))
+            targets = next(iter(batch.targets.values()))
+            losses: list = []
+            self.pp_schedule.step(inputs, targets, loss_fun, losses)
+            loss = self.pp_schedule.broadcast_mean_loss(losses)
+            grad_norm = None
+            if (micro_batch_id + 1) % self.gradient_acc_steps == 0:
+                if self.gradient_clipper is not None:
+                    grad_norm = self.gradient_clipper(model)
+                optimizer.step()
+                if scheduler is not None:
+                    scheduler.step()
+                optimizer.zero_grad()
+                return True, loss, grad_norm
+            return False, loss, None
         sharded = isinstance(model, XGMIShardedModel)
         result_batch = model_predict_batch(model, batch)
         loss = loss_fun(result_batch)
