@@ -34,6 +34,16 @@ def build_model_cfg(model_name: str):
             vocab_size=50304, n_layer=32, n_head_q=20, n_head_kv=20,
             n_embd=2560, ffn_hidden=10240, sequence_length=4096,
             activation_type="swiglu", use_weight_tying=False)
+    if model_name == "gpt2-8b":  # Llama-8B-like (BASELINE.json config 3)
+        return GPT2LLMConfig(
+            vocab_size=50304, n_layer=32, n_head_q=32, n_head_kv=8,
+            n_embd=4096, ffn_hidden=21504, sequence_length=8192,
+            activation_type="swiglu", use_weight_tying=False)
+    if model_name == "gpt2-70b":  # 288GB-HBM sizing config (BASELINE.json 5)
+        return GPT2LLMConfig(
+            vocab_size=50304, n_layer=80, n_head_q=64, n_head_kv=8,
+            n_embd=8192, ffn_hidden=43008, sequence_length=4096,
+            activation_type="swiglu", use_weight_tying=False)
     if model_name == "gpt2-tiny":  # smoke/debug
         return GPT2LLMConfig(
             vocab_size=50304, n_layer=4, n_head_q=4, n_head_kv=4,
@@ -47,10 +57,14 @@ def main():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--micro-batch", type=int, default=2)
-    p.add_argument("--seq-len", type=int, default=4096)
+    p.add_argument("--seq-len", type=int, default=None,
+                   help="default: the model config's native length")
     p.add_argument("--model", type=str, default="gpt2-2.7b")
     p.add_argument("--blocks-per-unit", type=int, default=4)
     p.add_argument("--reshard", action="store_true")
+    p.add_argument("--ac", action="store_true",
+                   help="full activation checkpointing")
+    p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -75,21 +89,48 @@ def main():
     from modalities_amd.utils.mfu import GPT2MFUCalculator, detect_device_peak_flops
 
     cfg = build_model_cfg(args.model)
+    if args.seq_len is None:
+        args.seq_len = cfg.sequence_length
     cfg.sequence_length = args.seq_len
     torch.manual_seed(1234)  # identical init on all ranks
     model = GPT2LLM(cfg)
     num_params = sum(p_.numel() for p_ in model.parameters())
 
-    sharded = XGMIShardedModel.from_transformer(
-        model, device, blocks_per_unit=args.blocks_per_unit,
-        param_dtype=torch.bfloat16 if on_gpu else torch.float32,
-        reshard_after_forward=args.reshard)
+    mesh = None
+    dp_world, dp_rank = world, rank
+    if args.tp > 1:
+        from modalities_amd.parallel.mesh import (ParallelismDegrees,
+                                                  get_device_mesh)
+        from modalities_amd.parallel.tp import \
+            get_gpt2_tensor_parallelized_model
+        mesh = get_device_mesh(world, rank, tensor_parallel_degree=args.tp)
+        model = get_gpt2_tensor_parallelized_model(model, device_mesh=mesh)
+        dp_world = mesh.dp_degree
+        dp_rank = mesh.dp_rank
+    if args.ac:
+        from modalities_amd.training.activation_checkpointing import (
+            ActivationCheckpointingVariant, apply_activation_checkpointing_)
+        apply_activation_checkpointing_(
+            model, ActivationCheckpointingVariant.FULL_ACTIVATION_CHECKPOINTING)
+
+    if mesh is not None:
+        from modalities_amd.models.model_factory import ModelFactory
+        sharded = ModelFactory.get_sharded_model(
+            model, device_mesh=mesh, blocks_per_unit=args.blocks_per_unit,
+            reshard_after_forward=args.reshard,
+            param_dtype="bf16" if on_gpu else "fp32", device=device)
+    else:
+        sharded = XGMIShardedModel.from_transformer(
+            model, device, blocks_per_unit=args.blocks_per_unit,
+            param_dtype=torch.bfloat16 if on_gpu else torch.float32,
+            reshard_after_forward=args.reshard)
     opt = get_adam_w(sharded, lr=3e-4, weight_decay=0.1)
 
     B, T, V = args.micro_batch, args.seq_len, cfg.vocab_size
 
     def make_batch(step: int):
-        g = torch.Generator().manual_seed(10_000 + step * world + rank)
+        # TP ranks in one dp group must see the SAME data
+        g = torch.Generator().manual_seed(10_000 + step * dp_world + dp_rank)
         ids = torch.randint(0, V, (B, T + 1), generator=g)
         return ids[:, :-1].to(device, non_blocking=True), \
             ids[:, 1:].to(device, non_blocking=True)
@@ -128,7 +169,7 @@ def main():
     elapsed = t.item()
 
     ms_per_step = elapsed / args.steps * 1000
-    tokens_per_step_global = B * T * world
+    tokens_per_step_global = B * T * dp_world
     tokens_per_s = tokens_per_step_global * args.steps / elapsed
 
     # Reference baseline: 2.7B 8xA100 FULL_SHARD mbs=2 -> 18.63 samples/s
@@ -161,7 +202,9 @@ def main():
             "mfu": mfu,
             "config": {"model": args.model, "global_batch": B * world,
                        "micro_batch": B, "seq_len": T,
-                       "parallelism": f"dp{world}_fullshard",
+                       "parallelism": (f"dp{dp_world}_tp{args.tp}" if args.tp > 1
+                                       else f"dp{world}_fullshard")
+                       + ("_ac" if args.ac else ""),
                        "num_params": num_params},
         }), flush=True)
 
