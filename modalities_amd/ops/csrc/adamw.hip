@@ -38,6 +38,41 @@ __global__ void adamw_masked_kernel(float* __restrict__ p,
   }
 }
 
+// Graph-safe variant: the Adam step count lives in DEVICE memory so a
+// hipGraph replay of the training step sees the advancing bias correction
+// (a host-side bc baked into kernel args would be frozen at capture).
+__global__ void adamw_masked_devstep_kernel(float* __restrict__ p,
+                                            const float* __restrict__ g,
+                                            float* __restrict__ m,
+                                            float* __restrict__ v,
+                                            const float* __restrict__ wd_mask,
+                                            const int* __restrict__ step,
+                                            long n4, float lr, float beta1,
+                                            float beta2, float eps, float wd) {
+  const float t = (float)*step;
+  const float bc1 = 1.0f - __powf(beta1, t);
+  const float bc2 = 1.0f - __powf(beta2, t);
+  const float step_size = lr / bc1;
+  const float inv_bc2 = 1.0f / bc2;
+  floatx4* p4 = reinterpret_cast<floatx4*>(p);
+  const floatx4* g4 = reinterpret_cast<const floatx4*>(g);
+  floatx4* m4 = reinterpret_cast<floatx4*>(m);
+  floatx4* v4 = reinterpret_cast<floatx4*>(v);
+  const floatx4* w4 = reinterpret_cast<const floatx4*>(wd_mask);
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    floatx4 pp = p4[i], gg = g4[i], mm = m4[i], vv = v4[i], ww = w4[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      pp[j] *= (1.0f - lr * wd * ww[j]);
+      mm[j] = beta1 * mm[j] + (1.0f - beta1) * gg[j];
+      vv[j] = beta2 * vv[j] + (1.0f - beta2) * gg[j] * gg[j];
+      pp[j] -= step_size * mm[j] / (sqrtf(vv[j] * inv_bc2) + eps);
+    }
+    p4[i] = pp; m4[i] = mm; v4[i] = vv;
+  }
+}
+
 __global__ void sqsum_kernel(const float* __restrict__ x, long n4,
                              float* __restrict__ out) {
   __shared__ float scratch[256 / WAVE_SIZE];
@@ -86,6 +121,25 @@ void fused_adamw_masked(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      m.data_ptr<float>(), v.data_ptr<float>(),
                      wd_mask.data_ptr<float>(), n4, (float)lr, (float)beta1,
                      (float)beta2, (float)eps, (float)wd, (float)bc1, (float)bc2);
+  HIP_CHECK_KERNEL();
+}
+
+void fused_adamw_masked_devstep(torch::Tensor p, torch::Tensor g,
+                                torch::Tensor m, torch::Tensor v,
+                                torch::Tensor wd_mask, torch::Tensor step,
+                                double lr, double beta1, double beta2,
+                                double eps, double wd) {
+  TORCH_CHECK(p.is_cuda() && p.dtype() == torch::kFloat32 && p.is_contiguous());
+  TORCH_CHECK(p.numel() % 4 == 0, "flat shard must be divisible by 4");
+  TORCH_CHECK(step.is_cuda() && step.dtype() == torch::kInt32);
+  long n4 = p.numel() / 4;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(adamw_masked_devstep_kernel, dim3(grid_for(n4, 256)),
+                     dim3(256), 0, stream, p.data_ptr<float>(),
+                     g.data_ptr<float>(), m.data_ptr<float>(),
+                     v.data_ptr<float>(), wd_mask.data_ptr<float>(),
+                     step.data_ptr<int>(), n4, (float)lr, (float)beta1,
+                     (float)beta2, (float)eps, (float)wd);
   HIP_CHECK_KERNEL();
 }
 

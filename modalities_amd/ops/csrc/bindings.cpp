@@ -19,6 +19,11 @@ void fused_adamw_masked(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                         torch::Tensor v, torch::Tensor wd_mask, double lr,
                         double beta1, double beta2, double eps, double wd,
                         double bc1, double bc2);
+void fused_adamw_masked_devstep(torch::Tensor p, torch::Tensor g,
+                                torch::Tensor m, torch::Tensor v,
+                                torch::Tensor wd_mask, torch::Tensor step,
+                                double lr, double beta1, double beta2,
+                                double eps, double wd);
 void fused_adamw(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                  std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
                  double lr, double beta1, double beta2, double eps, double wd,
@@ -53,6 +58,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw_masked", &fused_adamw_masked,
         "AdamW on a flat fp32 shard with per-element wd mask (K9)");
   m.def("fused_adamw", &fused_adamw, "multi-tensor AdamW (K9)");
+  m.def("fused_adamw_masked_devstep", &fused_adamw_masked_devstep,
+        "AdamW with device-resident step counter (hipGraph-safe)");
   m.def("multi_tensor_sqsum", &multi_tensor_sqsum, "sum of squares (K10)");
   m.def("multi_tensor_scale", &multi_tensor_scale, "in-place scale (K10)");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (K1)");

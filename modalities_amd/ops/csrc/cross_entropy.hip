@@ -56,7 +56,9 @@ __global__ void ce_bwd_kernel(const unsigned short* __restrict__ logits,
                               const long* __restrict__ targets,
                               const float* __restrict__ lse,
                               unsigned short* __restrict__ dlogits,
-                              float scale, int V, long ignore_index) {
+                              const float* __restrict__ scale_ptr, int V,
+                              long ignore_index) {
+  const float scale = *scale_ptr;  // device-resident: hipGraph-capture-safe
   const long row = blockIdx.x;
   const long tgt = targets[row];
   const unsigned short* lr = logits + row * (long)V;
@@ -113,12 +115,15 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor targets,
   const long N = logits.size(0);
   const int V = logits.size(1);
   auto dlogits = torch::empty_like(logits);
-  float scale_f = scale.item<float>();
+  // scale stays on device (a .item() here is a host sync that would break
+  // hipGraph capture of the training step)
+  auto scale_dev = scale.to(logits.device(), torch::kFloat32).contiguous();
   auto stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(ce_bwd_kernel, dim3(N), dim3(BLOCK), 0, stream,
                      (const unsigned short*)logits.data_ptr(),
                      targets.data_ptr<long>(), lse.data_ptr<float>(),
-                     (unsigned short*)dlogits.data_ptr(), scale_f, V, ignore_index);
+                     (unsigned short*)dlogits.data_ptr(),
+                     scale_dev.data_ptr<float>(), V, ignore_index);
   HIP_CHECK_KERNEL();
   return dlogits;
 }

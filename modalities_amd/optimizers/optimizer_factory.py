@@ -34,6 +34,11 @@ class ShardedAdamW(torch.optim.Optimizer):
             st["step"] = 0
             st["exp_avg"] = torch.zeros_like(u.master_shard)
             st["exp_avg_sq"] = torch.zeros_like(u.master_shard)
+        # Device-resident shared step counter (hipGraph-safe bias correction:
+        # a replayed training step advances it on-device; host-computed bc
+        # would freeze at capture). Lazily initialized from the CPU step so
+        # warmstart loads are honored.
+        self._step_dev = None
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -41,6 +46,17 @@ class ShardedAdamW(torch.optim.Optimizer):
         group = self.param_groups[0]
         lr, (beta1, beta2) = group["lr"], group["betas"]
         eps, wd = group["eps"], group["weight_decay"]
+        from modalities_amd.ops.backend import hip_available, hip_ext
+        on_gpu = hip_available() and self.sharded_model.units \
+            and self.sharded_model.units[0].master_shard.is_cuda
+        if on_gpu:
+            if self._step_dev is None:
+                start = max((self.state[u.master_shard].get("step", 0)
+                             for u in self.sharded_model.units), default=0)
+                self._step_dev = torch.tensor(
+                    start, dtype=torch.int32,
+                    device=self.sharded_model.units[0].master_shard.device)
+            self._step_dev += 1  # on-device: advances under hipGraph replay
         for u in self.sharded_model.units:
             if not u.grad_fresh:
                 continue  # no grads reduced for this unit this step
@@ -50,13 +66,13 @@ class ShardedAdamW(torch.optim.Optimizer):
             m, v = st["exp_avg"], st["exp_avg_sq"]
             g = u.grad_shard
             p = u.master_shard
-            bc1 = 1.0 - beta1 ** step
-            bc2 = 1.0 - beta2 ** step
-            from modalities_amd.ops.backend import hip_available, hip_ext
-            if hip_available() and p.is_cuda:
-                hip_ext().fused_adamw_masked(p, g, m, v, u.wd_mask_shard,
-                                             lr, beta1, beta2, eps, wd, bc1, bc2)
+            if on_gpu:
+                hip_ext().fused_adamw_masked_devstep(
+                    p, g, m, v, u.wd_mask_shard, self._step_dev,
+                    lr, beta1, beta2, eps, wd)
             else:
+                bc1 = 1.0 - beta1 ** step
+                bc2 = 1.0 - beta2 ** step
                 p.mul_(1.0 - lr * wd * u.wd_mask_shard)
                 m.mul_(beta1).add_(g, alpha=1 - beta1)
                 v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
