@@ -67,12 +67,19 @@ def convert_model_checkpoint(modalities_model: GPT2LLM):
     sd = {}
     src = modalities_model.state_dict()
     sd["model.embed_tokens.weight"] = src["wte.weight"]
+    head_dim_kv = cfg.n_embd // cfg.n_head_q * cfg.n_head_kv
     for i in range(cfg.n_layer):
         p, hp = f"blocks.{i}", f"model.layers.{i}"
         sd[f"{hp}.input_layernorm.weight"] = src[f"{p}.attention_norm.weight"]
-        sd[f"{hp}.self_attn.q_proj.weight"] = src[f"{p}.attn.q_attn.weight"]
-        sd[f"{hp}.self_attn.k_proj.weight"] = src[f"{p}.attn.k_attn.weight"]
-        sd[f"{hp}.self_attn.v_proj.weight"] = src[f"{p}.attn.v_attn.weight"]
+        if cfg.fused_qkv:
+            w = src[f"{p}.attn.qkv_attn.weight"]
+            sd[f"{hp}.self_attn.q_proj.weight"] = w[:cfg.n_embd]
+            sd[f"{hp}.self_attn.k_proj.weight"] = w[cfg.n_embd:cfg.n_embd + head_dim_kv]
+            sd[f"{hp}.self_attn.v_proj.weight"] = w[cfg.n_embd + head_dim_kv:]
+        else:
+            sd[f"{hp}.self_attn.q_proj.weight"] = src[f"{p}.attn.q_attn.weight"]
+            sd[f"{hp}.self_attn.k_proj.weight"] = src[f"{p}.attn.k_attn.weight"]
+            sd[f"{hp}.self_attn.v_proj.weight"] = src[f"{p}.attn.v_attn.weight"]
         sd[f"{hp}.self_attn.o_proj.weight"] = src[f"{p}.attn.c_proj.weight"]
         sd[f"{hp}.post_attention_layernorm.weight"] = src[f"{p}.ffn_norm.weight"]
         sd[f"{hp}.mlp.gate_proj.weight"] = src[f"{p}.mlp.W.weight"]

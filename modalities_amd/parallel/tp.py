@@ -279,6 +279,10 @@ def get_gpt2_tensor_parallelized_model(model, device_mesh=None, group=None,
 
     for block in model.blocks:
         attn = block.attn
+        if getattr(attn, "fused_qkv", False):
+            raise NotImplementedError(
+                "tensor parallelism requires the unfused q/k/v layout "
+                "(build the model with fused_qkv=False)")
         _shard_linear_(attn.q_attn, tp_rank, tp_size, dim=0)
         _shard_linear_(attn.k_attn, tp_rank, tp_size, dim=0)
         _shard_linear_(attn.v_attn, tp_rank, tp_size, dim=0)

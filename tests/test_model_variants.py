@@ -155,3 +155,21 @@ def test_steppable_profilers(tmp_path):
     # rank filtering
     assert isinstance(get_profiler("kernel", tmp_path, global_rank=1,
                                    tracked_ranks=[0]), SteppableNoProfiler)
+
+
+def test_fused_qkv_matches_unfused():
+    torch.manual_seed(0)
+    unfused = GPT2LLM(cfg())
+    fused = GPT2LLM(cfg(fused_qkv=True))
+    # copy unfused weights into the fused projection
+    with torch.no_grad():
+        for bu, bf_ in zip(unfused.blocks, fused.blocks):
+            cat = torch.cat([bu.attn.q_attn.weight, bu.attn.k_attn.weight,
+                             bu.attn.v_attn.weight], dim=0)
+            bf_.attn.qkv_attn.weight.copy_(cat)
+        sd = {k: v for k, v in unfused.state_dict().items()
+              if "q_attn" not in k and "k_attn" not in k and "v_attn" not in k}
+        fused.load_state_dict(sd, strict=False)
+    unfused.eval(), fused.eval()
+    o1, o2 = run_fwd(unfused), run_fwd(fused)
+    torch.testing.assert_close(o1, o2, rtol=1e-4, atol=1e-5)
