@@ -316,7 +316,7 @@ struct AttnFwdKernel {
 };
 
 template <int D, int ABL = 0>
-__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
+__global__ __launch_bounds__(NWAVES * 64, 2) void attn_fwd_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
     unsigned short* o, float* lse, int B, int T, int Hq, int Hkv, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
@@ -343,14 +343,14 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   auto stream = at::cuda::getCurrentHIPStream();
   const size_t smem = sizeof(unsigned short) * (attn::KVBLK * D + D * attn::KVBLK);
   if (D == 128) {
-    hipLaunchKernelGGL(attn::attn_fwd_kernel<128>, grid, dim3(256), smem, stream,
+    hipLaunchKernelGGL(attn::attn_fwd_kernel<128>, grid, dim3(attn::NWAVES * 64), smem, stream,
                        (const unsigned short*)q.data_ptr(),
                        (const unsigned short*)k.data_ptr(),
                        (const unsigned short*)v.data_ptr(),
                        (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
                        B, T, Hq, Hkv, scale);
   } else {
-    hipLaunchKernelGGL(attn::attn_fwd_kernel<64>, grid, dim3(256), smem, stream,
+    hipLaunchKernelGGL(attn::attn_fwd_kernel<64>, grid, dim3(attn::NWAVES * 64), smem, stream,
                        (const unsigned short*)q.data_ptr(),
                        (const unsigned short*)k.data_ptr(),
                        (const unsigned short*)v.data_ptr(),
@@ -375,7 +375,7 @@ std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
   auto stream = at::cuda::getCurrentHIPStream();
   const size_t smem = sizeof(unsigned short) * (attn::KVBLK * D + D * attn::KVBLK);
   auto launch = [&](auto kfn) {
-    hipLaunchKernelGGL(kfn, grid, dim3(256), smem, stream,
+    hipLaunchKernelGGL(kfn, grid, dim3(attn::NWAVES * 64), smem, stream,
                        (const unsigned short*)q.data_ptr(),
                        (const unsigned short*)k.data_ptr(),
                        (const unsigned short*)v.data_ptr(),
