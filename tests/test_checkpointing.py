@@ -184,3 +184,31 @@ def test_checkpoint_saving_combiner_deletes(tmp_path):
     saving.save_checkpoint_and_free_memory(progress_at(2), app)
     folders = sorted(p.name for p in (tmp_path / "e").iterdir() if p.is_dir())
     assert len(folders) == 1 and "seen_steps_2" in folders[0]
+
+
+def _load_w2_worker(rank, world, tmp_dir):
+    """Load a world-1 checkpoint into a world-2 model (reshard up)."""
+    sharded, opt, sched = build(world, rank)
+    app = AppState(sharded, opt, sched)
+    folder = read_last_checkpoint_info(f"{tmp_dir}/exp1")
+    ShardedCheckpointLoading(rank).load_checkpoint_(app, folder)
+    sd = sharded.gather_full_state_dict()
+    return {k: v.numpy() for k, v in sd.items()}
+
+
+def test_reshard_world1_to_world2(tmp_path):
+    sharded, opt, sched = build()
+    train_steps(sharded, opt, sched, range(3))
+    app = AppState(sharded, opt, sched)
+    ShardedCheckpointSaving(tmp_path, "exp1", 0).save_checkpoint(
+        app, progress_at(3))
+    ref_sd = sharded.gather_full_state_dict()
+
+    results = run_distributed(_load_w2_worker, world_size=2,
+                              port=find_free_port(), args=(str(tmp_path),))
+    for rank in (0, 1):
+        got = results[rank]
+        assert set(got) == set(ref_sd)
+        for k in ref_sd:
+            torch.testing.assert_close(torch.from_numpy(got[k]), ref_sd[k],
+                                       rtol=1e-6, atol=1e-7)
