@@ -76,6 +76,19 @@ def main():
         f"WORLD_SIZE={world} but --gpus={args.gpus}"
 
     on_gpu = torch.cuda.is_available()
+    # Load pre-tuned hipBLASLt GEMM algorithm choices (TunableOp) for the
+    # benchmark shapes; tuned offline on MI355X (assets/tunableop_gfx950.csv).
+    tuned_file = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                              "assets", "tunableop_gfx950.csv")
+    if on_gpu and os.path.exists(tuned_file):
+        try:
+            torch.cuda.tunable.enable(True)
+            torch.cuda.tunable.tuning_enable(False)
+            torch.cuda.tunable.set_filename(tuned_file, insert_device_ordinal=False)
+            torch.cuda.tunable.read_file(tuned_file)
+        except Exception as e:
+            import sys
+            print(f"# TunableOp load skipped: {e}", file=sys.stderr)
     if world > 1:
         dist.init_process_group("nccl" if on_gpu else "gloo")
     if on_gpu:
