@@ -154,7 +154,10 @@ struct DQKernel {
                                       dofrag[s][m] = (__bf16)0.f; }
       }
     }
-    const float my_lse = (qg < T) ? lse[(((long)b * Hq) + h) * T + qg] : 0.f;
+    // log2-domain lse (exp2 fold; see attention_fwd.hip)
+    const float my_lse2 = ((qg < T) ? lse[(((long)b * Hq) + h) * T + qg] : 0.f)
+                          * 1.44269504f;
+    const float scale2 = scale * 1.44269504f;
     const float my_delta = (qg < T) ? delta[(((long)b * Hq) + h) * T + qg] : 0.f;
 
     floatx16 acc_dq[NDBLK];
@@ -236,9 +239,9 @@ struct DQKernel {
         const int kg0 = kv0 + crow(r, hi);
         const int kg1 = kv0 + 32 + crow(r, hi);
         const float p0 = (kg0 <= qg && kg0 < T)
-                             ? __expf(s0[r] * scale - my_lse) : 0.f;
+                             ? __builtin_amdgcn_exp2f(s0[r] * scale2 - my_lse2) : 0.f;
         const float p1 = (kg1 <= qg && kg1 < T)
-                             ? __expf(s1[r] * scale - my_lse) : 0.f;
+                             ? __builtin_amdgcn_exp2f(s1[r] * scale2 - my_lse2) : 0.f;
         ds[r] = scale * p0 * (dp0[r] - my_delta);
         ds[16 + r] = scale * p1 * (dp1[r] - my_delta);
       }
@@ -367,7 +370,8 @@ struct DKDVKernel {
       }
     }
     if (tid < 32) {
-      r.stat = (q0 + tid < T) ? lse_h[q0 + tid] : 0.f;
+      // staged in the log2 domain (exp2 fold) - free at staging time
+      r.stat = (q0 + tid < T) ? lse_h[q0 + tid] * 1.44269504f : 0.f;
     } else if (tid < 64) {
       r.stat = (q0 + tid - 32 < T) ? delta_h[q0 + tid - 32] : 0.f;
     }
@@ -413,6 +417,7 @@ struct DKDVKernel {
     const int rep = Hq / Hkv;
     const int kvblk0 = blockIdx.x * 128;
     const int kvg = kvblk0 + wid * 32 + ln31;  // this lane's kv row
+    const float scale2k = scale * 1.44269504f;  // exp2 fold
 
     const long kv_base = (((long)b * T) * Hkv + hkv) * D;
 
@@ -533,7 +538,7 @@ struct DKDVKernel {
           for (int r = 0; r < 16; ++r) {
             const float lq = (ABL != 3) ? cur->lse[crow(r, hi)] : 0.f;
             const float dl = (ABL != 3) ? cur->delta[crow(r, hi)] : 0.f;
-            const float pv = __expf(s_acc[r] * scale - lq);
+            const float pv = __builtin_amdgcn_exp2f(s_acc[r] * scale2k - lq);
             p_c[r] = pv;
             ds_c[r] = scale * pv * (dp_acc[r] - dl);
           }
@@ -545,7 +550,7 @@ struct DKDVKernel {
           // LDS broadcast (same address across the 32 lanes of one half)
           const float lq = (ABL != 3) ? cur->lse[crow(r, hi)] : 0.f;
           const float dl = (ABL != 3) ? cur->delta[crow(r, hi)] : 0.f;
-          const float pv = ok ? __expf(s_acc[r] * scale - lq) : 0.f;
+          const float pv = ok ? __builtin_amdgcn_exp2f(s_acc[r] * scale2k - lq) : 0.f;
           p_c[r] = pv;
           ds_c[r] = scale * pv * (dp_acc[r] - dl);
         }

@@ -102,6 +102,10 @@ struct AttnFwdKernel {
     const long q_base = (((long)b * T) * Hq + h) * D;        // + t*Hq*D
     const long kv_base = (((long)b * T) * Hkv + hkv) * D;
     const int qg = qblk0 + wid * QBLK + ln31;  // this lane's q row
+    // exp2 fold: v_exp_f32 is base-2; folding log2(e) into the QK scale and
+    // keeping the softmax state in the log2 domain removes one VALU multiply
+    // per score element. lse is converted back to natural log on store.
+    const float scale2 = scale * 1.44269504f;
 
     // ---- Q preload: lane holds Q[qg][hi*8 + m + 16*s], s=0..NDSTEP-1 ----
     bf16x8 qfrag[NDSTEP];
@@ -189,8 +193,8 @@ struct AttnFwdKernel {
       for (int r = 0; r < 16; ++r) {
         const int kg0 = kv0 + crow(r, hi);
         const int kg1 = kv0 + 32 + crow(r, hi);
-        float v0 = (kg0 <= qg && kg0 < T) ? s0[r] * scale : -INFINITY;
-        float v1 = (kg1 <= qg && kg1 < T) ? s1[r] * scale : -INFINITY;
+        float v0 = (kg0 <= qg && kg0 < T) ? s0[r] * scale2 : -INFINITY;
+        float v1 = (kg1 <= qg && kg1 < T) ? s1[r] * scale2 : -INFINITY;
         p[r] = v0;
         p[16 + r] = v1;
         tmax = fmaxf(tmax, fmaxf(v0, v1));
@@ -210,7 +214,7 @@ struct AttnFwdKernel {
         const float m_new = fmaxf(m_run, tmax);
         const float m_safe = (m_new == -INFINITY) ? 0.f : m_new;
         // -inf guards: fully-masked tiles keep O/l at 0 without NaNs
-        const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_safe);
+        const float alpha = (m_run == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m_run - m_safe);
         l_run *= alpha;
 #pragma unroll
         for (int dblk = 0; dblk < NDBLK; ++dblk)
@@ -222,7 +226,7 @@ struct AttnFwdKernel {
       float psum = 0.f;
 #pragma unroll
       for (int r = 0; r < 32; ++r) {
-        p[r] = __expf(p[r] - m_safe2);
+        p[r] = __builtin_amdgcn_exp2f(p[r] - m_safe2);
         psum += p[r];
       }
       psum += __shfl_xor(psum, 32, 64);
@@ -309,7 +313,7 @@ struct AttnFwdKernel {
       }
       if (hi == 0) {
         lse[(((long)b * Hq) + h) * T + qg] =
-            (m_run == -INFINITY) ? 0.f : m_run + __logf(l_safe);
+            (m_run == -INFINITY) ? 0.f : m_run * 0.6931471806f + __logf(l_safe);
       }
     }
   }
