@@ -1,7 +1,6 @@
 """Batch types (capability parity with reference src/modalities/batch.py)."""
 
 from dataclasses import dataclass, field
-from enum import Enum
 from typing import Optional
 
 import torch

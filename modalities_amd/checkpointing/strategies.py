@@ -3,7 +3,6 @@ reference src/modalities/checkpointing/checkpoint_saving_strategies.py:36-120).
 """
 
 from dataclasses import dataclass, field
-from typing import Optional
 
 from modalities_amd.training.progress import TrainingProgress
 

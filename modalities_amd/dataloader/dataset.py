@@ -2,7 +2,6 @@
 src/modalities/dataloader/dataset.py:76-464)."""
 
 from pathlib import Path
-from typing import Optional
 
 import numpy as np
 import torch

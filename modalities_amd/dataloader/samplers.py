@@ -2,7 +2,7 @@
 src/modalities/dataloader/samplers.py:11-137): epoch-seeded shuffle,
 skip_num_global_samples for warmstart, pad/drop, rank-strided subsampling."""
 
-from typing import Iterator, Optional
+from typing import Iterator
 
 import torch
 from torch.utils.data import Dataset, Sampler

@@ -5,7 +5,6 @@ sink is intentionally not bundled — offline environment)."""
 
 import json
 from pathlib import Path
-from typing import Optional
 
 import torch
 

@@ -18,7 +18,7 @@ from modalities_amd.config.yaml_loader import load_app_config_dict
 from modalities_amd.logging_broker.broker import (MessageBroker, MessagePublisher,
                                                   MessageTypes)
 from modalities_amd.registry.components import get_default_registry
-from modalities_amd.registry.registry import ComponentEntity, Registry
+from modalities_amd.registry.registry import ComponentEntity
 from modalities_amd.running_env import global_rank, is_dist
 from modalities_amd.training.evaluator import Evaluator
 from modalities_amd.training.gym import Gym

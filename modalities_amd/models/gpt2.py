@@ -10,14 +10,12 @@ Attention head_dim is sized for MFMA tiles (multiples of 16; 64/128 fast)."""
 
 import math
 from enum import Enum
-from functools import partial
 from typing import Annotated, Optional
 
 import torch
 import torch.nn as nn
 from pydantic import BaseModel, Field, model_validator
 
-from modalities_amd.batch import InferenceResultBatch
 from modalities_amd.models.model import NNModel, SwiGLU
 from modalities_amd.ops import flash_attention, precompute_rope_cos_sin, rope_apply
 from modalities_amd.ops.rms_norm import RMSNorm

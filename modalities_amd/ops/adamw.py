@@ -5,8 +5,7 @@ HIP kernel: csrc/adamw.hip. Replaces torch.optim.AdamW(fused=True)
 clip_grads_with_norm_ (reference: training/gradient_clipping/
 fsdp_gradient_clipper.py:144-229)."""
 
-import math
-from typing import Iterable, Optional
+from typing import Iterable
 
 import torch
 

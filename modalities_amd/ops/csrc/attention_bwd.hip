@@ -8,8 +8,11 @@
 //   dQ = dS' K  ; dK = dS'^T Q
 // All matmuls are 32x32x16 bf16 MFMAs using the same fragment maps as the
 // forward (A[i=lane&31][k=hi*8+m], B[k=hi*8+m][j=lane&31], D col=lane&31 /
-// row=crow(reg,hi)); in-register layout transposes via cvt_pk +
-// permlane32_swap (guide T12).
+// row=crow(reg,hi)). In-register C->A/B layout transposes go through
+// cvt_pk + permlane32_swap (guide T12); transposed Q/dO/K operands come
+// from [panel][row][16] LDS images via ds_read_b64_tr_b16 (guide T10);
+// both kernels software-pipeline their tile staging (dkdv: double-buffered
+// two-deep unroll; dq: T14 issue-early/write-late).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
@@ -42,17 +45,6 @@ constexpr int KVBLK = 64;  // dQ kernel kv tile
 // Swizzles (see attention_fwd.hip for the derivation):
 __device__ __forceinline__ int swz(int row, int col) {   // row-major tiles
   return col ^ (((row & 3) | (((row >> 4) & 1) << 2)) << 3);
-}
-__device__ __forceinline__ int swz_t(int d, int r) {     // transposed, 64-wide
-  return r ^ ((((d >> 3) & 3) << 1 | ((d >> 1) & 1)) << 3);
-}
-__device__ __forceinline__ int swz32(int d, int r) {     // transposed, 32-wide
-  // Read pattern (dV/dK B-operands): lanes d = dblk*32 + ln31 all read the
-  // same r-range; row stride is 64 B so bank(dword) = (d*16 + r/2) % 64 —
-  // only d&3 separates lanes without a swizzle. Injecting (d>>2)&3 into the
-  // 4 dword slots makes lanes d, d+4, d+8, d+12 land on distinct slots
-  // (<=2-way from d vs d+16 only).
-  return r ^ (((d >> 2) & 3) << 3);
 }
 __device__ __forceinline__ int crow(int reg, int hi) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * hi;

@@ -6,17 +6,20 @@
 //
 // - workgroup = 4 waves; each wave owns a 32-row Q block (WG covers 128 rows)
 // - KV tiles of 64 staged in LDS by the whole WG:
-//     K   [64][D]  row-major, XOR-swizzled (guide T2: col ^= (row&7)<<3)
-//     V^T [D][64]  transposed+swizzled, so PV reads rows (consecutive kv)
+//     K  [64][D]          row-major, XOR-swizzled (guide T2)
+//     V  [D/16][64][16]   panel image read via ds_read_b64_tr_b16 (guide
+//                         T10 hardware transpose; staged with 16B writes)
 // - SWAPPED QK^T: S^T[kv][q] = mfma(A=K, B=Q^T) puts a full softmax row in
 //   one lane pair (q = lane&31) -> row reduce is in-register + 1 shfl_xor(32)
 // - SWAPPED PV:   O^T[d][q]  = mfma(A=V^T, B=P^T): the online-softmax O
 //   rescale stays lane-local (same q = lane&31 layout as the stats)
 // - P (f32, S-layout) -> PV A-fragment (bf16) via cvt_pk + permlane32_swap
 //   (guide T12) — no LDS round-trip for P.
+// - defer-max rescale skip (guide T13, THR=8) and exp2-domain softmax
+//   state (log2e folded into the QK scale).
 //
 // D in {64, 128}; causal only. bf16 in/out, f32 softmax state; lse saved
-// for the backward.
+// for the backward (natural-log domain).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
@@ -49,13 +52,6 @@ constexpr int KVBLK = 64;    // kv rows per LDS tile
 // writes are 8-lane same-row groups with distinct 16B slots -> conflict-free.
 __device__ __forceinline__ int swz(int row, int col) {
   return col ^ (((row & 3) | (((row >> 4) & 1) << 2)) << 3);
-}
-
-// Transposed-tile swizzle (scalar writes walk d = 8j+m at fixed kv row, so
-// the key needs (d>>3) entropy; reads need d1/d4): writes ~4-way (was 16),
-// reads ~1-way.
-__device__ __forceinline__ int swz_t(int d, int r) {
-  return r ^ ((((d >> 3) & 3) << 1 | ((d >> 1) & 1)) << 3);
 }
 
 // C/D register map of v_mfma_f32_32x32x16_bf16 (guide §3):

@@ -6,11 +6,10 @@ equivalent of the reference's regex weight-decay groups (reference:
 src/modalities/optimizers/optimizer_factory.py:22-215)."""
 
 import re
-from typing import Iterable, Optional
+from typing import Optional
 
 import torch
 
-from modalities_amd.ops.adamw import fused_adamw_step
 from modalities_amd.parallel.fsdp import XGMIShardedModel
 
 

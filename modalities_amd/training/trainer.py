@@ -6,7 +6,7 @@ profiler hook, evaluation/checkpointing callbacks."""
 
 import gc
 import time
-from typing import Callable, Optional
+from typing import Callable
 
 import torch
 import torch.distributed as dist
