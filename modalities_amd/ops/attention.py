@@ -43,17 +43,19 @@ def _attention_ref(q, k, v, causal=True, q_offset=None):
 
 class _FlashAttnHip(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal):
-        o, lse = hip_ext().attn_fwd(q, k, v, causal)
+    def forward(ctx, q, k, v, causal, q_offset):
+        o, lse = hip_ext().attn_fwd(q, k, v, causal, q_offset)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
+        ctx.q_offset = q_offset
         return o
 
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = hip_ext().attn_bwd(do.contiguous(), q, k, v, o, lse, ctx.causal)
-        return dq, dk, dv, None
+        dq, dk, dv = hip_ext().attn_bwd(do.contiguous(), q, k, v, o, lse,
+                                        ctx.causal, ctx.q_offset)
+        return dq, dk, dv, None, None
 
 
 from typing import Optional
@@ -66,13 +68,15 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
     q_offset (context parallelism): global key position of q row 0; None =
     classic alignment (q block ends at the last key, i.e. offset S-T).
-    Offset attention currently runs the composed-ops path on device;
-    teaching the HIP kernels the offset is a planned optimization."""
-    if use_hip(q, k, v) and q.shape[1] == k.shape[1] and q_offset in (None, 0):
-        return _FlashAttnHip.apply(q.contiguous(), k.contiguous(), v.contiguous(), causal)
+    The HIP kernels handle Tq != Tkv and offsets natively."""
+    if use_hip(q, k, v):
+        off = q_offset
+        if off is None:
+            off = k.shape[1] - q.shape[1]  # classic alignment (q at the end)
+        return _FlashAttnHip.apply(q.contiguous(), k.contiguous(),
+                                   v.contiguous(), causal, off)
     if q.is_cuda:
-        # SDPA with an explicit offset-causal mask (keeps CP functional on
-        # GPU; slower than K1 — kernel offset support is the follow-up)
+        # SDPA fallback (MODALITIES_AMD_FORCE_EAGER debug path)
         B, T, Hq, D = q.shape
         S, Hkv = k.shape[1], k.shape[2]
         rep = Hq // Hkv

@@ -118,8 +118,9 @@ struct DQKernel {
   static __device__ void run(const unsigned short* q, const unsigned short* k,
                              const unsigned short* v, const unsigned short* dout,
                              const float* lse, const float* delta,
-                             unsigned short* dq, int B, int T, int Hq, int Hkv,
-                             float scale, char* smem_raw) {
+                             unsigned short* dq, int B, int Tq, int Tkv,
+                             int q_off, int Hq, int Hkv, float scale,
+                             char* smem_raw) {
     Smem* sm = reinterpret_cast<Smem*>(smem_raw);
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
@@ -127,17 +128,18 @@ struct DQKernel {
     const int h = blockIdx.y, b = blockIdx.z;
     const int hkv = h / (Hq / Hkv);
     const int qblk0 = blockIdx.x * 128;
-    const int qg = qblk0 + wid * 32 + ln31;
+    const int qg = qblk0 + wid * 32 + ln31;   // LOCAL q row
+    const int qgl = qg + q_off;               // global position vs keys
 
-    const long q_base = (((long)b * T) * Hq + h) * D;
-    const long kv_base = (((long)b * T) * Hkv + hkv) * D;
+    const long q_base = (((long)b * Tq) * Hq + h) * D;
+    const long kv_base = (((long)b * Tkv) * Hkv + hkv) * D;
 
     bf16x8 qfrag[NDSTEP], dofrag[NDSTEP];
     const unsigned short* qr = q + q_base + (long)qg * Hq * D;
     const unsigned short* dor = dout + q_base + (long)qg * Hq * D;
 #pragma unroll
     for (int s = 0; s < NDSTEP; ++s) {
-      if (qg < T) {
+      if (qg < Tq) {
         qfrag[s] = *reinterpret_cast<const bf16x8*>(qr + hi * 8 + 16 * s);
         dofrag[s] = *reinterpret_cast<const bf16x8*>(dor + hi * 8 + 16 * s);
       } else {
@@ -147,10 +149,11 @@ struct DQKernel {
       }
     }
     // log2-domain lse (exp2 fold; see attention_fwd.hip)
-    const float my_lse2 = ((qg < T) ? lse[(((long)b * Hq) + h) * T + qg] : 0.f)
-                          * 1.44269504f;
+    const float my_lse2 = ((qg < Tq) ? lse[(((long)b * Hq) + h) * Tq + qg]
+                                      : 0.f) * 1.44269504f;
     const float scale2 = scale * 1.44269504f;
-    const float my_delta = (qg < T) ? delta[(((long)b * Hq) + h) * T + qg] : 0.f;
+    const float my_delta = (qg < Tq)
+        ? delta[(((long)b * Hq) + h) * Tq + qg] : 0.f;
 
     floatx16 acc_dq[NDBLK];
 #pragma unroll
@@ -158,8 +161,8 @@ struct DQKernel {
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc_dq[dblk][r] = 0.f;
 
-    const int q_hi_wg = min(qblk0 + 127, T - 1);
-    const int n_tiles = q_hi_wg / KVBLK + 1;
+    const int q_hi_wg = min(qblk0 + 127, Tq - 1) + q_off;  // global kv bound
+    const int n_tiles = min(q_hi_wg, Tkv - 1) / KVBLK + 1;
 
     // T14 issue-early/write-late with ONE register set and ONE LDS buffer:
     // the next tile's global loads are issued before this tile's compute
@@ -175,7 +178,7 @@ struct DQKernel {
       for (int it = 0; it < NCH; ++it) {
         const int i = tid + it * 256;
         const int r = (i * 8) / D, c = (i * 8) % D;
-        if (kv0 + r < T) {
+        if (kv0 + r < Tkv) {
           kreg[it] = *reinterpret_cast<const shortx8*>(
               k + kv_base + (long)(kv0 + r) * Hkv * D + c);
           vreg[it] = *reinterpret_cast<const shortx8*>(
@@ -230,9 +233,9 @@ struct DQKernel {
       for (int r = 0; r < 16; ++r) {
         const int kg0 = kv0 + crow(r, hi);
         const int kg1 = kv0 + 32 + crow(r, hi);
-        const float p0 = (kg0 <= qg && kg0 < T)
+        const float p0 = (kg0 <= qgl && kg0 < Tkv)
                              ? __builtin_amdgcn_exp2f(s0[r] * scale2 - my_lse2) : 0.f;
-        const float p1 = (kg1 <= qg && kg1 < T)
+        const float p1 = (kg1 <= qgl && kg1 < Tkv)
                              ? __builtin_amdgcn_exp2f(s1[r] * scale2 - my_lse2) : 0.f;
         ds[r] = scale * p0 * (dp0[r] - my_delta);
         ds[16 + r] = scale * p1 * (dp1[r] - my_delta);
@@ -278,7 +281,7 @@ struct DQKernel {
       }
     }
 
-    if (qg < T) {
+    if (qg < Tq) {
       unsigned short* dqr = dq + q_base + (long)qg * Hq * D;
 #pragma unroll
       for (int dblk = 0; dblk < NDBLK; ++dblk)
@@ -344,14 +347,14 @@ struct DKDVKernel {
   static __device__ void stage_load(StageRegs& r, const unsigned short* q,
                                     const unsigned short* dout, long q_base,
                                     const float* lse_h, const float* delta_h,
-                                    int q0, int T, int Hq) {
+                                    int q0, int Tq, int Hq) {
     const int tid = threadIdx.x;
 #pragma unroll
     for (int it = 0; it < NCH; ++it) {
       const int i = tid + it * 256;
       const int row = (i * 8) / D, c = (i * 8) % D;
       if (row >= 32) continue;
-      if (q0 + row < T) {
+      if (q0 + row < Tq) {
         r.qq[it] = *reinterpret_cast<const shortx8*>(
             q + q_base + (long)(q0 + row) * Hq * D + c);
         r.dd[it] = *reinterpret_cast<const shortx8*>(
@@ -363,9 +366,9 @@ struct DKDVKernel {
     }
     if (tid < 32) {
       // staged in the log2 domain (exp2 fold) - free at staging time
-      r.stat = (q0 + tid < T) ? lse_h[q0 + tid] * 1.44269504f : 0.f;
+      r.stat = (q0 + tid < Tq) ? lse_h[q0 + tid] * 1.44269504f : 0.f;
     } else if (tid < 64) {
-      r.stat = (q0 + tid - 32 < T) ? delta_h[q0 + tid - 32] : 0.f;
+      r.stat = (q0 + tid - 32 < Tq) ? delta_h[q0 + tid - 32] : 0.f;
     }
   }
 
@@ -399,8 +402,8 @@ struct DKDVKernel {
                              const unsigned short* v, const unsigned short* dout,
                              const float* lse, const float* delta,
                              unsigned short* dk, unsigned short* dv,
-                             int B, int T, int Hq, int Hkv, float scale,
-                             char* smem_raw) {
+                             int B, int Tq, int Tkv, int q_off, int Hq,
+                             int Hkv, float scale, char* smem_raw) {
     Smem* sm = reinterpret_cast<Smem*>(smem_raw);
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
@@ -411,7 +414,7 @@ struct DKDVKernel {
     const int kvg = kvblk0 + wid * 32 + ln31;  // this lane's kv row
     const float scale2k = scale * 1.44269504f;  // exp2 fold
 
-    const long kv_base = (((long)b * T) * Hkv + hkv) * D;
+    const long kv_base = (((long)b * Tkv) * Hkv + hkv) * D;
 
     bf16x8 kfrag[NDSTEP], vfrag[NDSTEP];
     {
@@ -419,7 +422,7 @@ struct DKDVKernel {
       const unsigned short* vr = v + kv_base + (long)kvg * Hkv * D;
 #pragma unroll
       for (int s = 0; s < NDSTEP; ++s) {
-        if (kvg < T) {
+        if (kvg < Tkv) {
           kfrag[s] = *reinterpret_cast<const bf16x8*>(kr + hi * 8 + 16 * s);
           vfrag[s] = *reinterpret_cast<const bf16x8*>(vr + hi * 8 + 16 * s);
         } else {
@@ -440,8 +443,9 @@ struct DKDVKernel {
     // iter t computes tile t from LDS buf[t&1] while tile t+1's staged
     // registers are written to buf[(t+1)&1] and tile t+2's global loads are
     // issued (T14 issue-early/write-late; one barrier per iteration).
-    const int first_qtile = kvblk0 / 32;  // causal: q >= kv
-    const int n_qtiles = (T + 31) / 32;
+    // causal in LOCAL q coords: need q_local >= kv - q_off
+    const int first_qtile = max(0, kvblk0 - q_off) / 32;
+    const int n_qtiles = (Tq + 31) / 32;
     const int tiles_per_head = n_qtiles - first_qtile;
     const int n_iter = rep * tiles_per_head;
     const int q0_first = first_qtile * 32;
@@ -458,17 +462,17 @@ struct DKDVKernel {
     auto cursor_init = [&](Cursor& c) {
       const int hq = hkv * rep;
       c.q0 = q0_first;
-      c.q_base = (((long)b * T) * Hq + hq) * D;
-      c.lse_h = lse + (((long)b * Hq) + hq) * T;
-      c.delta_h = delta + (((long)b * Hq) + hq) * T;
+      c.q_base = (((long)b * Tq) * Hq + hq) * D;
+      c.lse_h = lse + (((long)b * Hq) + hq) * Tq;
+      c.delta_h = delta + (((long)b * Hq) + hq) * Tq;
     };
     auto cursor_next = [&](Cursor& c) {
       c.q0 += 32;
       if (c.q0 >= n_qtiles * 32) {  // next q-head of the GQA group
         c.q0 = q0_first;
         c.q_base += (long)D;
-        c.lse_h += T;
-        c.delta_h += T;
+        c.lse_h += Tq;
+        c.delta_h += Tq;
       }
     };
 
@@ -477,11 +481,11 @@ struct DKDVKernel {
     cursor_init(cc);
     cursor_init(cp);
     {
-      stage_load(rA, q, dout, cc.q_base, cc.lse_h, cc.delta_h, cc.q0, T, Hq);
+      stage_load(rA, q, dout, cc.q_base, cc.lse_h, cc.delta_h, cc.q0, Tq, Hq);
       stage_write(rA, &sm->t[0]);
       if (n_iter > 1) {
         cursor_next(cp);
-        stage_load(rA, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0, T, Hq);
+        stage_load(rA, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0, Tq, Hq);
       }
       __syncthreads();
     }
@@ -496,7 +500,7 @@ struct DKDVKernel {
         stage_write(wr, &sm->t[(t + 1) & 1]);
         if (t + 2 < n_iter) {
           cursor_next(cp);
-          stage_load(ld, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0, T, Hq);
+          stage_load(ld, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0, Tq, Hq);
         }
       }
       const int q0 = cc.q0;
@@ -523,7 +527,7 @@ struct DKDVKernel {
 
       float p_c[16], ds_c[16];
       if constexpr (ABL != 5) {
-        if (q0 >= kvblk0 + 128 && q0 + 32 <= T) {
+        if (q0 + q_off >= kvblk0 + 128 && q0 + 32 <= Tq) {
           // interior tile: every q row of this tile is > every kv row of
           // the workgroup and in-range -> no mask compares needed
 #pragma unroll
@@ -537,8 +541,8 @@ struct DKDVKernel {
         } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const int qg = q0 + crow(r, hi);
-          const bool ok = (qg < T) && (qg >= kvg) && (kvg < T);
+          const int qg = q0 + crow(r, hi);       // local
+          const bool ok = (qg < Tq) && (qg + q_off >= kvg) && (kvg < Tkv);
           // LDS broadcast (same address across the 32 lanes of one half)
           const float lq = (ABL != 3) ? cur->lse[crow(r, hi)] : 0.f;
           const float dl = (ABL != 3) ? cur->delta[crow(r, hi)] : 0.f;
@@ -616,13 +620,13 @@ struct DKDVKernel {
     // rows. Pack pairs along r&3 into uint stores per row is not possible
     // (different rows). Keep scalar stores but hoist the bounds check out
     // of the loop (interior workgroups skip all 128 compares).
-    const bool all_in = (kvblk0 + 128 <= T);
+    const bool all_in = (kvblk0 + 128 <= Tkv);
 #pragma unroll
     for (int dblk = 0; dblk < NDBLK; ++dblk)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kvrow = kvblk0 + wid * 32 + crow(r, hi);
-        if (all_in || kvrow < T) {
+        if (all_in || kvrow < Tkv) {
           const long off = kv_base + (long)kvrow * Hkv * D + dblk * 32 + ln31;
           dk[off] = f32_to_bf16(acc_dk[dblk][r]);
           dv[off] = f32_to_bf16(acc_dv[dblk][r]);
@@ -635,21 +639,22 @@ template <int D>
 __global__ __launch_bounds__(256) void dq_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
     const unsigned short* dout, const float* lse, const float* delta,
-    unsigned short* dq, int B, int T, int Hq, int Hkv, float scale) {
+    unsigned short* dq, int B, int Tq, int Tkv, int q_off, int Hq, int Hkv,
+    float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  DQKernel<D>::run(q, k, v, dout, lse, delta, dq, B, T, Hq, Hkv, scale,
-                   smem_raw);
+  DQKernel<D>::run(q, k, v, dout, lse, delta, dq, B, Tq, Tkv, q_off, Hq, Hkv,
+                   scale, smem_raw);
 }
 
 template <int D, int ABL = 0>
 __global__ __launch_bounds__(256) void dkdv_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
     const unsigned short* dout, const float* lse, const float* delta,
-    unsigned short* dk, unsigned short* dv, int B, int T, int Hq, int Hkv,
-    float scale) {
+    unsigned short* dk, unsigned short* dv, int B, int Tq, int Tkv, int q_off,
+    int Hq, int Hkv, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  DKDVKernel<D, ABL>::run(q, k, v, dout, lse, delta, dk, dv, B, T, Hq, Hkv,
-                          scale, smem_raw);
+  DKDVKernel<D, ABL>::run(q, k, v, dout, lse, delta, dk, dv, B, Tq, Tkv,
+                          q_off, Hq, Hkv, scale, smem_raw);
 }
 
 }  // namespace attnbwd
@@ -657,11 +662,12 @@ __global__ __launch_bounds__(256) void dkdv_kernel(
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    bool causal) {
+                                    bool causal, long q_offset) {
   TORCH_CHECK(causal, "attn_bwd: only causal attention is implemented");
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
   const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
-  const int Hkv = k.size(2);
+  const int Tkv = k.size(1), Hkv = k.size(2);
+  const int q_off = (int)q_offset;
   const float scale = 1.0f / sqrtf((float)D);
   auto stream = at::cuda::getCurrentHIPStream();
 
@@ -684,7 +690,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   auto dv = torch::empty_like(v);
 
   const int n_qblk = (T + 127) / 128;
-  const int n_kvblk = (T + 127) / 128;
+  const int n_kvblk = (Tkv + 127) / 128;
   if (D == 128) {
     size_t smem_dq = sizeof(typename attnbwd::DQKernel<128>::Smem);
     hipLaunchKernelGGL(attnbwd::dq_kernel<128>, dim3(n_qblk, Hq, B), dim3(256),
@@ -693,7 +699,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (const unsigned short*)v.data_ptr(),
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       (unsigned short*)dq.data_ptr(), B, T, Hq, Hkv, scale);
+                       (unsigned short*)dq.data_ptr(), B, T, Tkv, q_off, Hq, Hkv, scale);
     HIP_CHECK_KERNEL();
     size_t smem_kv = sizeof(typename attnbwd::DKDVKernel<128>::Smem);
     hipLaunchKernelGGL(attnbwd::dkdv_kernel<128>, dim3(n_kvblk, Hkv, B),
@@ -704,7 +710,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (unsigned short*)dk.data_ptr(),
-                       (unsigned short*)dv.data_ptr(), B, T, Hq, Hkv, scale);
+                       (unsigned short*)dv.data_ptr(), B, T, Tkv, q_off, Hq, Hkv, scale);
     HIP_CHECK_KERNEL();
   } else if (D == 64) {
     size_t smem_dq = sizeof(typename attnbwd::DQKernel<64>::Smem);
@@ -714,7 +720,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (const unsigned short*)v.data_ptr(),
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       (unsigned short*)dq.data_ptr(), B, T, Hq, Hkv, scale);
+                       (unsigned short*)dq.data_ptr(), B, T, Tkv, q_off, Hq, Hkv, scale);
     HIP_CHECK_KERNEL();
     size_t smem_kv = sizeof(typename attnbwd::DKDVKernel<64>::Smem);
     hipLaunchKernelGGL(attnbwd::dkdv_kernel<64>, dim3(n_kvblk, Hkv, B),
@@ -725,7 +731,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (unsigned short*)dk.data_ptr(),
-                       (unsigned short*)dv.data_ptr(), B, T, Hq, Hkv, scale);
+                       (unsigned short*)dv.data_ptr(), B, T, Tkv, q_off, Hq, Hkv, scale);
     HIP_CHECK_KERNEL();
   } else {
     TORCH_CHECK(false, "attn_bwd: head_dim must be 64 or 128");
@@ -759,7 +765,8 @@ std::vector<torch::Tensor> attn_bwd_dkdv_ablate(torch::Tensor dout,
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (unsigned short*)dk.data_ptr(),
-                       (unsigned short*)dv.data_ptr(), B, T, Hq, Hkv, scale);
+                       (unsigned short*)dv.data_ptr(), B, T, T, 0, Hq, Hkv,
+                       scale);
   };
   switch (mode) {
     case 0: launch(attnbwd::dkdv_kernel<128, 0>); break;

@@ -82,8 +82,8 @@ struct AttnFwdKernel {
                              const unsigned short* __restrict__ v,
                              unsigned short* __restrict__ o,
                              float* __restrict__ lse,
-                             int B, int T, int Hq, int Hkv, float scale,
-                             char* smem_raw) {
+                             int B, int Tq, int Tkv, int q_off, int Hq,
+                             int Hkv, float scale, char* smem_raw) {
     Smem* sm = reinterpret_cast<Smem*>(smem_raw);
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
@@ -95,9 +95,10 @@ struct AttnFwdKernel {
     const int b = blockIdx.z;
     const int hkv = h / (Hq / Hkv);
 
-    const long q_base = (((long)b * T) * Hq + h) * D;        // + t*Hq*D
-    const long kv_base = (((long)b * T) * Hkv + hkv) * D;
-    const int qg = qblk0 + wid * QBLK + ln31;  // this lane's q row
+    const long q_base = (((long)b * Tq) * Hq + h) * D;       // + t*Hq*D
+    const long kv_base = (((long)b * Tkv) * Hkv + hkv) * D;
+    const int qg = qblk0 + wid * QBLK + ln31;  // this lane's LOCAL q row
+    const int qgl = qg + q_off;                // global position vs keys
     // exp2 fold: v_exp_f32 is base-2; folding log2(e) into the QK scale and
     // keeping the softmax state in the log2 domain removes one VALU multiply
     // per score element. lse is converted back to natural log on store.
@@ -109,7 +110,7 @@ struct AttnFwdKernel {
       const unsigned short* qr = q + q_base + (long)qg * Hq * D;
 #pragma unroll
       for (int s = 0; s < NDSTEP; ++s) {
-        if (qg < T) {
+        if (qg < Tq) {
           qfrag[s] = *reinterpret_cast<const bf16x8*>(qr + hi * 8 + 16 * s);
         } else {
           for (int m = 0; m < 8; ++m) qfrag[s][m] = (__bf16)0.f;
@@ -125,8 +126,9 @@ struct AttnFwdKernel {
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc_o[dblk][r] = 0.f;
 
-    const int q_hi_wg = min(qblk0 + WG_Q - 1, T - 1);  // causal upper bound
-    const int n_tiles = (min(q_hi_wg, T - 1)) / KVBLK + 1;
+    // causal upper bound in GLOBAL key coords
+    const int q_hi_wg = min(qblk0 + WG_Q - 1, Tq - 1) + q_off;
+    const int n_tiles = (min(q_hi_wg, Tkv - 1)) / KVBLK + 1;
 
     // V goes into a panel image [D/16][KVBLK][16] consumed via hardware
     // transpose reads (vectorized writes; the v1 scalar scatter stores had
@@ -143,7 +145,7 @@ struct AttnFwdKernel {
         const int i = tid + it * NWAVES * 64;
         const int r = (i * 8) / D, c = (i * 8) % D;
         shortx8 kk, vv;
-        if (kv0 + r < T) {
+        if (kv0 + r < Tkv) {
           kk = *reinterpret_cast<const shortx8*>(
               k + kv_base + (long)(kv0 + r) * Hkv * D + c);
           vv = *reinterpret_cast<const shortx8*>(
@@ -189,8 +191,8 @@ struct AttnFwdKernel {
       for (int r = 0; r < 16; ++r) {
         const int kg0 = kv0 + crow(r, hi);
         const int kg1 = kv0 + 32 + crow(r, hi);
-        float v0 = (kg0 <= qg && kg0 < T) ? s0[r] * scale2 : -INFINITY;
-        float v1 = (kg1 <= qg && kg1 < T) ? s1[r] * scale2 : -INFINITY;
+        float v0 = (kg0 <= qgl && kg0 < Tkv) ? s0[r] * scale2 : -INFINITY;
+        float v1 = (kg1 <= qgl && kg1 < Tkv) ? s1[r] * scale2 : -INFINITY;
         p[r] = v0;
         p[16 + r] = v1;
         tmax = fmaxf(tmax, fmaxf(v0, v1));
@@ -287,7 +289,7 @@ struct AttnFwdKernel {
     // ---- epilogue: normalize, write O (scatter by q row) + lse ----------
     const float l_safe = (l_run > 0.f) ? l_run : 1.f;
     const float inv_l = 1.f / l_safe;
-    if (qg < T) {
+    if (qg < Tq) {
       unsigned short* orow = o + q_base + (long)qg * Hq * D;
       // PV output: col = ln31 = this lane's q row; rows crow(r,hi) = d.
       // crow runs in 4-consecutive groups (r&3), so pack 4 bf16 -> one
@@ -308,7 +310,7 @@ struct AttnFwdKernel {
         }
       }
       if (hi == 0) {
-        lse[(((long)b * Hq) + h) * T + qg] =
+        lse[(((long)b * Hq) + h) * Tq + qg] =
             (m_run == -INFINITY) ? 0.f : m_run * 0.6931471806f + __logf(l_safe);
       }
     }
@@ -318,27 +320,35 @@ struct AttnFwdKernel {
 template <int D, int ABL = 0>
 __global__ __launch_bounds__(NWAVES * 64, 2) void attn_fwd_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
-    unsigned short* o, float* lse, int B, int T, int Hq, int Hkv, float scale) {
+    unsigned short* o, float* lse, int B, int Tq, int Tkv, int q_off, int Hq,
+    int Hkv, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  AttnFwdKernel<D, ABL>::run(q, k, v, o, lse, B, T, Hq, Hkv, scale, smem_raw);
+  AttnFwdKernel<D, ABL>::run(q, k, v, o, lse, B, Tq, Tkv, q_off, Hq, Hkv,
+                             scale, smem_raw);
 }
 
 }  // namespace attn
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v, bool causal) {
+                                    torch::Tensor v, bool causal,
+                                    long q_offset) {
+  // q: [B, Tq, Hq, D]; k/v: [B, Tkv, Hkv, D]. q_offset = global key
+  // position of q row 0 (context parallelism); Tq == Tkv, q_offset == 0 is
+  // the classic square causal case.
   TORCH_CHECK(causal, "attn_fwd: only causal attention is implemented");
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 4
               && q.is_contiguous(), "q must be contiguous bf16 [B,T,Hq,D]");
   TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
-  const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
-  const int Hkv = k.size(2);
+  const int B = q.size(0), Tq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Tkv = k.size(1), Hkv = k.size(2);
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
   TORCH_CHECK(D == 64 || D == 128, "attn_fwd: head_dim must be 64 or 128");
+  TORCH_CHECK(q_offset >= 0 && q_offset + Tq <= Tkv + Tq,
+              "q_offset out of range");
   auto o = torch::empty_like(q);
-  auto lse = torch::empty({B, Hq, T}, q.options().dtype(torch::kFloat32));
+  auto lse = torch::empty({B, Hq, Tq}, q.options().dtype(torch::kFloat32));
   const float scale = 1.0f / sqrtf((float)D);
-  const int n_qblk = (T + attn::WG_Q - 1) / attn::WG_Q;
+  const int n_qblk = (Tq + attn::WG_Q - 1) / attn::WG_Q;
   dim3 grid(n_qblk, Hq, B);
   auto stream = at::cuda::getCurrentHIPStream();
   const size_t smem = sizeof(unsigned short) * (attn::KVBLK * D + D * attn::KVBLK);
@@ -348,14 +358,14 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                        (const unsigned short*)k.data_ptr(),
                        (const unsigned short*)v.data_ptr(),
                        (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
-                       B, T, Hq, Hkv, scale);
+                       B, Tq, Tkv, (int)q_offset, Hq, Hkv, scale);
   } else {
     hipLaunchKernelGGL(attn::attn_fwd_kernel<64>, grid, dim3(attn::NWAVES * 64), smem, stream,
                        (const unsigned short*)q.data_ptr(),
                        (const unsigned short*)k.data_ptr(),
                        (const unsigned short*)v.data_ptr(),
                        (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
-                       B, T, Hq, Hkv, scale);
+                       B, Tq, Tkv, (int)q_offset, Hq, Hkv, scale);
   }
   HIP_CHECK_KERNEL();
   return {o, lse};
@@ -380,7 +390,7 @@ std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
                        (const unsigned short*)k.data_ptr(),
                        (const unsigned short*)v.data_ptr(),
                        (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
-                       B, T, Hq, Hkv, scale);
+                       B, T, T, 0, Hq, Hkv, scale);
   };
   switch (mode) {
     case 0: launch(attn::attn_fwd_kernel<128, 0>); break;

@@ -31,11 +31,12 @@ void fused_adamw(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
 torch::Tensor multi_tensor_sqsum(std::vector<torch::Tensor> tensors);
 void multi_tensor_scale(std::vector<torch::Tensor> tensors, torch::Tensor scale);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v, bool causal);
+                                    torch::Tensor v, bool causal,
+                                    long q_offset);
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    bool causal);
+                                    bool causal, long q_offset);
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
 torch::Tensor tr16_probe(long mode);
 std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
@@ -62,8 +63,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "AdamW with device-resident step counter (hipGraph-safe)");
   m.def("multi_tensor_sqsum", &multi_tensor_sqsum, "sum of squares (K10)");
   m.def("multi_tensor_scale", &multi_tensor_scale, "in-place scale (K10)");
-  m.def("attn_fwd", &attn_fwd, "flash attention forward (K1)");
-  m.def("attn_bwd", &attn_bwd, "flash attention backward (K1)");
+  m.def("attn_fwd", &attn_fwd, "flash attention forward (K1)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"),
+        py::arg("q_offset") = 0);
+  m.def("attn_bwd", &attn_bwd, "flash attention backward (K1)",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("o"), py::arg("lse"), py::arg("causal"),
+        py::arg("q_offset") = 0);
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16,
         "MFMA fragment-layout probe (verification)");

@@ -9,7 +9,8 @@ one all-gather uses all 7 links in parallel; KV bytes per layer at seq 8192 /
 GQA-20 heads are ~84 MB — transfer hides behind the block's attention
 compute), and each rank computes its Q chunk against the full K/V with an
 OFFSET-causal mask (rank r's queries sit at global positions
-[r*T_local, (r+1)*T_local)).
+[r*T_local, (r+1)*T_local)) — on device this runs the K1 HIP flash kernels,
+which support Tq != Tkv and the query offset natively.
 
 Memory: block activations, attention state and logits stay 1/cp of the full
 sequence; only the transient K/V gather and the embedding output are full-T.
