@@ -99,3 +99,55 @@ def test_cp2_matches_full_sequence():
     wte_sum = torch.from_numpy(results[0][3]) + torch.from_numpy(results[1][3])
     torch.testing.assert_close(wte_sum, torch.from_numpy(ref_grads["wte.weight"]),
                                rtol=1e-4, atol=1e-5)
+
+
+def _cp_ulysses_worker(rank, world):
+    import torch.distributed as dist
+
+    from modalities_amd.parallel.cp import (get_gpt2_context_parallel_model,
+                                            slice_targets_for_cp)
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    model = get_gpt2_context_parallel_model(model, group=dist.group.WORLD,
+                                            cp_rank=rank, cp_size=world,
+                                            variant="ulysses")
+    x, y = make_batch()
+    out = model({"input_ids": x})["logits"]
+    return out.detach().numpy()
+
+
+def test_cp2_ulysses_matches_full_sequence():
+    ref_out, _, _ = reference_run()
+    results = run_distributed(_cp_ulysses_worker, world_size=2,
+                              port=find_free_port())
+    outs = [torch.from_numpy(results[r]) for r in range(2)]
+    torch.testing.assert_close(torch.cat(outs, dim=1), ref_out,
+                               rtol=1e-4, atol=1e-4)
+
+
+def _ulysses_unit_worker(rank, world):
+    import torch.distributed as dist
+
+    from modalities_amd.ops.attention import _attention_ref
+    from modalities_amd.parallel.cp import cp_attention_ulysses
+    torch.manual_seed(1)
+    B, Tn, Hq, Hkv, D = 2, 32, 4, 2, 16
+    q = torch.randn(B, Tn, Hq, D)
+    k = torch.randn(B, Tn, Hkv, D)
+    v = torch.randn(B, Tn, Hkv, D)
+    tl = Tn // world
+    out = cp_attention_ulysses(q[:, rank * tl:(rank + 1) * tl],
+                               k[:, rank * tl:(rank + 1) * tl],
+                               v[:, rank * tl:(rank + 1) * tl],
+                               dist.group.WORLD, rank, world)
+    full = _attention_ref(q, k, v, causal=True)
+    torch.testing.assert_close(out, full[:, rank * tl:(rank + 1) * tl],
+                               rtol=1e-4, atol=1e-5)
+    return True
+
+
+def test_ulysses_attention_unit():
+    """cp_attention_ulysses == full attention, world 2 (head scatter)."""
+    results = run_distributed(_ulysses_unit_worker, world_size=2,
+                              port=find_free_port())
+    assert all(results.values())
