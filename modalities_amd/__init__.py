@@ -17,3 +17,6 @@ re-designed for AMD MI355X (gfx950, CDNA4):
 """
 
 __version__ = "0.1.0"
+
+def version() -> str:
+    return __version__

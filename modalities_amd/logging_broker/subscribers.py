@@ -5,6 +5,7 @@ sink is intentionally not bundled — offline environment)."""
 
 import json
 from pathlib import Path
+from typing import Optional
 
 import torch
 
@@ -50,6 +51,56 @@ class RichResultSubscriber(MessageSubscriberIF):
     def consume_dict(self, message_dict: dict) -> None:
         if self.enabled:
             print(json.dumps(message_dict, default=str), flush=True)
+
+
+class RichProgressSubscriber(MessageSubscriberIF):
+    """Live progress bars per dataloader tag (rank 0), built on `rich`
+    (reference: logging_broker/subscriber_impl/progress_subscriber.py:22-120).
+    Falls back to a no-op when rich is unavailable or stdout is not a TTY
+    environment worth animating."""
+
+    def __init__(self, num_ranks: int = 1, global_rank: int = 0,
+                 train_split_lengths: Optional[dict] = None,
+                 eval_split_lengths: Optional[dict] = None):
+        self.enabled = global_rank == 0
+        self._tasks: dict = {}
+        self._progress = None
+        if not self.enabled:
+            return
+        try:
+            from rich.progress import (BarColumn, MofNCompleteColumn, Progress,
+                                       TaskProgressColumn, TextColumn,
+                                       TimeElapsedColumn)
+            self._progress = Progress(
+                TextColumn("[progress.description]{task.description}"),
+                BarColumn(), MofNCompleteColumn(), TaskProgressColumn(),
+                TimeElapsedColumn(), refresh_per_second=2)
+            self._progress.start()
+            for tag, total in (train_split_lengths or {}).items():
+                self._tasks[tag] = self._progress.add_task(tag, total=total)
+            for tag, total in (eval_split_lengths or {}).items():
+                self._tasks[tag] = self._progress.add_task(tag, total=total)
+        except Exception:
+            self._progress = None
+
+    def consume_message(self, message: Message) -> None:
+        if self._progress is None:
+            return
+        payload = message.payload
+        tag = getattr(payload, "dataloader_tag", None)
+        step = getattr(payload, "num_steps_done", None)
+        if tag is None or step is None:
+            return
+        if tag not in self._tasks:
+            self._tasks[tag] = self._progress.add_task(tag, total=None)
+        self._progress.update(self._tasks[tag], completed=step)
+
+    def consume_dict(self, message_dict: dict) -> None:
+        pass
+
+    def close(self):
+        if self._progress is not None:
+            self._progress.stop()
 
 
 class ResultsToDiscSubscriber(MessageSubscriberIF):

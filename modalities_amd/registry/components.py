@@ -27,6 +27,7 @@ from modalities_amd.loss_functions import CLMCrossEntropyLoss, NCELoss
 from modalities_amd.logging_broker.subscribers import (DummyProgressSubscriber,
                                                        DummyResultSubscriber,
                                                        ResultsToDiscSubscriber,
+                                                       RichProgressSubscriber,
                                                        RichResultSubscriber)
 from modalities_amd.models.coca import CoCa
 from modalities_amd.models.gpt2 import GPT2LLM, GPT2LLMConfig
@@ -211,7 +212,7 @@ COMPONENTS: list[ComponentEntity] = [
     ComponentEntity("debugging_enriched_model", "default",
                     get_debugging_enriched_model, None),
     ComponentEntity("progress_subscriber", "dummy", DummyProgressSubscriber, None),
-    ComponentEntity("progress_subscriber", "rich", DummyProgressSubscriber, None),
+    ComponentEntity("progress_subscriber", "rich", RichProgressSubscriber, None),
     ComponentEntity("results_subscriber", "dummy", DummyResultSubscriber, None),
     ComponentEntity("results_subscriber", "rich", RichResultSubscriber, None),
     ComponentEntity("results_subscriber", "save_to_disc", ResultsToDiscSubscriber, None),
