@@ -689,6 +689,20 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
 
+  // dq and dkdv are independent given delta: run them on two streams so
+  // their partially-filled tail waves overlap (each launches ~1280 blocks
+  // over 256 CUs). Raw fork/join events (created once, reused; no timing)
+  // keep this hipGraph-capturable.
+  static hipStream_t side = nullptr;
+  static hipEvent_t ev_fork = nullptr, ev_join = nullptr;
+  if (side == nullptr) {
+    hipStreamCreateWithFlags(&side, hipStreamNonBlocking);
+    hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming);
+    hipEventCreateWithFlags(&ev_join, hipEventDisableTiming);
+  }
+  hipEventRecord(ev_fork, stream);
+  hipStreamWaitEvent(side, ev_fork, 0);
+
   const int n_qblk = (T + 127) / 128;
   const int n_kvblk = (Tkv + 127) / 128;
   if (D == 128) {
@@ -703,7 +717,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
     HIP_CHECK_KERNEL();
     size_t smem_kv = sizeof(typename attnbwd::DKDVKernel<128>::Smem);
     hipLaunchKernelGGL(attnbwd::dkdv_kernel<128>, dim3(n_kvblk, Hkv, B),
-                       dim3(256), smem_kv, stream,
+                       dim3(256), smem_kv, side,
                        (const unsigned short*)q.data_ptr(),
                        (const unsigned short*)k.data_ptr(),
                        (const unsigned short*)v.data_ptr(),
@@ -724,7 +738,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
     HIP_CHECK_KERNEL();
     size_t smem_kv = sizeof(typename attnbwd::DKDVKernel<64>::Smem);
     hipLaunchKernelGGL(attnbwd::dkdv_kernel<64>, dim3(n_kvblk, Hkv, B),
-                       dim3(256), smem_kv, stream,
+                       dim3(256), smem_kv, side,
                        (const unsigned short*)q.data_ptr(),
                        (const unsigned short*)k.data_ptr(),
                        (const unsigned short*)v.data_ptr(),
@@ -736,6 +750,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   } else {
     TORCH_CHECK(false, "attn_bwd: head_dim must be 64 or 128");
   }
+  hipEventRecord(ev_join, side);
+  hipStreamWaitEvent(stream, ev_join, 0);
   return {dq, dk, dv};
 }
 
