@@ -41,12 +41,15 @@ __global__ void adamw_masked_kernel(float* __restrict__ p,
 // Graph-safe variant: the Adam step count lives in DEVICE memory so a
 // hipGraph replay of the training step sees the advancing bias correction
 // (a host-side bc baked into kernel args would be frozen at capture).
+// Also writes the bf16 working copy (bf16_out) in the same pass — fusing
+// the publish cast saves a full fp32 re-read + bf16 write of the params.
 __global__ void adamw_masked_devstep_kernel(float* __restrict__ p,
                                             const float* __restrict__ g,
                                             float* __restrict__ m,
                                             float* __restrict__ v,
                                             const float* __restrict__ wd_mask,
                                             const int* __restrict__ step,
+                                            unsigned short* __restrict__ bf16_out,
                                             long n4, float lr, float beta1,
                                             float beta2, float eps, float wd) {
   const float t = (float)*step;
@@ -62,14 +65,17 @@ __global__ void adamw_masked_devstep_kernel(float* __restrict__ p,
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
        i += (long)gridDim.x * blockDim.x) {
     floatx4 pp = p4[i], gg = g4[i], mm = m4[i], vv = v4[i], ww = w4[i];
+    shortx4 h;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       pp[j] *= (1.0f - lr * wd * ww[j]);
       mm[j] = beta1 * mm[j] + (1.0f - beta1) * gg[j];
       vv[j] = beta2 * vv[j] + (1.0f - beta2) * gg[j] * gg[j];
       pp[j] -= step_size * mm[j] / (sqrtf(vv[j] * inv_bc2) + eps);
+      h[j] = (short)f32_to_bf16(pp[j]);
     }
     p4[i] = pp; m4[i] = mm; v4[i] = vv;
+    reinterpret_cast<shortx4*>(bf16_out)[i] = h;
   }
 }
 
@@ -127,19 +133,23 @@ void fused_adamw_masked(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 void fused_adamw_masked_devstep(torch::Tensor p, torch::Tensor g,
                                 torch::Tensor m, torch::Tensor v,
                                 torch::Tensor wd_mask, torch::Tensor step,
-                                double lr, double beta1, double beta2,
-                                double eps, double wd) {
+                                torch::Tensor bf16_out, double lr,
+                                double beta1, double beta2, double eps,
+                                double wd) {
   TORCH_CHECK(p.is_cuda() && p.dtype() == torch::kFloat32 && p.is_contiguous());
   TORCH_CHECK(p.numel() % 4 == 0, "flat shard must be divisible by 4");
   TORCH_CHECK(step.is_cuda() && step.dtype() == torch::kInt32);
+  TORCH_CHECK(bf16_out.dtype() == torch::kBFloat16
+              && bf16_out.numel() == p.numel());
   long n4 = p.numel() / 4;
   auto stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(adamw_masked_devstep_kernel, dim3(grid_for(n4, 256)),
                      dim3(256), 0, stream, p.data_ptr<float>(),
                      g.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), wd_mask.data_ptr<float>(),
-                     step.data_ptr<int>(), n4, (float)lr, (float)beta1,
-                     (float)beta2, (float)eps, (float)wd);
+                     step.data_ptr<int>(),
+                     (unsigned short*)bf16_out.data_ptr(), n4, (float)lr,
+                     (float)beta1, (float)beta2, (float)eps, (float)wd);
   HIP_CHECK_KERNEL();
 }
 

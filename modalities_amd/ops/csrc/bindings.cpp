@@ -22,8 +22,9 @@ void fused_adamw_masked(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 void fused_adamw_masked_devstep(torch::Tensor p, torch::Tensor g,
                                 torch::Tensor m, torch::Tensor v,
                                 torch::Tensor wd_mask, torch::Tensor step,
-                                double lr, double beta1, double beta2,
-                                double eps, double wd);
+                                torch::Tensor bf16_out, double lr,
+                                double beta1, double beta2, double eps,
+                                double wd);
 void fused_adamw(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                  std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
                  double lr, double beta1, double beta2, double eps, double wd,

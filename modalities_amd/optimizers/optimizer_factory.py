@@ -65,10 +65,19 @@ class ShardedAdamW(torch.optim.Optimizer):
             m, v = st["exp_avg"], st["exp_avg_sq"]
             g = u.grad_shard
             p = u.master_shard
-            if on_gpu:
+            if on_gpu and u.bf16_shard.dtype == torch.bfloat16:
+                # fused publish: the kernel writes the bf16 working shard in
+                # the same pass (saves a full param re-read + cast)
                 hip_ext().fused_adamw_masked_devstep(
                     p, g, m, v, u.wd_mask_shard, self._step_dev,
-                    lr, beta1, beta2, eps, wd)
+                    u.bf16_shard, lr, beta1, beta2, eps, wd)
+            elif on_gpu:  # fp32 working copy on GPU: unfused publish below
+                bc1 = 1.0 - beta1 ** step
+                bc2 = 1.0 - beta2 ** step
+                hip_ext().fused_adamw_masked(p, g, m, v, u.wd_mask_shard,
+                                             lr, beta1, beta2, eps, wd,
+                                             bc1, bc2)
+                u.publish_master()
             else:
                 bc1 = 1.0 - beta1 ** step
                 bc2 = 1.0 - beta2 ** step
@@ -77,7 +86,10 @@ class ShardedAdamW(torch.optim.Optimizer):
                 v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
                 denom = (v / bc2).sqrt().add_(eps)
                 p.add_(-lr / bc1 * m / denom)
-        self.sharded_model.publish_master()
+        if on_gpu:
+            self.sharded_model.free_stale_fulls()
+        else:
+            self.sharded_model.publish_master()
         return loss
 
     def zero_grad(self, set_to_none: bool = True):

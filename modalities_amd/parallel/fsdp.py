@@ -455,6 +455,16 @@ class XGMIShardedModel(nn.Module):
             u.publish_master()
 
     @torch.no_grad()
+    def free_stale_fulls(self):
+        """Post-optimizer bookkeeping when the optimizer kernel already
+        wrote the bf16 shards itself (fused publish): only drop the stale
+        gathered buffers (world 1 aliases the shard — nothing to do)."""
+        if self.world > 1:
+            for u in self.units:
+                if u.full_buf is not None:
+                    u.free_full()
+
+    @torch.no_grad()
     def clip_grad_norm_(self, max_norm: Optional[float],
                         norm_type: float = 2.0) -> torch.Tensor:
         from modalities_amd.ops.adamw import multi_tensor_l2norm, multi_tensor_scale_
