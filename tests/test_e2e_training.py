@@ -59,6 +59,9 @@ def test_e2e_training_run(e2e_config, tmp_path):
     train_records = [r for r in records if r.get("dataloader_tag") == "train"]
     assert train_records, records
     assert "CLMCrossEntropyLoss average" in train_records[-1]["losses"]
+    # the evaluator ran on the eval dataloader at the configured interval
+    val_records = [r for r in records if r.get("dataloader_tag") == "val"]
+    assert val_records and "CLMCrossEntropyLoss" in val_records[-1]["losses"]
 
     # resolved config copied into the experiment folder
     assert (ckpt_root / "config.yaml.resolved").exists()
