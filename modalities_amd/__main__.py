@@ -112,6 +112,21 @@ def CMD_entry_point_convert_pytorch_to_hf_checkpoint(
                                          output_hf_checkpoint_dir, prediction_key)
 
 
+@main.command(name="convert_checkpoint_to_full")
+@click.option("--checkpoint_folder_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+@click.option("--config_file_path", type=click.Path(exists=True, path_type=Path),
+              required=True, help="config containing the `model` component")
+@click.option("--output_path", type=click.Path(path_type=Path), required=True)
+def CMD_convert_checkpoint_to_full(checkpoint_folder_path, config_file_path,
+                                   output_path):
+    """Reassemble a sharded training checkpoint into one full state-dict
+    file (for text generation / HF export)."""
+    api.convert_sharded_checkpoint_to_full(checkpoint_folder_path,
+                                           config_file_path, output_path)
+    click.echo(f"wrote {output_path}")
+
+
 # ---- data subcommands -------------------------------------------------------
 
 @main.group(name="data")

@@ -188,3 +188,23 @@ def convert_pytorch_to_hf_checkpoint(config_path: Path, output_hf_checkpoint_dir
     from modalities_amd.conversion.convert_gpt2 import convert_gpt2_to_hf
     return convert_gpt2_to_hf(Path(config_path), Path(output_hf_checkpoint_dir),
                               prediction_key)
+
+
+def convert_sharded_checkpoint_to_full(checkpoint_folder_path: Path,
+                                       config_path: Path,
+                                       output_path: Path) -> None:
+    """Reassemble a sharded training checkpoint into a single full fp32
+    state-dict file loadable by the `model`/`checkpointed` component."""
+    import torch
+
+    from modalities_amd.checkpointing.loading import \
+        load_full_model_state_from_checkpoint
+    from modalities_amd.config.component_factory import ComponentFactory
+    from modalities_amd.config.yaml_loader import load_app_config_dict
+    from modalities_amd.registry.components import get_default_registry
+
+    config_dict = load_app_config_dict(Path(config_path))
+    factory = ComponentFactory(get_default_registry())
+    model = factory.build_component_by_key(config_dict, "model")
+    load_full_model_state_from_checkpoint(Path(checkpoint_folder_path), model)
+    torch.save(model.state_dict(), Path(output_path))

@@ -57,7 +57,7 @@ class AppState:
             u.master_shard.copy_(shards[f"model.{u.name}.master_shard"])
             st = self.optimizer.state[u.master_shard]
             key = f"optim.{u.name}.exp_avg"
-            if key in shards:
+            if key in shards and "exp_avg" in st:
                 st["exp_avg"].copy_(shards[key])
                 st["exp_avg_sq"].copy_(shards[f"optim.{u.name}.exp_avg_sq"])
             st["step"] = optim_steps.get(u.name, 0)

@@ -103,3 +103,43 @@ class ShardedCheckpointLoading:
             if max(s_lo, n_lo) < min(s_hi, n_hi):
                 return True
         return False
+
+
+@torch.no_grad()
+def load_full_model_state_from_checkpoint(folder: Path, model) -> None:
+    """Single-process path: reassemble the flat fp32 master weights of a
+    sharded checkpoint and load them into a PLAIN (unsharded) module in
+    place. Used by text generation / HF export (reference analog: the
+    "torch" checkpoint_loading variant).
+
+    The flat layout is reconstructed exactly as XGMIShardedModel built it
+    (same unit grouping): we wrap a throwaway sharded view of `model` at
+    world size 1 to recover offsets, then fill its master shards from the
+    saved rank files and publish back into the module parameters."""
+    from modalities_amd.checkpointing.app_state import AppState
+    from modalities_amd.parallel.fsdp import XGMIShardedModel
+
+    folder = Path(folder)
+    meta = read_checkpoint_meta(folder)
+    device = next(model.parameters()).device
+    sharded = XGMIShardedModel.from_transformer(
+        model, device, rank=0, world_size=1,
+        param_dtype=next(model.parameters()).dtype
+        if next(model.parameters()).dtype in (torch.float32, torch.bfloat16)
+        else torch.float32)
+    app = AppState(sharded, _NullOptimizer(sharded))
+    ShardedCheckpointLoading(0).load_checkpoint_(app, folder)
+    # materialize full params back onto the module
+    sd = sharded.gather_full_state_dict()
+    for name, p in model.named_parameters():
+        p.data = sd[name].to(p.dtype)
+    return None
+
+
+class _NullOptimizer:
+    """Minimal stand-in so AppState can load model shards without a real
+    optimizer (moments in the checkpoint are skipped)."""
+
+    def __init__(self, sharded):
+        self.state = {u.master_shard: {} for u in sharded.units}
+        self.param_groups = []

@@ -212,3 +212,23 @@ def test_reshard_world1_to_world2(tmp_path):
         for k in ref_sd:
             torch.testing.assert_close(torch.from_numpy(got[k]), ref_sd[k],
                                        rtol=1e-6, atol=1e-7)
+
+
+def test_full_state_reassembly_single_process(tmp_path):
+    """Sharded checkpoint -> plain module weights (the inference/export
+    path)."""
+    from modalities_amd.checkpointing.loading import \
+        load_full_model_state_from_checkpoint
+    sharded, opt, sched = build()
+    train_steps(sharded, opt, sched, range(2))
+    app = AppState(sharded, opt, sched)
+    ShardedCheckpointSaving(tmp_path, "expf", 0).save_checkpoint(
+        app, progress_at(2))
+    ref_sd = sharded.gather_full_state_dict()
+
+    torch.manual_seed(123)  # different init; must be overwritten
+    fresh = GPT2LLM(tiny_cfg())
+    folder = read_last_checkpoint_info(tmp_path / "expf")
+    load_full_model_state_from_checkpoint(folder, fresh)
+    for name, p in fresh.named_parameters():
+        torch.testing.assert_close(p.data, ref_sd[name], rtol=1e-6, atol=1e-7)
