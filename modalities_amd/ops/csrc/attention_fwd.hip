@@ -187,6 +187,20 @@ struct AttnFwdKernel {
         l_run += 1.f;
       } else {
       float tmax = -INFINITY;
+      // interior fast path: this wave's lowest q row covers the whole kv
+      // tile and the tile is in range -> no causal/range compares (the
+      // kernel is VALU-issue-bound; dropping 64 cmp+cndmask per tile)
+      const int q_lo_wave = qblk0 + wid * QBLK + q_off;
+      if (kv0 + KVBLK - 1 <= q_lo_wave && kv0 + KVBLK <= Tkv && qg < Tq) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float v0 = s0[r] * scale2;
+          float v1 = s1[r] * scale2;
+          p[r] = v0;
+          p[16 + r] = v1;
+          tmax = fmaxf(tmax, fmaxf(v0, v1));
+        }
+      } else {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kg0 = kv0 + crow(r, hi);
@@ -196,6 +210,7 @@ struct AttnFwdKernel {
         p[r] = v0;
         p[16 + r] = v1;
         tmax = fmaxf(tmax, fmaxf(v0, v1));
+      }
       }
       tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));  // other kv half, same q
 
