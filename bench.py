@@ -168,7 +168,11 @@ def main():
 
     one_step = eager_step
     graph_mode = False
-    if on_gpu and not args.no_graph:
+    # Graph capture only at world 1: RCCL-collective capture support is not
+    # risked in the multi-GPU scaling run (a capture hang there would cost
+    # the whole run; the measured replay gain is ~1%). Comm/compute overlap
+    # at N>1 comes from the engine's dedicated HIP streams either way.
+    if on_gpu and world == 1 and not args.no_graph:
         # hipGraph-capture the whole training step (launch-bound gaps were
         # ~10% of step time): static input buffers, device-side Adam step
         # counter so replays advance bias correction. Fresh random data is

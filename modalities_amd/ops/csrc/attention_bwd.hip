@@ -229,6 +229,19 @@ struct DQKernel {
 
       // dS'^T = scale * P .* (dP - delta), P = exp(scale*S' - lse)
       float ds[32];
+      // interior fast path: the whole kv tile is strictly below this
+      // wave's lowest (global) q row -> skip the causal/range compares
+      // (VALU-issue-bound, like the forward)
+      const int q_lo_wave = qblk0 + wid * 32 + q_off;
+      if (kv0 + KVBLK - 1 <= q_lo_wave && kv0 + KVBLK <= Tkv && qg < Tq) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float p0 = __builtin_amdgcn_exp2f(s0[r] * scale2 - my_lse2);
+          const float p1 = __builtin_amdgcn_exp2f(s1[r] * scale2 - my_lse2);
+          ds[r] = scale * p0 * (dp0[r] - my_delta);
+          ds[16 + r] = scale * p1 * (dp1[r] - my_delta);
+        }
+      } else {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kg0 = kv0 + crow(r, hi);
@@ -239,6 +252,7 @@ struct DQKernel {
                              ? __builtin_amdgcn_exp2f(s1[r] * scale2 - my_lse2) : 0.f;
         ds[r] = scale * p0 * (dp0[r] - my_delta);
         ds[16 + r] = scale * p1 * (dp1[r] - my_delta);
+      }
       }
       bf16x8 dsfrag[KVBLK / 16];
       c_layout_to_frags(ds, dsfrag, KVBLK / 16);
