@@ -18,6 +18,7 @@ from pydantic import BaseModel, Field, model_validator
 
 from modalities_amd.models.model import NNModel, SwiGLU
 from modalities_amd.ops import flash_attention, precompute_rope_cos_sin, rope_apply
+from modalities_amd.ops.linear import TwoStreamLinear
 from modalities_amd.ops.rms_norm import RMSNorm
 
 
@@ -114,12 +115,12 @@ class CausalSelfAttention(nn.Module):
         self.fused_qkv = fused_qkv
         kv_dim = self.head_dim * n_head_kv
         if fused_qkv:
-            self.qkv_attn = nn.Linear(n_embd, n_embd + 2 * kv_dim, bias=bias)
+            self.qkv_attn = TwoStreamLinear(n_embd, n_embd + 2 * kv_dim, bias=bias)
         else:
-            self.q_attn = nn.Linear(n_embd, n_embd, bias=bias)
-            self.k_attn = nn.Linear(n_embd, kv_dim, bias=bias)
-            self.v_attn = nn.Linear(n_embd, kv_dim, bias=bias)
-        self.c_proj = nn.Linear(n_embd, n_embd, bias=bias)
+            self.q_attn = TwoStreamLinear(n_embd, n_embd, bias=bias)
+            self.k_attn = TwoStreamLinear(n_embd, kv_dim, bias=bias)
+            self.v_attn = TwoStreamLinear(n_embd, kv_dim, bias=bias)
+        self.c_proj = TwoStreamLinear(n_embd, n_embd, bias=bias)
         self.resid_dropout = nn.Dropout(dropout)
         self.dropout = dropout
         if use_qk_norm:
@@ -176,9 +177,9 @@ class CausalSelfAttention(nn.Module):
 class TransformerMLP(nn.Module):
     def __init__(self, n_embd: int, ffn_hidden: int, bias: bool, dropout: float):
         super().__init__()
-        self.c_fc = nn.Linear(n_embd, ffn_hidden, bias=bias)
+        self.c_fc = TwoStreamLinear(n_embd, ffn_hidden, bias=bias)
         self.gelu = nn.GELU(approximate="tanh")
-        self.c_proj = nn.Linear(ffn_hidden, n_embd, bias=bias)
+        self.c_proj = TwoStreamLinear(ffn_hidden, n_embd, bias=bias)
         self.dropout = nn.Dropout(dropout)
 
     def forward(self, x):
@@ -231,7 +232,7 @@ class GPT2LLM(NNModel):
         self.drop = nn.Dropout(cfg.dropout)
         self.blocks = nn.ModuleList(GPT2Block(cfg) for _ in range(cfg.n_layer))
         self.lm_head_norm = make_norm(cfg.lm_head_norm_config, cfg.n_embd)
-        self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)
+        self.lm_head = TwoStreamLinear(cfg.n_embd, cfg.vocab_size, bias=False)
         if cfg.use_weight_tying:
             self.lm_head.weight = self.wte.weight
 
