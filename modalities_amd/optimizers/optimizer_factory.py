@@ -56,7 +56,7 @@ class ShardedAdamW(torch.optim.Optimizer):
                     start, dtype=torch.int32,
                     device=self.sharded_model.units[0].master_shard.device)
             self._step_dev += 1  # on-device: advances under hipGraph replay
-        for u in self.sharded_model.units:
+        for ui, u in enumerate(self.sharded_model.units):
             if not u.grad_fresh:
                 continue  # no grads reduced for this unit this step
             st = self.state[u.master_shard]
@@ -71,6 +71,9 @@ class ShardedAdamW(torch.optim.Optimizer):
                 hip_ext().fused_adamw_masked_devstep(
                     p, g, m, v, u.wd_mask_shard, self._step_dev,
                     u.bf16_shard, lr, beta1, beta2, eps, wd)
+                if u.full_buf is not None and self.sharded_model.world > 1:
+                    u.free_full()  # stale gathered params
+                self.sharded_model.prefetch_unit_gather(ui)
             elif on_gpu:  # fp32 working copy on GPU: unfused publish below
                 bc1 = 1.0 - beta1 ** step
                 bc2 = 1.0 - beta2 ** step
@@ -87,9 +90,16 @@ class ShardedAdamW(torch.optim.Optimizer):
                 denom = (v / bc2).sqrt().add_(eps)
                 p.add_(-lr / bc1 * m / denom)
         if on_gpu:
-            self.sharded_model.free_stale_fulls()
+            # stale gathered buffers were already freed per unit (before the
+            # next-step prefetch gathers, which must NOT be freed here)
+            pass
         else:
             self.sharded_model.publish_master()
+            # same next-step gather prefetch as the GPU path (synchronous on
+            # CPU/gloo, which lets the world-2 tests exercise the identical
+            # collective ordering the RCCL path uses)
+            for ui in range(len(self.sharded_model.units)):
+                self.sharded_model.prefetch_unit_gather(ui)
         return loss
 
     def zero_grad(self, set_to_none: bool = True):

@@ -455,6 +455,19 @@ class XGMIShardedModel(nn.Module):
             u.publish_master()
 
     @torch.no_grad()
+    def prefetch_unit_gather(self, idx: int):
+        """Kick off unit idx's next-step all-gather on the comm stream (call
+        right after the optimizer updated that unit's bf16 shard: the first
+        unit's gather then hides behind the remaining units' optimizer
+        work instead of stalling the next forward). No-op when resharding
+        after forward (the gather there must stay just-in-time)."""
+        if self.reshard_after_forward or self.world == 1:
+            return
+        u = self.units[idx]
+        if not u.is_gathered:
+            u.gather(self.streams)
+
+    @torch.no_grad()
     def free_stale_fulls(self):
         """Post-optimizer bookkeeping when the optimizer kernel already
         wrote the bf16 shards itself (fused publish): only drop the stale
