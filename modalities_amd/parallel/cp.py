@@ -167,6 +167,9 @@ class _CPAttentionForward:
         if self.variant == "ulysses":
             y = cp_attention_ulysses(q, k, v, self.group, self.cp_rank,
                                      self.cp_size)
+        elif self.variant == "ring":
+            y = cp_attention_ring(q, k, v, self.group, self.cp_rank,
+                                  self.cp_size)
         else:
             y = cp_attention(q, k, v, self.group, self.cp_rank, self.cp_size)
         y = y.reshape(B, Tl, attn.n_head_q * attn.head_dim)
@@ -188,7 +191,7 @@ def get_gpt2_context_parallel_model(model, device_mesh=None, group=None,
         group, cp_rank, cp_size = dim.group, dim.rank, dim.size
     if cp_size in (None, 1):
         return model
-    if variant not in ("allgather", "ulysses"):
+    if variant not in ("allgather", "ulysses", "ring"):
         raise ValueError(f"Unknown CP variant {variant!r}")
     for block in model.blocks:
         block.attn.forward = _CPAttentionForward(block.attn, group, cp_rank,
@@ -211,3 +214,127 @@ def cp_grad_allreduce_(model, group) -> None:
     for p in model.parameters():
         if p.grad is not None:
             dist.all_reduce(p.grad, group=group)
+
+
+class _RingExchange(torch.autograd.Function):
+    """One ring step: send a [B, Tl, H, D] K/V chunk to rank+1, receive from
+    rank-1 (the xGMI p2p link pattern — each step uses one neighbor link
+    while attention computes). Backward reverses the ring direction."""
+
+    @staticmethod
+    def _shift(x, group, direction: int):
+        world = dist.get_world_size(group)
+        rank = dist.get_rank(group)
+        ranks = dist.get_process_group_ranks(group) if group is not None \
+            else list(range(world))
+        dst = ranks[(rank + direction) % world]
+        src = ranks[(rank - direction) % world]
+        recv = torch.empty_like(x)
+        send_op = dist.P2POp(dist.isend, x.contiguous(), dst, group=group)
+        recv_op = dist.P2POp(dist.irecv, recv, src, group=group)
+        for work in dist.batch_isend_irecv([send_op, recv_op]):
+            work.wait()
+        return recv
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return _RingExchange._shift(x, group, +1)
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _RingExchange._shift(grad.contiguous(), ctx.group, -1), None
+
+
+class _GraphTie(torch.autograd.Function):
+    """Identity on `out` that adds a zero-gradient dependency on `aux`."""
+
+    @staticmethod
+    def forward(ctx, out, aux):
+        ctx.aux_shape = aux.shape
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        return grad, grad.new_zeros(ctx.aux_shape)
+
+
+def _merge_partials(o_a, lse_a, o_b, lse_b):
+    """Online-softmax combination of two partial attentions (o in
+    [B,T,H,D], lse in [B,H,T])."""
+    lse_new = torch.logaddexp(lse_a, lse_b)
+    w_a = torch.exp(lse_a - lse_new).permute(0, 2, 1).unsqueeze(-1)
+    w_b = torch.exp(lse_b - lse_new).permute(0, 2, 1).unsqueeze(-1)
+    return o_a * w_a + o_b * w_b, lse_new
+
+
+def _flash_with_lse(q, k, v, q_offset):
+    """flash attention that also returns lse (natural log), differentiably.
+    On device this is the K1 kernel pair; on CPU a fp32 composed path."""
+    from modalities_amd.ops.backend import use_hip
+    if use_hip(q, k, v):
+        from modalities_amd.ops.backend import hip_ext
+
+        class _WithLse(torch.autograd.Function):
+            @staticmethod
+            def forward(ctx, q_, k_, v_):
+                o, lse = hip_ext().attn_fwd(q_, k_, v_, True, q_offset)
+                ctx.save_for_backward(q_, k_, v_, o, lse)
+                return o, lse
+
+            @staticmethod
+            def backward(ctx, do, dlse):
+                q_, k_, v_, o, lse = ctx.saved_tensors
+                dq, dk, dv = hip_ext().attn_bwd(do.contiguous(), q_, k_, v_,
+                                                o, lse, True, q_offset)
+                return dq, dk, dv
+
+        return _WithLse.apply(q.contiguous(), k.contiguous(), v.contiguous())
+    # composed fp32 path (CPU tests): compute scores once, derive o and lse
+    import math
+    B, T, Hq, D = q.shape
+    S, Hkv = k.shape[1], k.shape[2]
+    rep = Hq // Hkv
+    qf = q.permute(0, 2, 1, 3).float()
+    kf = k.permute(0, 2, 1, 3).float().repeat_interleave(rep, dim=1)
+    vf = v.permute(0, 2, 1, 3).float().repeat_interleave(rep, dim=1)
+    att = qf @ kf.transpose(-2, -1) / math.sqrt(D)
+    mask = torch.ones(T, S, dtype=torch.bool, device=q.device).tril(q_offset)
+    att = att.masked_fill(~mask, float("-inf"))
+    lse = att.logsumexp(-1)                                  # [B,Hq,T]
+    o = (att.softmax(-1) @ vf).permute(0, 2, 1, 3).to(q.dtype)
+    return o, lse
+
+
+def cp_attention_ring(q_local, k_local, v_local, group, cp_rank: int,
+                      cp_size: int) -> torch.Tensor:
+    """Ring CP (SURVEY.md §7 stage 9): K/V chunks rotate around the ring;
+    at step s this rank holds chunk (cp_rank - s) mod cp_size and computes
+    a partial attention — full (unmasked) for chunks strictly before its
+    queries, causal for its own chunk, skipped for later chunks — merging
+    partials with the online-softmax rule. The unmasked case reuses the
+    offset-causal kernel with q_offset = Tkv (mask allows every key)."""
+    if cp_size == 1:
+        return flash_attention(q_local, k_local, v_local, causal=True)
+    Tl = q_local.shape[1]
+    o = lse = None
+    # K and V travel as ONE stacked tensor: a single sequential exchange
+    # chain keeps every rank's backward collectives in the same order (two
+    # independent chains could interleave k/v sends differently per rank).
+    kv_cur = torch.stack((k_local, v_local))
+    for s in range(cp_size):
+        chunk = (cp_rank - s) % cp_size
+        if chunk <= cp_rank:
+            off = Tl if chunk < cp_rank else 0  # full vs diagonal-causal
+            o_s, lse_s = _flash_with_lse(q_local, kv_cur[0], kv_cur[1], off)
+            if o is None:
+                o, lse = o_s, lse_s
+            else:
+                o, lse = _merge_partials(o, lse, o_s, lse_s)
+        if s + 1 < cp_size:  # rotate K/V to the next rank
+            kv_cur = _RingExchange.apply(kv_cur, group)
+    # Tie the output to the end of the exchange chain: ranks whose queries
+    # never attend to a received chunk still must run that exchange's
+    # backward (it relays the K/V grads around the ring) — without the tie
+    # their autograd graph skips it and the ring deadlocks.
+    return _GraphTie.apply(o, kv_cur)

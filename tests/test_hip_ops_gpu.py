@@ -363,3 +363,25 @@ def test_attn_offset_causal_fwd_bwd(Tq, Tkv, off):
     torch.testing.assert_close(dq.float().cpu(), qf.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(dk.float().cpu(), kf.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(dv.float().cpu(), vf.grad, rtol=5e-2, atol=5e-2)
+
+
+@pytest.mark.gpu
+def test_ring_partial_flash_with_lse():
+    """The ring-CP building block on the HIP path: _flash_with_lse with
+    q_offset = Tkv (full, unmasked attention against a remote K/V chunk)
+    and the online-softmax merge of two partials equal one full causal
+    attention over the concatenated K/V."""
+    from modalities_amd.parallel.cp import _flash_with_lse, _merge_partials
+    from modalities_amd.ops.attention import _attention_ref
+    torch.manual_seed(11)
+    B, Tl, Hq, Hkv, D = 2, 128, 4, 2, 128
+    dev = "cuda"
+    q1 = torch.randn(B, Tl, Hq, D, device=dev, dtype=torch.bfloat16)
+    k = torch.randn(B, 2 * Tl, Hkv, D, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(B, 2 * Tl, Hkv, D, device=dev, dtype=torch.bfloat16)
+    # rank-1's view: full attention over chunk 0, causal over chunk 1
+    o_full, lse_full = _flash_with_lse(q1, k[:, :Tl], v[:, :Tl], Tl)
+    o_diag, lse_diag = _flash_with_lse(q1, k[:, Tl:], v[:, Tl:], 0)
+    o, _ = _merge_partials(o_full.float(), lse_full, o_diag.float(), lse_diag)
+    ref = _attention_ref(q1, k, v, causal=True, q_offset=Tl)
+    torch.testing.assert_close(o.to(torch.bfloat16), ref, rtol=2e-2, atol=2e-2)
