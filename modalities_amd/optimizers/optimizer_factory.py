@@ -5,7 +5,6 @@ fused HIP AdamW kernel (K9) and a per-element weight-decay mask — the flat
 equivalent of the reference's regex weight-decay groups (reference:
 src/modalities/optimizers/optimizer_factory.py:22-215)."""
 
-import re
 from typing import Optional
 
 import torch
