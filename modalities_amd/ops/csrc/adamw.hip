@@ -50,8 +50,13 @@ __global__ void adamw_masked_devstep_kernel(float* __restrict__ p,
                                             const float* __restrict__ wd_mask,
                                             const int* __restrict__ step,
                                             unsigned short* __restrict__ bf16_out,
+                                            const float* __restrict__ gscale,
                                             long n4, float lr, float beta1,
                                             float beta2, float eps, float wd) {
+  // gscale (device scalar, e.g. the grad-clip coefficient) is folded into
+  // the grad read here — saves the separate full read+write scale pass
+  // over every grad shard that multi_tensor_scale_ would cost.
+  const float gs = gscale ? *gscale : 1.0f;
   const float t = (float)*step;
   const float bc1 = 1.0f - __powf(beta1, t);
   const float bc2 = 1.0f - __powf(beta2, t);
@@ -68,9 +73,10 @@ __global__ void adamw_masked_devstep_kernel(float* __restrict__ p,
     shortx4 h;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
+      const float gj = gg[j] * gs;
       pp[j] *= (1.0f - lr * wd * ww[j]);
-      mm[j] = beta1 * mm[j] + (1.0f - beta1) * gg[j];
-      vv[j] = beta2 * vv[j] + (1.0f - beta2) * gg[j] * gg[j];
+      mm[j] = beta1 * mm[j] + (1.0f - beta1) * gj;
+      vv[j] = beta2 * vv[j] + (1.0f - beta2) * gj * gj;
       pp[j] -= step_size * mm[j] / (sqrtf(vv[j] * inv_bc2) + eps);
       h[j] = (short)f32_to_bf16(pp[j]);
     }
@@ -133,9 +139,16 @@ void fused_adamw_masked(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 void fused_adamw_masked_devstep(torch::Tensor p, torch::Tensor g,
                                 torch::Tensor m, torch::Tensor v,
                                 torch::Tensor wd_mask, torch::Tensor step,
-                                torch::Tensor bf16_out, double lr,
+                                torch::Tensor bf16_out,
+                                c10::optional<torch::Tensor> gscale, double lr,
                                 double beta1, double beta2, double eps,
                                 double wd) {
+  const float* gs_ptr = nullptr;
+  if (gscale.has_value()) {
+    TORCH_CHECK(gscale->is_cuda() && gscale->dtype() == torch::kFloat32
+                && gscale->numel() == 1, "gscale must be a cuda fp32 scalar");
+    gs_ptr = gscale->data_ptr<float>();
+  }
   TORCH_CHECK(p.is_cuda() && p.dtype() == torch::kFloat32 && p.is_contiguous());
   TORCH_CHECK(p.numel() % 4 == 0, "flat shard must be divisible by 4");
   TORCH_CHECK(step.is_cuda() && step.dtype() == torch::kInt32);
@@ -148,8 +161,9 @@ void fused_adamw_masked_devstep(torch::Tensor p, torch::Tensor g,
                      g.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), wd_mask.data_ptr<float>(),
                      step.data_ptr<int>(),
-                     (unsigned short*)bf16_out.data_ptr(), n4, (float)lr,
-                     (float)beta1, (float)beta2, (float)eps, (float)wd);
+                     (unsigned short*)bf16_out.data_ptr(), gs_ptr, n4,
+                     (float)lr, (float)beta1, (float)beta2, (float)eps,
+                     (float)wd);
   HIP_CHECK_KERNEL();
 }
 

@@ -22,7 +22,8 @@ void fused_adamw_masked(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 void fused_adamw_masked_devstep(torch::Tensor p, torch::Tensor g,
                                 torch::Tensor m, torch::Tensor v,
                                 torch::Tensor wd_mask, torch::Tensor step,
-                                torch::Tensor bf16_out, double lr,
+                                torch::Tensor bf16_out,
+                                c10::optional<torch::Tensor> gscale, double lr,
                                 double beta1, double beta2, double eps,
                                 double wd);
 void fused_adamw(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,

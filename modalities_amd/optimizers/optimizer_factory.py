@@ -24,6 +24,10 @@ class ShardedAdamW(torch.optim.Optimizer):
                  betas: tuple[float, float] = (0.9, 0.95), eps: float = 1e-8,
                  weight_decay: float = 0.1):
         self.sharded_model = sharded_model
+        # clip_grad_norm_ checks this flag to defer the clip-coefficient
+        # multiply into our fused kernel instead of a separate grad pass
+        sharded_model._optimizer_consumes_grad_scale = True
+        sharded_model._pending_grad_scale = None
         params = [u.master_shard for u in sharded_model.units]
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
@@ -55,6 +59,8 @@ class ShardedAdamW(torch.optim.Optimizer):
                     start, dtype=torch.int32,
                     device=self.sharded_model.units[0].master_shard.device)
             self._step_dev += 1  # on-device: advances under hipGraph replay
+        gscale = getattr(self.sharded_model, "_pending_grad_scale", None)
+        self.sharded_model._pending_grad_scale = None
         for ui, u in enumerate(self.sharded_model.units):
             if not u.grad_fresh:
                 continue  # no grads reduced for this unit this step
@@ -69,11 +75,13 @@ class ShardedAdamW(torch.optim.Optimizer):
                 # the same pass (saves a full param re-read + cast)
                 hip_ext().fused_adamw_masked_devstep(
                     p, g, m, v, u.wd_mask_shard, self._step_dev,
-                    u.bf16_shard, lr, beta1, beta2, eps, wd)
+                    u.bf16_shard, gscale, lr, beta1, beta2, eps, wd)
                 if u.full_buf is not None and self.sharded_model.world > 1:
                     u.free_full()  # stale gathered params
                 self.sharded_model.prefetch_unit_gather(ui)
             elif on_gpu:  # fp32 working copy on GPU: unfused publish below
+                if gscale is not None:
+                    g = g.mul(gscale)
                 bc1 = 1.0 - beta1 ** step
                 bc2 = 1.0 - beta2 ** step
                 hip_ext().fused_adamw_masked(p, g, m, v, u.wd_mask_shard,
@@ -81,6 +89,8 @@ class ShardedAdamW(torch.optim.Optimizer):
                                              bc1, bc2)
                 u.publish_master()
             else:
+                if gscale is not None:
+                    g = g.mul(gscale)
                 bc1 = 1.0 - beta1 ** step
                 bc2 = 1.0 - beta2 ** step
                 p.mul_(1.0 - lr * wd * u.wd_mask_shard)

@@ -490,7 +490,14 @@ class XGMIShardedModel(nn.Module):
         total = local.sqrt()
         if max_norm is not None and max_norm > 0 and fresh:
             clip = (max_norm / (total + 1e-6)).clamp(max=1.0)
-            multi_tensor_scale_(fresh, clip)
+            if getattr(self, "_optimizer_consumes_grad_scale", False) \
+                    and total.is_cuda:
+                # ShardedAdamW folds the clip coefficient into its fused
+                # kernel's grad read — deferring skips a full read+write
+                # pass over every grad shard here.
+                self._pending_grad_scale = clip
+            else:
+                multi_tensor_scale_(fresh, clip)
         return total
 
     # -- state (sharded checkpoints) --------------------------------------

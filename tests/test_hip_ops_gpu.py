@@ -385,3 +385,42 @@ def test_ring_partial_flash_with_lse():
     o, _ = _merge_partials(o_full.float(), lse_full, o_diag.float(), lse_diag)
     ref = _attention_ref(q1, k, v, causal=True, q_offset=Tl)
     torch.testing.assert_close(o.to(torch.bfloat16), ref, rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.gpu
+def test_deferred_clip_scale_matches_eager():
+    """ShardedAdamW folds the grad-clip coefficient into the fused kernel's
+    grad read; the resulting master weights must match applying
+    multi_tensor_scale_ eagerly before an unscaled step."""
+    from modalities_amd.models.gpt2 import GPT2LLM, GPT2LLMConfig
+    from modalities_amd.parallel.fsdp import XGMIShardedModel
+    from modalities_amd.optimizers.optimizer_factory import get_adam_w
+    from modalities_amd.ops import fused_cross_entropy
+
+    def run(defer: bool):
+        torch.manual_seed(0)
+        cfg = GPT2LLMConfig(vocab_size=512, n_layer=2, n_head_q=4,
+                            n_head_kv=2, n_embd=512, ffn_hidden=2048,
+                            sequence_length=256)
+        model = GPT2LLM(cfg)
+        sharded = XGMIShardedModel.from_transformer(
+            model, torch.device(DEV), blocks_per_unit=1,
+            param_dtype=torch.bfloat16)
+        opt = get_adam_w(sharded, lr=5e-4)
+        sharded._optimizer_consumes_grad_scale = defer
+        g = torch.Generator().manual_seed(7)
+        ids = torch.randint(0, 512, (2, 257), generator=g).to(DEV)
+        x, y = ids[:, :-1], ids[:, 1:]
+        for _ in range(3):
+            loss = fused_cross_entropy(sharded({"input_ids": x})["logits"], y)
+            loss.backward()
+            sharded.backward_epilogue()
+            # tight max_norm so the clip coefficient is far from 1.0
+            sharded.clip_grad_norm_(0.01)
+            opt.step()
+            opt.zero_grad()
+        return [u.master_shard.clone() for u in sharded.units]
+
+    a, b = run(True), run(False)
+    for ta, tb in zip(a, b):
+        torch.testing.assert_close(ta, tb, rtol=1e-6, atol=1e-7)
