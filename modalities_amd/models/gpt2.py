@@ -18,6 +18,8 @@ from pydantic import BaseModel, Field, model_validator
 
 from modalities_amd.models.model import NNModel, SwiGLU
 from modalities_amd.ops import flash_attention, precompute_rope_cos_sin, rope_apply
+from modalities_amd.ops.attention import fused_qkv_rope_attention
+from modalities_amd.ops.backend import use_hip
 from modalities_amd.ops.linear import TwoStreamLinear
 from modalities_amd.ops.rms_norm import RMSNorm
 
@@ -135,6 +137,14 @@ class CausalSelfAttention(nn.Module):
         kv_dim = self.head_dim * self.n_head_kv
         if self.fused_qkv:
             qkv = self.qkv_attn(x)
+            if (self.attention_impl == AttentionImplementation.HIP_FLASH
+                    and self.q_norm is None and rope_cos is not None
+                    and use_hip(qkv)):
+                # fused split+RoPE+attention (joint dqkv assembly in bwd)
+                y = fused_qkv_rope_attention(qkv, rope_cos, rope_sin,
+                                             self.n_head_q, self.n_head_kv,
+                                             self.head_dim)
+                return self.resid_dropout(self.c_proj(y.reshape(B, T, C)))
             q, k, v = qkv.split([C, kv_dim, kv_dim], dim=-1)
             q = q.view(B, T, self.n_head_q, self.head_dim)
             k = k.view(B, T, self.n_head_kv, self.head_dim)

@@ -412,12 +412,16 @@ struct DKDVKernel {
     else if (tid < 64) tl->delta[tid - 32] = r.stat;
   }
 
+  // dvp/dvc: row pitch (elements per (b,t) row) and column offset for the
+  // dv store — lets dv land directly in a fused-QKV gradient buffer
+  // [B,T,Cq+Ck+Cv] instead of its own tensor (kills the backward cat).
   static __device__ void run(const unsigned short* q, const unsigned short* k,
                              const unsigned short* v, const unsigned short* dout,
                              const float* lse, const float* delta,
                              unsigned short* dk, unsigned short* dv,
                              int B, int Tq, int Tkv, int q_off, int Hq,
-                             int Hkv, float scale, char* smem_raw) {
+                             int Hkv, float scale, long dvp, long dvc,
+                             char* smem_raw) {
     Smem* sm = reinterpret_cast<Smem*>(smem_raw);
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
@@ -642,8 +646,10 @@ struct DKDVKernel {
         const int kvrow = kvblk0 + wid * 32 + crow(r, hi);
         if (all_in || kvrow < Tkv) {
           const long off = kv_base + (long)kvrow * Hkv * D + dblk * 32 + ln31;
+          const long dvoff = ((long)b * Tkv + kvrow) * dvp + dvc
+              + (long)hkv * D + dblk * 32 + ln31;
           dk[off] = f32_to_bf16(acc_dk[dblk][r]);
-          dv[off] = f32_to_bf16(acc_dv[dblk][r]);
+          dv[dvoff] = f32_to_bf16(acc_dv[dblk][r]);
         }
       }
   }
@@ -665,18 +671,18 @@ __global__ __launch_bounds__(256) void dkdv_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
     const unsigned short* dout, const float* lse, const float* delta,
     unsigned short* dk, unsigned short* dv, int B, int Tq, int Tkv, int q_off,
-    int Hq, int Hkv, float scale) {
+    int Hq, int Hkv, float scale, long dvp, long dvc) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   DKDVKernel<D, ABL>::run(q, k, v, dout, lse, delta, dk, dv, B, Tq, Tkv,
-                          q_off, Hq, Hkv, scale, smem_raw);
+                          q_off, Hq, Hkv, scale, dvp, dvc, smem_raw);
 }
 
 }  // namespace attnbwd
 
-std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
-                                    torch::Tensor k, torch::Tensor v,
-                                    torch::Tensor o, torch::Tensor lse,
-                                    bool causal, long q_offset) {
+static std::vector<torch::Tensor> attn_bwd_impl(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor o, torch::Tensor lse, bool causal, long q_offset,
+    torch::Tensor dv, long dvp, long dvc) {
   TORCH_CHECK(causal, "attn_bwd: only causal attention is implemented");
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
   const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
@@ -701,7 +707,6 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
 
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
 
   // dq and dkdv are independent given delta: run them on two streams so
   // their partially-filled tail waves overlap (each launches ~1280 blocks
@@ -738,7 +743,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (unsigned short*)dk.data_ptr(),
-                       (unsigned short*)dv.data_ptr(), B, T, Tkv, q_off, Hq, Hkv, scale);
+                       (unsigned short*)dv.data_ptr(), B, T, Tkv, q_off, Hq, Hkv, scale, dvp, dvc);
     HIP_CHECK_KERNEL();
   } else if (D == 64) {
     size_t smem_dq = sizeof(typename attnbwd::DQKernel<64>::Smem);
@@ -759,7 +764,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (unsigned short*)dk.data_ptr(),
-                       (unsigned short*)dv.data_ptr(), B, T, Tkv, q_off, Hq, Hkv, scale);
+                       (unsigned short*)dv.data_ptr(), B, T, Tkv, q_off, Hq, Hkv, scale, dvp, dvc);
     HIP_CHECK_KERNEL();
   } else {
     TORCH_CHECK(false, "attn_bwd: head_dim must be 64 or 128");
@@ -767,6 +772,36 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   hipEventRecord(ev_join, side);
   hipStreamWaitEvent(stream, ev_join, 0);
   return {dq, dk, dv};
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    bool causal, long q_offset) {
+  auto dv = torch::empty_like(v);
+  const long Hkv = v.size(2), D = v.size(3);
+  auto r = attn_bwd_impl(dout, q, k, v, o, lse, causal, q_offset, dv,
+                         Hkv * D, 0);
+  return {r[0], r[1], dv};
+}
+
+// Fused-QKV variant: dv is written directly into its column block of the
+// joint gradient buffer dqkv [B, T, Cq+Ck+Cv] (column offset dv_col_off);
+// returns {dq, dk} for the RoPE-backward stage to scatter the same way.
+std::vector<torch::Tensor> attn_bwd_qkvjoint(torch::Tensor dout,
+                                             torch::Tensor q, torch::Tensor k,
+                                             torch::Tensor v, torch::Tensor o,
+                                             torch::Tensor lse, bool causal,
+                                             long q_offset, torch::Tensor dqkv,
+                                             long dv_col_off) {
+  TORCH_CHECK(dqkv.is_cuda() && dqkv.dtype() == torch::kBFloat16
+              && dqkv.dim() == 3 && dqkv.is_contiguous());
+  const long Ctot = dqkv.size(2), Hkv = v.size(2), D = v.size(3);
+  TORCH_CHECK(dqkv.size(0) == v.size(0) && dqkv.size(1) == v.size(1)
+              && dv_col_off + Hkv * D <= Ctot);
+  auto r = attn_bwd_impl(dout, q, k, v, o, lse, causal, q_offset, dqkv,
+                         Ctot, dv_col_off);
+  return {r[0], r[1]};
 }
 
 // Ablation entry: times ONLY the dkdv kernel at the given mode (D=128).
@@ -796,7 +831,7 @@ std::vector<torch::Tensor> attn_bwd_dkdv_ablate(torch::Tensor dout,
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (unsigned short*)dk.data_ptr(),
                        (unsigned short*)dv.data_ptr(), B, T, T, 0, Hq, Hkv,
-                       scale);
+                       scale, (long)Hkv * 128, 0L);
   };
   switch (mode) {
     case 0: launch(attnbwd::dkdv_kernel<128, 0>); break;

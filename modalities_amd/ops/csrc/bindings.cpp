@@ -6,6 +6,11 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor w, torch::Tensor invrms);
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t,
                        bool neg);
+torch::Tensor rope_fwd_slice(torch::Tensor qkv, torch::Tensor cos_t,
+                             torch::Tensor sin_t, long col_off, long Hn,
+                             long D, bool neg);
+void rope_bwd_slice(torch::Tensor dx, torch::Tensor cos_t, torch::Tensor sin_t,
+                    torch::Tensor dqkv, long col_off);
 torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u);
 std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dout, torch::Tensor g,
                                         torch::Tensor u);
@@ -43,6 +48,12 @@ torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
 torch::Tensor tr16_probe(long mode);
 std::vector<torch::Tensor> attn_fwd_ablate(torch::Tensor q, torch::Tensor k,
                                            torch::Tensor v, long mode);
+std::vector<torch::Tensor> attn_bwd_qkvjoint(torch::Tensor dout,
+                                             torch::Tensor q, torch::Tensor k,
+                                             torch::Tensor v, torch::Tensor o,
+                                             torch::Tensor lse, bool causal,
+                                             long q_offset, torch::Tensor dqkv,
+                                             long dv_col_off);
 std::vector<torch::Tensor> attn_bwd_dkdv_ablate(torch::Tensor dout,
                                                 torch::Tensor q,
                                                 torch::Tensor k,
@@ -54,6 +65,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (K5)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (K5)");
   m.def("rope_fwd", &rope_fwd, "RoPE rotate-half (K4); neg=true => inverse");
+  m.def("rope_fwd_slice", &rope_fwd_slice,
+        "RoPE a head-block slice of fused QKV into a contiguous [B,T,H,D]");
+  m.def("rope_bwd_slice", &rope_bwd_slice,
+        "inverse-RoPE a [B,T,H,D] grad into a fused-QKV grad buffer slice");
   m.def("silu_mul_fwd", &silu_mul_fwd, "silu(g)*u forward (K6)");
   m.def("silu_mul_bwd", &silu_mul_bwd, "silu(g)*u backward (K6)");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE forward (K8)");
@@ -77,6 +92,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA fragment-layout probe (verification)");
   m.def("attn_fwd_ablate", &attn_fwd_ablate,
         "attention fwd cost-attribution ablation (timing only)");
+  m.def("attn_bwd_qkvjoint", &attn_bwd_qkvjoint,
+        "attention backward writing dv into a fused-QKV grad buffer");
   m.def("attn_bwd_dkdv_ablate", &attn_bwd_dkdv_ablate,
         "attention bwd dkdv cost-attribution ablation (timing only)");
 }

@@ -14,20 +14,28 @@ namespace {
 // out_{j+D/2} = x_{j+D/2} * c_j + x_j * s_j
 // backward = rotation by -theta (neg=true flips the sin sign).
 // One thread handles 2 adjacent (lo, hi) pairs -> 2x ushort2 loads per half.
+// in/out may be SLICES of a wider row (fused-QKV layout [B,T,Cq+Ck+Cv]):
+// ipitch/opitch are elements between consecutive (b,t) rows, ioff/ooff the
+// column offset of this slice's head block. Contiguous [B,T,H,D] tensors
+// pass pitch=Hn*D, off=0.
 __global__ void rope_kernel(const unsigned short* __restrict__ x,
                             const float* __restrict__ cos_t,
                             const float* __restrict__ sin_t,
                             unsigned short* __restrict__ out,
                             long total_pairs2,  // B*T*H*(D/2)/2
-                            int T, int Hn, int D, int neg) {
+                            int T, int Hn, int D, int neg,
+                            long ipitch, long ioff, long opitch, long ooff) {
   const int halfD = D / 2;
   const int pairs_per_head2 = halfD / 2;
   for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x;
        idx < total_pairs2; idx += (long)gridDim.x * blockDim.x) {
     const long head_idx = idx / pairs_per_head2;       // b*T*H + t*H + h
     const int j2 = (int)(idx % pairs_per_head2) * 2;   // pair start in [0, D/2)
-    const int t = (int)((head_idx / Hn) % T);
-    const long base = head_idx * (long)D;
+    const int h = (int)(head_idx % Hn);
+    const long bt = head_idx / Hn;
+    const int t = (int)(bt % T);
+    const long base = bt * ipitch + ioff + (long)h * D;
+    const long obase = bt * opitch + ooff + (long)h * D;
 
     const unsigned int* lo_p = reinterpret_cast<const unsigned int*>(x + base + j2);
     const unsigned int* hi_p =
@@ -49,8 +57,8 @@ __global__ void rope_kernel(const unsigned short* __restrict__ x,
     float o_hi0 = hi0 * cs0.x + lo0 * s0;
     float o_hi1 = hi1 * cs1.x + lo1 * s1;
 
-    unsigned int* out_lo = reinterpret_cast<unsigned int*>(out + base + j2);
-    unsigned int* out_hi = reinterpret_cast<unsigned int*>(out + base + halfD + j2);
+    unsigned int* out_lo = reinterpret_cast<unsigned int*>(out + obase + j2);
+    unsigned int* out_hi = reinterpret_cast<unsigned int*>(out + obase + halfD + j2);
     *out_lo = (unsigned int)f32_to_bf16(o_lo0) |
               ((unsigned int)f32_to_bf16(o_lo1) << 16);
     *out_hi = (unsigned int)f32_to_bf16(o_hi0) |
@@ -129,9 +137,57 @@ torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t
   hipLaunchKernelGGL(rope_kernel, dim3(grid_for(total2, 256)), dim3(256), 0, stream,
                      (const unsigned short*)x.data_ptr(),
                      cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
-                     (unsigned short*)out.data_ptr(), total2, T, Hn, D, neg ? 1 : 0);
+                     (unsigned short*)out.data_ptr(), total2, T, Hn, D, neg ? 1 : 0,
+                     (long)Hn * D, 0L, (long)Hn * D, 0L);
   HIP_CHECK_KERNEL();
   return out;
+}
+
+// RoPE a head-block slice of a fused-QKV activation [B, T, Ctot] into a
+// contiguous [B, T, Hn, D] tensor (no .contiguous() copy of the slice).
+torch::Tensor rope_fwd_slice(torch::Tensor qkv, torch::Tensor cos_t,
+                             torch::Tensor sin_t, long col_off, long Hn,
+                             long D, bool neg) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16
+              && qkv.dim() == 3 && qkv.is_contiguous());
+  const int B = qkv.size(0), T = qkv.size(1);
+  const long Ctot = qkv.size(2);
+  TORCH_CHECK(col_off % 2 == 0 && (D % 4) == 0 && col_off + Hn * D <= Ctot);
+  TORCH_CHECK(cos_t.size(0) >= T && cos_t.size(1) == D / 2
+              && cos_t.dtype() == torch::kFloat32 && cos_t.is_contiguous());
+  auto out = torch::empty({B, T, Hn, D}, qkv.options());
+  long total2 = (long)B * T * Hn * (D / 2) / 2;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(rope_kernel, dim3(grid_for(total2, 256)), dim3(256), 0,
+                     stream, (const unsigned short*)qkv.data_ptr(),
+                     cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+                     (unsigned short*)out.data_ptr(), total2, T, (int)Hn,
+                     (int)D, neg ? 1 : 0, Ctot, col_off, Hn * D, 0L);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
+// Inverse-RoPE a contiguous [B, T, Hn, D] gradient into a head-block slice
+// of a fused-QKV gradient buffer [B, T, Ctot] (replaces the backward cat).
+void rope_bwd_slice(torch::Tensor dx, torch::Tensor cos_t, torch::Tensor sin_t,
+                    torch::Tensor dqkv, long col_off) {
+  TORCH_CHECK(dx.is_cuda() && dx.dtype() == torch::kBFloat16 && dx.dim() == 4
+              && dx.is_contiguous());
+  TORCH_CHECK(dqkv.is_cuda() && dqkv.dtype() == torch::kBFloat16
+              && dqkv.dim() == 3 && dqkv.is_contiguous());
+  const int B = dx.size(0), T = dx.size(1), Hn = dx.size(2), D = dx.size(3);
+  const long Ctot = dqkv.size(2);
+  TORCH_CHECK(dqkv.size(0) == B && dqkv.size(1) == T
+              && col_off + (long)Hn * D <= Ctot && col_off % 2 == 0);
+  TORCH_CHECK(cos_t.size(0) >= T && cos_t.size(1) == D / 2);
+  long total2 = (long)B * T * Hn * (D / 2) / 2;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(rope_kernel, dim3(grid_for(total2, 256)), dim3(256), 0,
+                     stream, (const unsigned short*)dx.data_ptr(),
+                     cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+                     (unsigned short*)dqkv.data_ptr(), total2, T, Hn, D, 1,
+                     (long)Hn * D, 0L, Ctot, col_off);
+  HIP_CHECK_KERNEL();
 }
 
 torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u) {

@@ -9,12 +9,20 @@ in attention_bwd.hip, which measured -16% there). State-dict compatible
 with nn.Linear; inactive on CPU.
 """
 
+import os
 from typing import Optional
 
 import torch
 import torch.nn as nn
 
 _side_stream: Optional[torch.cuda.Stream] = None
+
+# Overlap only pays when one GEMM underfills the 256-CU chip: for very large
+# weights (lm_head: 50304x2560) both wgrad and dgrad already use
+# chip-filling persistent kernels, and co-running them degrades each more
+# than the overlap saves. Weights above this element count run sequentially.
+_TWO_STREAM_MAX_WEIGHT = int(os.environ.get(
+    "MODALITIES_AMD_TWO_STREAM_MAX_WEIGHT", 64 * 1024 * 1024))
 
 
 def _get_side_stream() -> torch.cuda.Stream:
@@ -61,6 +69,7 @@ class _TwoStreamLinearFn(torch.autograd.Function):
 class TwoStreamLinear(nn.Linear):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if x.is_cuda and torch.is_grad_enabled() and (
-                x.requires_grad or self.weight.requires_grad):
+                x.requires_grad or self.weight.requires_grad) and \
+                self.weight.numel() <= _TWO_STREAM_MAX_WEIGHT:
             return _TwoStreamLinearFn.apply(x, self.weight, self.bias)
         return torch.nn.functional.linear(x, self.weight, self.bias)
