@@ -85,6 +85,60 @@ __global__ void adamw_masked_devstep_kernel(float* __restrict__ p,
   }
 }
 
+template <int W>
+__global__ void adamw_devstep_wide_kernel(float* __restrict__ p,
+                                          const float* __restrict__ g,
+                                          float* __restrict__ m,
+                                          float* __restrict__ v,
+                                          const float* __restrict__ wd_mask,
+                                          const int* __restrict__ step,
+                                          unsigned short* __restrict__ bf16_out,
+                                          const float* __restrict__ gscale,
+                                          long n4, float lr, float beta1,
+                                          float beta2, float eps, float wd) {
+  const float gs = gscale ? *gscale : 1.0f;
+  const float t = (float)*step;
+  const float bc1 = 1.0f - __powf(beta1, t);
+  const float bc2 = 1.0f - __powf(beta2, t);
+  const float step_size = lr / bc1;
+  const float inv_bc2 = 1.0f / bc2;
+  floatx4* p4 = reinterpret_cast<floatx4*>(p);
+  const floatx4* g4 = reinterpret_cast<const floatx4*>(g);
+  floatx4* m4 = reinterpret_cast<floatx4*>(m);
+  floatx4* v4 = reinterpret_cast<floatx4*>(v);
+  const floatx4* w4 = reinterpret_cast<const floatx4*>(wd_mask);
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * W;
+       i0 < n4; i0 += stride * W) {
+    floatx4 pp[W], gg[W], mm[W], vv[W], ww[W];
+    shortx4 h[W];
+#pragma unroll
+    for (int u = 0; u < W; ++u)
+      if (i0 + u < n4) {
+        pp[u] = p4[i0 + u]; gg[u] = g4[i0 + u]; mm[u] = m4[i0 + u];
+        vv[u] = v4[i0 + u]; ww[u] = w4[i0 + u];
+      }
+#pragma unroll
+    for (int u = 0; u < W; ++u) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float gj = gg[u][j] * gs;
+        pp[u][j] *= (1.0f - lr * wd * ww[u][j]);
+        mm[u][j] = beta1 * mm[u][j] + (1.0f - beta1) * gj;
+        vv[u][j] = beta2 * vv[u][j] + (1.0f - beta2) * gj * gj;
+        pp[u][j] -= step_size * mm[u][j] / (sqrtf(vv[u][j] * inv_bc2) + eps);
+        h[u][j] = (short)f32_to_bf16(pp[u][j]);
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < W; ++u)
+      if (i0 + u < n4) {
+        p4[i0 + u] = pp[u]; m4[i0 + u] = mm[u]; v4[i0 + u] = vv[u];
+        reinterpret_cast<shortx4*>(bf16_out)[i0 + u] = h[u];
+      }
+  }
+}
+
 __global__ void sqsum_kernel(const float* __restrict__ x, long n4,
                              float* __restrict__ out) {
   __shared__ float scratch[256 / WAVE_SIZE];
@@ -156,14 +210,29 @@ void fused_adamw_masked_devstep(torch::Tensor p, torch::Tensor g,
               && bf16_out.numel() == p.numel());
   long n4 = p.numel() / 4;
   auto stream = at::cuda::getCurrentHIPStream();
-  hipLaunchKernelGGL(adamw_masked_devstep_kernel, dim3(grid_for(n4, 256)),
-                     dim3(256), 0, stream, p.data_ptr<float>(),
-                     g.data_ptr<float>(), m.data_ptr<float>(),
-                     v.data_ptr<float>(), wd_mask.data_ptr<float>(),
-                     step.data_ptr<int>(),
-                     (unsigned short*)bf16_out.data_ptr(), gs_ptr, n4,
-                     (float)lr, (float)beta1, (float)beta2, (float)eps,
-                     (float)wd);
+  static const int wide = []() {
+    const char* e = getenv("MODALITIES_AMD_ADAMW_WIDE");
+    return e ? atoi(e) : 2;  // 2x float4 per thread measured fastest
+  }();
+  if (wide >= 2) {
+    hipLaunchKernelGGL(adamw_devstep_wide_kernel<2>,
+                       dim3(grid_for((n4 + 1) / 2, 256)), dim3(256), 0, stream,
+                       p.data_ptr<float>(), g.data_ptr<float>(),
+                       m.data_ptr<float>(), v.data_ptr<float>(),
+                       wd_mask.data_ptr<float>(), step.data_ptr<int>(),
+                       (unsigned short*)bf16_out.data_ptr(), gs_ptr, n4,
+                       (float)lr, (float)beta1, (float)beta2, (float)eps,
+                       (float)wd);
+  } else {
+    hipLaunchKernelGGL(adamw_masked_devstep_kernel, dim3(grid_for(n4, 256)),
+                       dim3(256), 0, stream, p.data_ptr<float>(),
+                       g.data_ptr<float>(), m.data_ptr<float>(),
+                       v.data_ptr<float>(), wd_mask.data_ptr<float>(),
+                       step.data_ptr<int>(),
+                       (unsigned short*)bf16_out.data_ptr(), gs_ptr, n4,
+                       (float)lr, (float)beta1, (float)beta2, (float)eps,
+                       (float)wd);
+  }
   HIP_CHECK_KERNEL();
 }
 
