@@ -22,48 +22,44 @@ __global__ void rope_kernel(const unsigned short* __restrict__ x,
                             const float* __restrict__ cos_t,
                             const float* __restrict__ sin_t,
                             unsigned short* __restrict__ out,
-                            long total_pairs2,  // B*T*H*(D/2)/2
                             int T, int Hn, int D, int neg,
                             long ipitch, long ioff, long opitch, long ooff) {
+  // grid = (ceil(Hn*pairs2 / block), T, B): t and b come straight from
+  // blockIdx (no 64-bit div/mod per element — that serialized the old
+  // grid-stride form well below HBM bandwidth).
   const int halfD = D / 2;
-  const int pairs_per_head2 = halfD / 2;
-  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x;
-       idx < total_pairs2; idx += (long)gridDim.x * blockDim.x) {
-    const long head_idx = idx / pairs_per_head2;       // b*T*H + t*H + h
-    const int j2 = (int)(idx % pairs_per_head2) * 2;   // pair start in [0, D/2)
-    const int h = (int)(head_idx % Hn);
-    const long bt = head_idx / Hn;
-    const int t = (int)(bt % T);
-    const long base = bt * ipitch + ioff + (long)h * D;
-    const long obase = bt * opitch + ooff + (long)h * D;
+  const int pairs2 = halfD / 2;
+  const int w = blockIdx.x * blockDim.x + threadIdx.x;
+  if (w >= Hn * pairs2) return;
+  const int h = w / pairs2;
+  const int j2 = (w % pairs2) * 2;
+  const int t = blockIdx.y;
+  const long bt = (long)blockIdx.z * T + t;
+  const long base = bt * ipitch + ioff + (long)h * D;
+  const long obase = bt * opitch + ooff + (long)h * D;
 
-    const unsigned int* lo_p = reinterpret_cast<const unsigned int*>(x + base + j2);
-    const unsigned int* hi_p =
-        reinterpret_cast<const unsigned int*>(x + base + halfD + j2);
-    unsigned int lo_u = *lo_p, hi_u = *hi_p;
-    float lo0 = bf16_to_f32((unsigned short)(lo_u & 0xffff));
-    float lo1 = bf16_to_f32((unsigned short)(lo_u >> 16));
-    float hi0 = bf16_to_f32((unsigned short)(hi_u & 0xffff));
-    float hi1 = bf16_to_f32((unsigned short)(hi_u >> 16));
+  const unsigned int lo_u =
+      *reinterpret_cast<const unsigned int*>(x + base + j2);
+  const unsigned int hi_u =
+      *reinterpret_cast<const unsigned int*>(x + base + halfD + j2);
+  const float lo0 = bf16_to_f32((unsigned short)(lo_u & 0xffff));
+  const float lo1 = bf16_to_f32((unsigned short)(lo_u >> 16));
+  const float hi0 = bf16_to_f32((unsigned short)(hi_u & 0xffff));
+  const float hi1 = bf16_to_f32((unsigned short)(hi_u >> 16));
 
-    const float2 cs0 = make_float2(cos_t[t * halfD + j2], sin_t[t * halfD + j2]);
-    const float2 cs1 = make_float2(cos_t[t * halfD + j2 + 1],
-                                   sin_t[t * halfD + j2 + 1]);
-    const float s0 = neg ? -cs0.y : cs0.y;
-    const float s1 = neg ? -cs1.y : cs1.y;
+  const float c0 = cos_t[t * halfD + j2], c1 = cos_t[t * halfD + j2 + 1];
+  float s0 = sin_t[t * halfD + j2], s1 = sin_t[t * halfD + j2 + 1];
+  if (neg) { s0 = -s0; s1 = -s1; }
 
-    float o_lo0 = lo0 * cs0.x - hi0 * s0;
-    float o_lo1 = lo1 * cs1.x - hi1 * s1;
-    float o_hi0 = hi0 * cs0.x + lo0 * s0;
-    float o_hi1 = hi1 * cs1.x + lo1 * s1;
+  const float o_lo0 = lo0 * c0 - hi0 * s0;
+  const float o_lo1 = lo1 * c1 - hi1 * s1;
+  const float o_hi0 = hi0 * c0 + lo0 * s0;
+  const float o_hi1 = hi1 * c1 + lo1 * s1;
 
-    unsigned int* out_lo = reinterpret_cast<unsigned int*>(out + obase + j2);
-    unsigned int* out_hi = reinterpret_cast<unsigned int*>(out + obase + halfD + j2);
-    *out_lo = (unsigned int)f32_to_bf16(o_lo0) |
-              ((unsigned int)f32_to_bf16(o_lo1) << 16);
-    *out_hi = (unsigned int)f32_to_bf16(o_hi0) |
-              ((unsigned int)f32_to_bf16(o_hi1) << 16);
-  }
+  *reinterpret_cast<unsigned int*>(out + obase + j2) =
+      (unsigned int)f32_to_bf16(o_lo0) | ((unsigned int)f32_to_bf16(o_lo1) << 16);
+  *reinterpret_cast<unsigned int*>(out + obase + halfD + j2) =
+      (unsigned int)f32_to_bf16(o_hi0) | ((unsigned int)f32_to_bf16(o_hi1) << 16);
 }
 
 // ---------------- SiLU * mul ----------------------------------------------
@@ -129,15 +125,15 @@ torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t
               && x.is_contiguous(), "rope: x must be contiguous bf16 [B,T,H,D]");
   const int B = x.size(0), T = x.size(1), Hn = x.size(2), D = x.size(3);
   TORCH_CHECK(D % 4 == 0, "rope: head_dim must be divisible by 4");
-  TORCH_CHECK(cos_t.size(0) >= T && cos_t.size(1) == D / 2);
+  TORCH_CHECK(T <= 65535 && cos_t.size(0) >= T && cos_t.size(1) == D / 2);
   TORCH_CHECK(cos_t.dtype() == torch::kFloat32 && cos_t.is_contiguous());
   auto out = torch::empty_like(x);
-  long total2 = (long)B * T * Hn * (D / 2) / 2;
   auto stream = at::cuda::getCurrentHIPStream();
-  hipLaunchKernelGGL(rope_kernel, dim3(grid_for(total2, 256)), dim3(256), 0, stream,
-                     (const unsigned short*)x.data_ptr(),
+  const int work = Hn * (D / 4);
+  hipLaunchKernelGGL(rope_kernel, dim3((work + 255) / 256, T, B), dim3(256),
+                     0, stream, (const unsigned short*)x.data_ptr(),
                      cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
-                     (unsigned short*)out.data_ptr(), total2, T, Hn, D, neg ? 1 : 0,
+                     (unsigned short*)out.data_ptr(), T, Hn, D, neg ? 1 : 0,
                      (long)Hn * D, 0L, (long)Hn * D, 0L);
   HIP_CHECK_KERNEL();
   return out;
@@ -153,15 +149,15 @@ torch::Tensor rope_fwd_slice(torch::Tensor qkv, torch::Tensor cos_t,
   const int B = qkv.size(0), T = qkv.size(1);
   const long Ctot = qkv.size(2);
   TORCH_CHECK(col_off % 2 == 0 && (D % 4) == 0 && col_off + Hn * D <= Ctot);
-  TORCH_CHECK(cos_t.size(0) >= T && cos_t.size(1) == D / 2
+  TORCH_CHECK(T <= 65535 && cos_t.size(0) >= T && cos_t.size(1) == D / 2
               && cos_t.dtype() == torch::kFloat32 && cos_t.is_contiguous());
   auto out = torch::empty({B, T, Hn, D}, qkv.options());
-  long total2 = (long)B * T * Hn * (D / 2) / 2;
   auto stream = at::cuda::getCurrentHIPStream();
-  hipLaunchKernelGGL(rope_kernel, dim3(grid_for(total2, 256)), dim3(256), 0,
-                     stream, (const unsigned short*)qkv.data_ptr(),
+  const int work = (int)(Hn * (D / 4));
+  hipLaunchKernelGGL(rope_kernel, dim3((work + 255) / 256, T, B), dim3(256),
+                     0, stream, (const unsigned short*)qkv.data_ptr(),
                      cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
-                     (unsigned short*)out.data_ptr(), total2, T, (int)Hn,
+                     (unsigned short*)out.data_ptr(), T, (int)Hn,
                      (int)D, neg ? 1 : 0, Ctot, col_off, Hn * D, 0L);
   HIP_CHECK_KERNEL();
   return out;
@@ -179,13 +175,13 @@ void rope_bwd_slice(torch::Tensor dx, torch::Tensor cos_t, torch::Tensor sin_t,
   const long Ctot = dqkv.size(2);
   TORCH_CHECK(dqkv.size(0) == B && dqkv.size(1) == T
               && col_off + (long)Hn * D <= Ctot && col_off % 2 == 0);
-  TORCH_CHECK(cos_t.size(0) >= T && cos_t.size(1) == D / 2);
-  long total2 = (long)B * T * Hn * (D / 2) / 2;
+  TORCH_CHECK(T <= 65535 && cos_t.size(0) >= T && cos_t.size(1) == D / 2);
   auto stream = at::cuda::getCurrentHIPStream();
-  hipLaunchKernelGGL(rope_kernel, dim3(grid_for(total2, 256)), dim3(256), 0,
-                     stream, (const unsigned short*)dx.data_ptr(),
+  const int work = Hn * (D / 4);
+  hipLaunchKernelGGL(rope_kernel, dim3((work + 255) / 256, T, B), dim3(256),
+                     0, stream, (const unsigned short*)dx.data_ptr(),
                      cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
-                     (unsigned short*)dqkv.data_ptr(), total2, T, Hn, D, 1,
+                     (unsigned short*)dqkv.data_ptr(), T, Hn, D, 1,
                      (long)Hn * D, 0L, Ctot, col_off);
   HIP_CHECK_KERNEL();
 }

@@ -217,9 +217,10 @@ def cp_grad_allreduce_(model, group) -> None:
 
 
 class _RingExchange(torch.autograd.Function):
-    """One ring step: send a [B, Tl, H, D] K/V chunk to rank+1, receive from
-    rank-1 (the xGMI p2p link pattern — each step uses one neighbor link
-    while attention computes). Backward reverses the ring direction."""
+    """One ring step: send the stacked [2, B, Tl, H, D] K/V chunk to rank+1,
+    receive from rank-1 (the xGMI p2p link pattern — each step uses one
+    neighbor link while attention computes). Backward reverses the ring
+    direction, relaying the chunk's gradient back toward its owner."""
 
     @staticmethod
     def _shift(x, group, direction: int):
