@@ -62,6 +62,7 @@ __global__ void rmsnorm_fwd_kernel(const unsigned short* __restrict__ x,
 // dx_j = r*w_j*dy_j - r^3 * x_j / H * sum_i(dy_i * w_i * x_i)
 // dw_j = sum_rows dy_j * x_j * r   (accumulated per-block in LDS, one
 // atomicAdd per element per block — guide G12 contention rule).
+template <int SLOTS>
 __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
                                    const unsigned short* __restrict__ x,
                                    const unsigned short* __restrict__ w,
@@ -69,66 +70,103 @@ __global__ void rmsnorm_bwd_kernel(const unsigned short* __restrict__ dy,
                                    unsigned short* __restrict__ dx,
                                    float* __restrict__ dw,  // f32 accumulator
                                    int H, long N, int rows_per_block) {
+  // One WAVE per row, and dw contributions accumulate in REGISTERS: lane l
+  // always touches vec8 columns {l, l+64, ...} (SLOTS of them), so its dw
+  // partials live in dwacc[][] across the whole row loop and spill to the
+  // per-wave LDS slab exactly once. (v1 block-per-row paid a barrier per
+  // row; v2 wave-per-row paid 16 LDS read-modify-writes per 16B of HBM
+  // traffic — both measured well under 20% of HBM bandwidth.)
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* dw_local = reinterpret_cast<float*>(smem_raw);            // H floats
-  float* scratch = dw_local + H;                                   // BLOCK/64
-
-  for (int i = threadIdx.x; i < H; i += BLOCK) dw_local[i] = 0.f;
-  __syncthreads();
+  float* slab = reinterpret_cast<float*>(smem_raw);
+  const int nwaves = blockDim.x / WAVE_SIZE;
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int wid = threadIdx.x / WAVE_SIZE;
+  float* my_slab = slab + (long)wid * H;
 
   const long row0 = (long)blockIdx.x * rows_per_block;
   const long row1 = min(row0 + rows_per_block, N);
   const int vecH = H / 8;
 
-  for (long row = row0; row < row1; ++row) {
-    const unsigned short* xr = x + row * (long)H;
-    const unsigned short* dyr = dy + row * (long)H;
-    unsigned short* dxr = dx + row * (long)H;
+  float dwacc[SLOTS][8];
+#pragma unroll
+  for (int it = 0; it < SLOTS; ++it)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dwacc[it][j] = 0.f;
+
+  const shortx8* wv = reinterpret_cast<const shortx8*>(w);
+  for (long row = row0 + wid; row < row1; row += nwaves) {
+    const shortx8* xv = reinterpret_cast<const shortx8*>(x + row * (long)H);
+    const shortx8* dyv = reinterpret_cast<const shortx8*>(dy + row * (long)H);
+    shortx8* dxv = reinterpret_cast<shortx8*>(dx + row * (long)H);
     const float r = invrms[row];
 
-    const shortx8* xv = reinterpret_cast<const shortx8*>(xr);
-    const shortx8* dyv = reinterpret_cast<const shortx8*>(dyr);
-    const shortx8* wv = reinterpret_cast<const shortx8*>(w);
-
     float dot = 0.f;
-    for (int i = threadIdx.x; i < vecH; i += BLOCK) {
-      shortx8 xvv = xv[i], dyvv = dyv[i], wvv = wv[i];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        dot += bf16_to_f32((unsigned short)dyvv[j]) *
-               bf16_to_f32((unsigned short)wvv[j]) *
-               bf16_to_f32((unsigned short)xvv[j]);
+    for (int it = 0; it < SLOTS; ++it) {
+      const int i = lane + it * WAVE_SIZE;
+      if (i < vecH) {
+        shortx8 xvv = xv[i], dyvv = dyv[i], wvv = wv[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          dot += bf16_to_f32((unsigned short)dyvv[j]) *
+                 bf16_to_f32((unsigned short)wvv[j]) *
+                 bf16_to_f32((unsigned short)xvv[j]);
+        }
       }
     }
-    for (int i = vecH * 8 + threadIdx.x; i < H; i += BLOCK)
-      dot += bf16_to_f32(dyr[i]) * bf16_to_f32(w[i]) * bf16_to_f32(xr[i]);
-    dot = block_reduce_sum(dot, scratch);
+    dot = wave_reduce_sum(dot);
     const float k = r * r * r * dot / H;
 
-    shortx8* dxv = reinterpret_cast<shortx8*>(dxr);
-    for (int i = threadIdx.x; i < vecH; i += BLOCK) {
-      shortx8 xvv = xv[i], dyvv = dyv[i], wvv = wv[i], o;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float xf = bf16_to_f32((unsigned short)xvv[j]);
-        float dyf = bf16_to_f32((unsigned short)dyvv[j]);
-        float wf = bf16_to_f32((unsigned short)wvv[j]);
-        o[j] = (short)f32_to_bf16(r * wf * dyf - k * xf);
-        dw_local[i * 8 + j] += dyf * xf * r;
+    for (int it = 0; it < SLOTS; ++it) {
+      const int i = lane + it * WAVE_SIZE;
+      if (i < vecH) {
+        shortx8 xvv = xv[i], dyvv = dyv[i], wvv = wv[i], o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xf = bf16_to_f32((unsigned short)xvv[j]);
+          float dyf = bf16_to_f32((unsigned short)dyvv[j]);
+          float wf = bf16_to_f32((unsigned short)wvv[j]);
+          o[j] = (short)f32_to_bf16(r * wf * dyf - k * xf);
+          dwacc[it][j] += dyf * xf * r;
+        }
+        dxv[i] = o;
       }
-      dxv[i] = o;
     }
-    for (int i = vecH * 8 + threadIdx.x; i < H; i += BLOCK) {
-      float xf = bf16_to_f32(xr[i]), dyf = bf16_to_f32(dyr[i]),
-            wf = bf16_to_f32(w[i]);
-      dxr[i] = f32_to_bf16(r * wf * dyf - k * xf);
-      dw_local[i] += dyf * xf * r;
-    }
-    __syncthreads();
   }
-  for (int i = threadIdx.x; i < H; i += BLOCK) {
-    if (dw_local[i] != 0.f) atomicAdd(&dw[i], dw_local[i]);
+
+#pragma unroll
+  for (int it = 0; it < SLOTS; ++it) {
+    const int i = lane + it * WAVE_SIZE;
+    if (i < vecH)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) my_slab[i * 8 + j] = dwacc[it][j];
   }
+  __syncthreads();
+  // Per-block partial row into global scratch (dw here = dw_part, laid out
+  // [nblocks, H]). A direct atomicAdd merge serializes nblocks (~2048)
+  // fp32 atomics per address — measured as the kernel's dominant cost —
+  // and is nondeterministic; the two-phase reduce is neither.
+  for (int i = threadIdx.x; i < H; i += blockDim.x) {
+    float acc = 0.f;
+    for (int sl = 0; sl < nwaves; ++sl) acc += slab[(long)sl * H + i];
+    dw[(long)blockIdx.x * H + i] = acc;
+  }
+}
+
+// Column reduce of dw_part [nblocks, H] -> dw [H]. grid.y chunks the block
+// dimension; the few cross-chunk merges go through atomicAdd (into zeros).
+__global__ void dw_reduce_kernel(const float* __restrict__ dw_part,
+                                 float* __restrict__ dw, int H, int nblocks,
+                                 int rows_per_chunk) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= H) return;
+  const int b0 = blockIdx.y * rows_per_chunk;
+  const int b1 = min(b0 + rows_per_chunk, nblocks);
+  float acc = 0.f;
+  for (int b = b0; b < b1; ++b) acc += dw_part[(long)b * H + i];
+  if (gridDim.y == 1) dw[i] = acc;
+  else atomicAdd(&dw[i], acc);
 }
 
 }  // namespace
@@ -157,19 +195,42 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   const int H = x.size(1);
   auto dx = torch::empty_like(x);
   auto dw = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  torch::Tensor dw_part;  // allocated after nblocks is known
   // Grid-stride over rows: cap blocks so the LDS dw accumulator amortizes.
-  int nblocks = (int)min((long)2048, N);
+  int nblocks = (int)min((long)2048, (N + 3) / 4);
   int rows_per_block = (int)((N + nblocks - 1) / nblocks);
   nblocks = (int)((N + rows_per_block - 1) / rows_per_block);
-  size_t smem = (H + BLOCK / WAVE_SIZE) * sizeof(float);
+  const int nwaves = BLOCK / WAVE_SIZE;
+  size_t smem = (size_t)H * nwaves * sizeof(float);
+  TORCH_CHECK(smem <= 160 * 1024, "rmsnorm_bwd: H too large for LDS slabs");
+  TORCH_CHECK(H % 8 == 0, "rmsnorm_bwd: H must be divisible by 8");
+  const int slots = (H / 8 + WAVE_SIZE - 1) / WAVE_SIZE;
+  dw_part = torch::empty({(long)nblocks, (long)H},
+                         x.options().dtype(torch::kFloat32));
   auto stream = at::cuda::getCurrentHIPStream();
-  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(nblocks), dim3(BLOCK), smem, stream,
-                     (const unsigned short*)dy.data_ptr(),
-                     (const unsigned short*)x.data_ptr(),
-                     (const unsigned short*)w.data_ptr(),
-                     invrms.data_ptr<float>(),
-                     (unsigned short*)dx.data_ptr(),
-                     dw.data_ptr<float>(), H, N, rows_per_block);
+  auto launch = [&](auto* kfn) {
+    hipLaunchKernelGGL(kfn, dim3(nblocks), dim3(BLOCK), smem, stream,
+                       (const unsigned short*)dy.data_ptr(),
+                       (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       invrms.data_ptr<float>(),
+                       (unsigned short*)dx.data_ptr(),
+                       dw_part.data_ptr<float>(), H, N, rows_per_block);
+  };
+  if (slots <= 1) launch(rmsnorm_bwd_kernel<1>);
+  else if (slots <= 2) launch(rmsnorm_bwd_kernel<2>);
+  else if (slots <= 5) launch(rmsnorm_bwd_kernel<5>);
+  else if (slots <= 8) launch(rmsnorm_bwd_kernel<8>);
+  else if (slots <= 16) launch(rmsnorm_bwd_kernel<16>);
+  else TORCH_CHECK(false, "rmsnorm_bwd: H too large (max 8192)");
+  {
+    const int chunks = (int)min((long)32, (long)((nblocks + 63) / 64));
+    const int rpc = (nblocks + chunks - 1) / chunks;
+    hipLaunchKernelGGL(dw_reduce_kernel,
+                       dim3((H + BLOCK - 1) / BLOCK, chunks), dim3(BLOCK), 0,
+                       stream, dw_part.data_ptr<float>(),
+                       dw.data_ptr<float>(), H, nblocks, rpc);
+  }
   HIP_CHECK_KERNEL();
   return {dx, dw};
 }
