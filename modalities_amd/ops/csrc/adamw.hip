@@ -212,9 +212,21 @@ void fused_adamw_masked_devstep(torch::Tensor p, torch::Tensor g,
   auto stream = at::cuda::getCurrentHIPStream();
   static const int wide = []() {
     const char* e = getenv("MODALITIES_AMD_ADAMW_WIDE");
-    return e ? atoi(e) : 2;  // 2x float4 per thread measured fastest
+    // Isolated micro (320M elems): 1x float4 5.6 TB/s, 2x 5.1, 4x 3.9 —
+    // the narrow grid-stride form wins; wall-clock A/B showed no
+    // difference (the optimizer phase partially hides behind step tail).
+    return e ? atoi(e) : 1;
   }();
-  if (wide >= 2) {
+  if (wide >= 4) {
+    hipLaunchKernelGGL(adamw_devstep_wide_kernel<4>,
+                       dim3(grid_for((n4 + 3) / 4, 256)), dim3(256), 0, stream,
+                       p.data_ptr<float>(), g.data_ptr<float>(),
+                       m.data_ptr<float>(), v.data_ptr<float>(),
+                       wd_mask.data_ptr<float>(), step.data_ptr<int>(),
+                       (unsigned short*)bf16_out.data_ptr(), gs_ptr, n4,
+                       (float)lr, (float)beta1, (float)beta2, (float)eps,
+                       (float)wd);
+  } else if (wide >= 2) {
     hipLaunchKernelGGL(adamw_devstep_wide_kernel<2>,
                        dim3(grid_for((n4 + 1) / 2, 256)), dim3(256), 0, stream,
                        p.data_ptr<float>(), g.data_ptr<float>(),
