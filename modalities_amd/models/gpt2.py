@@ -125,6 +125,12 @@ class CausalSelfAttention(nn.Module):
         self.c_proj = TwoStreamLinear(n_embd, n_embd, bias=bias)
         self.resid_dropout = nn.Dropout(dropout)
         self.dropout = dropout
+        if dropout > 0.0 and attention_impl == AttentionImplementation.HIP_FLASH:
+            raise ValueError(
+                "attention dropout is not implemented in the HIP flash "
+                "kernel (K1); use attention_impl=pytorch_flash for "
+                "attention dropout, or set dropout=0 (residual dropout is "
+                "always applied)")
         if use_qk_norm:
             self.q_norm = make_norm(norm_cfg, self.head_dim)
             self.k_norm = make_norm(norm_cfg, self.head_dim)
