@@ -221,7 +221,7 @@ def test_ring_attention_unit(world):
         torch.testing.assert_close(gv_r, v.grad[:, sl], rtol=1e-4, atol=1e-5)
 
 
-def _cp2_dp2_worker(rank, world):
+def _cp2_dp2_worker(rank, world, variant="allgather"):
     """CP=2 x DP_shard=2 over a world-4 mesh with the sharding engine:
     CP-partial grads are summed over the cp group on the engine's flat
     grad shards (after backward_epilogue, before the optimizer), composing
@@ -240,7 +240,8 @@ def _cp2_dp2_worker(rank, world):
     torch.manual_seed(0)
     model = GPT2LLM(tiny_cfg())
     model = get_gpt2_context_parallel_model(model, group=cp.group,
-                                            cp_rank=cp.rank, cp_size=cp.size)
+                                            cp_rank=cp.rank, cp_size=cp.size,
+                                            variant=variant)
     sharded = XGMIShardedModel.from_transformer(
         model, torch.device("cpu"), process_group=dp.group,
         param_dtype=torch.float32, rank=dp.rank, world_size=dp.size)
@@ -268,7 +269,8 @@ def _cp2_dp2_worker(rank, world):
     return losses
 
 
-def test_cp2_dp2_matches_single_process():
+@pytest.mark.parametrize("variant", ["allgather", "ring"])
+def test_cp2_dp2_matches_single_process(variant):
     torch.manual_seed(0)
     ref_model = GPT2LLM(tiny_cfg())
     ref_opt = torch.optim.AdamW(ref_model.parameters(), lr=1e-3,
@@ -284,7 +286,7 @@ def test_cp2_dp2_matches_single_process():
         ref_opt.zero_grad()
         ref_losses.append(loss.item())
     results = run_distributed(_cp2_dp2_worker, world_size=4,
-                              port=find_free_port())
+                              port=find_free_port(), args=(variant,))
     for r in range(4):
         assert results[r] == pytest.approx(ref_losses, rel=2e-4), \
             (results[r], ref_losses)
