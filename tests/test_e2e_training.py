@@ -75,3 +75,37 @@ def test_e2e_losses_decrease(e2e_config):
     import torch
     for u in components.wrapped_model.units:
         assert torch.isfinite(u.master_shard).all()
+
+
+def test_bench_contract_multirank_cpu(tmp_path):
+    """The driver launches bench.py via torch.distributed.run at round end;
+    validate that exact path on CPU/gloo world 2: both ranks run, rank 0
+    prints exactly one JSON line with the contract fields, value aggregates
+    over the whole job."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    from tests.conftest import find_free_port
+    env = dict(os.environ)
+    env.pop("CUDA_VISIBLE_DEVICES", None)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(find_free_port()),
+         os.path.join(repo, "bench.py"), "--gpus", "2", "--steps", "2",
+         "--warmup", "1", "--model", "gpt2-tiny", "--micro-batch", "1",
+         "--seq-len", "64"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=repo)
+    assert r.returncode == 0, r.stderr[-2000:]
+    json_lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1, r.stdout
+    d = json.loads(json_lines[0])
+    assert d["n_gpus"] == 2 and d["steps"] == 2
+    assert d["metric"] == "train_tokens_per_s"
+    assert d["config"]["global_batch"] == 2  # micro_batch 1 x dp 2
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["scaling"] == "weak"
+    assert d["dtype"] == "fp32"  # CPU fallback dtype (bf16 on GPU)
