@@ -135,3 +135,54 @@ class ResultsToDiscSubscriber(MessageSubscriberIF):
     def consume_dict(self, message_dict: dict) -> None:
         if self.enabled:
             self._write(message_dict)
+
+
+class WandBEvaluationResultSubscriber(MessageSubscriberIF):
+    """WandB sink for evaluation results (reference:
+    logging_broker/subscriber_impl/results_subscriber.py:61-117). wandb is
+    imported lazily — this environment has no network, so construction
+    raises a clear error unless wandb is installed and configured."""
+
+    def __init__(self, project: str, experiment_id: str, mode: str = "OFFLINE",
+                 directory: Optional[str] = None,
+                 config_file_path: Optional[str] = None,
+                 global_rank: int = 0):
+        self.enabled = global_rank == 0
+        if not self.enabled:
+            self._run = None
+            return
+        try:
+            import wandb
+        except ImportError as e:
+            raise ImportError(
+                "results_subscriber/wandb requires the wandb package; use "
+                "variant_key 'rich' or 'save_all' in offline environments"
+            ) from e
+        self._run = wandb.init(project=project, name=experiment_id,
+                               mode=mode.lower(), dir=directory)
+        if config_file_path is not None:
+            self._run.save(str(config_file_path))
+
+    def consume_message(self, message: Message) -> None:
+        if not self.enabled or self._run is None:
+            return
+        payload = message.payload
+        if isinstance(payload, EvaluationResultBatch):
+            step = payload.num_train_steps_done
+            prefix = payload.dataloader_tag
+
+            def val(item):
+                v = item.value
+                return v.item() if isinstance(v, torch.Tensor) and \
+                    v.numel() == 1 else v
+
+            logs = {f"{prefix} {k}": val(v) for k, v in payload.losses.items()}
+            logs.update({f"{prefix} {k}": val(v)
+                         for k, v in payload.metrics.items()})
+            logs.update({f"{prefix} {k}": val(v)
+                         for k, v in payload.throughput_metrics.items()})
+            self._run.log(data=logs, step=step)
+
+    def consume_dict(self, message_dict: dict) -> None:
+        if self.enabled and self._run is not None:
+            self._run.log(data=message_dict)

@@ -28,7 +28,7 @@ from modalities_amd.logging_broker.subscribers import (DummyProgressSubscriber,
                                                        DummyResultSubscriber,
                                                        ResultsToDiscSubscriber,
                                                        RichProgressSubscriber,
-                                                       RichResultSubscriber)
+                                                       RichResultSubscriber, WandBEvaluationResultSubscriber)
 from modalities_amd.models.coca import CoCa
 from modalities_amd.models.gpt2 import GPT2LLM, GPT2LLMConfig
 from modalities_amd.models.vision_transformer import VisionTransformer
@@ -216,6 +216,7 @@ COMPONENTS: list[ComponentEntity] = [
     ComponentEntity("results_subscriber", "dummy", DummyResultSubscriber, None),
     ComponentEntity("results_subscriber", "rich", RichResultSubscriber, None),
     ComponentEntity("results_subscriber", "save_to_disc", ResultsToDiscSubscriber, None),
+    ComponentEntity("results_subscriber", "wandb", WandBEvaluationResultSubscriber, None),
     # number conversion (config-time arithmetic components)
     ComponentEntity("number_conversion", "local_num_batches_from_num_samples",
                     NumberConversion.get_local_num_batches_from_num_samples, None),
