@@ -48,6 +48,13 @@ class StepProfile(BaseModel):
     gradient_accumulation_steps: int = 1
     local_train_micro_batch_size: int = 1
     sequence_length: int = 1
+    # Data-parallel degree (dp_replicate * dp_shard). Tokens per train step
+    # scale with the number of DATA-parallel replicas, not world_size: a
+    # composed tp/cp/pp config has world_size > dp_degree and using
+    # world_size would inflate token accounting by the tp*cp*pp factor
+    # (reference main.py:168-173 multiplies by dp_degree). None = derive
+    # from the device mesh at runtime, falling back to world_size.
+    dp_degree: Optional[int] = None
 
 
 class ConsistencyEnforcement(BaseModel):
@@ -81,9 +88,11 @@ class TrainingSettings(BaseModel):
         tokens into the target steps (reference
         instantiation_models.py:110-179)."""
         sp = self.step_profile
+        dp_degree = sp.dp_degree if sp.dp_degree is not None \
+            else self.cuda_env.world_size
         tokens_per_step = (sp.gradient_accumulation_steps
                            * sp.local_train_micro_batch_size * sp.sequence_length
-                           * self.cuda_env.world_size)
+                           * dp_degree)
         tt = self.training_target
         if tt.num_target_steps > 0 and tt.num_target_tokens > 0:
             expected = tokens_per_step * tt.num_target_steps
@@ -143,9 +152,11 @@ class TrainingReportGenerator:
     def get_report(self) -> str:
         s = self.settings
         sp = s.step_profile
+        dp_degree = sp.dp_degree if sp.dp_degree is not None \
+            else s.cuda_env.world_size
         tokens_per_step = (sp.gradient_accumulation_steps
                            * sp.local_train_micro_batch_size * sp.sequence_length
-                           * s.cuda_env.world_size)
+                           * dp_degree)
         lines = [
             "==== Training Report ====",
             f"experiment_id:        {s.experiment_id}",

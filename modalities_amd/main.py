@@ -85,9 +85,16 @@ class Main:
             components.progress_subscriber, components.evaluation_subscriber)
 
         sp = settings.step_profile
+        # tokens/step scale with the DATA-parallel degree, not world_size
+        # (reference main.py:168-173): tp/cp/pp ranks see the same tokens.
+        dp_degree = sp.dp_degree
+        if dp_degree is None and components.device_mesh is not None:
+            dp_degree = components.device_mesh.dp_degree
+        if dp_degree is None:
+            dp_degree = settings.cuda_env.world_size
         global_num_tokens_per_train_step = (
             sp.gradient_accumulation_steps * sp.local_train_micro_batch_size
-            * sp.sequence_length * settings.cuda_env.world_size)
+            * sp.sequence_length * dp_degree)
 
         import torch
         device = (torch.device("cuda", settings.cuda_env.local_rank)

@@ -59,6 +59,17 @@ class ShardedAdamW(torch.optim.Optimizer):
                     start, dtype=torch.int32,
                     device=self.sharded_model.units[0].master_shard.device)
             self._step_dev += 1  # on-device: advances under hipGraph replay
+            # The shared device counter assumes every unit steps every call
+            # (true for full-model backward: all grads reduce each step). A
+            # unit skipping would silently diverge its bias correction from
+            # the per-unit CPU counter persisted in checkpoints (ADVICE r1
+            # #4) — fail loudly instead.
+            fresh = [u.grad_fresh for u in self.sharded_model.units]
+            if any(fresh) and not all(fresh):
+                raise RuntimeError(
+                    "fused AdamW devstep: mixed grad_fresh across units "
+                    f"({sum(fresh)}/{len(fresh)}) — the shared device step "
+                    "counter requires all units to step together")
         gscale = getattr(self.sharded_model, "_pending_grad_scale", None)
         self.sharded_model._pending_grad_scale = None
         for ui, u in enumerate(self.sharded_model.units):

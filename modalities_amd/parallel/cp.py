@@ -209,8 +209,15 @@ def slice_targets_for_cp(targets: torch.Tensor, cp_rank: int, cp_size: int
 
 @torch.no_grad()
 def cp_grad_allreduce_(model, group) -> None:
-    """Sum partial CP gradients across the cp group (call after backward,
-    before the optimizer / DP reduce)."""
+    """Sum partial CP gradients across the cp group (call after backward +
+    backward_epilogue, before the optimizer). With the sharding engine the
+    accumulated grads live on the flat per-unit shards; otherwise on
+    p.grad."""
+    units = getattr(model, "units", None)
+    if units is not None:  # XGMIShardedModel: reduce the flat grad shards
+        for u in units:
+            dist.all_reduce(u.grad_shard, group=group)
+        return
     for p in model.parameters():
         if p.grad is not None:
             dist.all_reduce(p.grad, group=group)
@@ -262,11 +269,13 @@ class _GraphTie(torch.autograd.Function):
 
 def _merge_partials(o_a, lse_a, o_b, lse_b):
     """Online-softmax combination of two partial attentions (o in
-    [B,T,H,D], lse in [B,H,T])."""
+    [B,T,H,D], lse in [B,H,T]). Merged in fp32 (the lse weights are fp32;
+    multiplying bf16 partials by them would silently promote the output)
+    and cast back to the partials' dtype by the caller."""
     lse_new = torch.logaddexp(lse_a, lse_b)
     w_a = torch.exp(lse_a - lse_new).permute(0, 2, 1).unsqueeze(-1)
     w_b = torch.exp(lse_b - lse_new).permute(0, 2, 1).unsqueeze(-1)
-    return o_a * w_a + o_b * w_b, lse_new
+    return o_a.float() * w_a + o_b.float() * w_b, lse_new
 
 
 def _flash_with_lse(q, k, v, q_offset):
@@ -337,5 +346,8 @@ def cp_attention_ring(q_local, k_local, v_local, group, cp_rank: int,
     # Tie the output to the end of the exchange chain: ranks whose queries
     # never attend to a received chunk still must run that exchange's
     # backward (it relays the K/V grads around the ring) — without the tie
-    # their autograd graph skips it and the ring deadlocks.
-    return _GraphTie.apply(o, kv_cur)
+    # their autograd graph skips it and the ring deadlocks. Cast back to
+    # the input dtype: merged partials are fp32 (ADVICE r1 #1 — on the bf16
+    # GPU path every rank that merged >=2 partials would otherwise hand
+    # fp32 activations to the following c_proj GEMM).
+    return _GraphTie.apply(o.to(q_local.dtype), kv_cur)

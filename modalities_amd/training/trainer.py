@@ -94,15 +94,34 @@ class Trainer:
                 return True, loss, grad_norm
             return False, loss, None
         sharded = isinstance(model, XGMIShardedModel)
+        cp_info = self._cp_info_of(model)
+        if cp_info is not None:
+            # CP: logits come back seq-sharded [B, T/cp, V]; slice the
+            # targets identically so the loss shapes match.
+            from modalities_amd.parallel.cp import slice_targets_for_cp
+            group, cp_rank, cp_size = cp_info
+            batch = DatasetBatch(
+                samples=batch.samples,
+                targets={k: slice_targets_for_cp(v, cp_rank, cp_size)
+                         for k, v in batch.targets.items()})
         result_batch = model_predict_batch(model, batch)
         loss = loss_fun(result_batch)
-        (loss / self.gradient_acc_steps).backward()
+        # With CP the local mean covers 1/cp of the tokens: backward through
+        # loss/cp so the cp-SUM of grads equals the full-sequence gradient
+        # (matches tests/test_context_parallelism.py loss normalization).
+        loss_scale = self.gradient_acc_steps * (cp_info[2] if cp_info else 1)
+        (loss / loss_scale).backward()
         if sharded:
             model.backward_epilogue()
 
         grad_norm = None
         step_performed = False
         if (micro_batch_id + 1) % self.gradient_acc_steps == 0:
+            if cp_info is not None:
+                # sum CP-partial grads once per optimizer step (linear in the
+                # accumulated grads), before clipping / the optimizer.
+                from modalities_amd.parallel.cp import cp_grad_allreduce_
+                cp_grad_allreduce_(model, cp_info[0])
             if self.gradient_clipper is not None:
                 grad_norm = self.gradient_clipper(model)
             optimizer.step()
@@ -139,6 +158,7 @@ class Trainer:
                 step_performed, loss, grad_norm = self._train_batch(
                     batch, model, optimizer, scheduler, loss_fun, micro_batch_id)
                 cumulated_losses[0] += loss.item()
+                cumulated_losses[1] = loss.item()  # true last-batch loss
                 cumulated_losses[-1] += 1
                 if grad_norm is not None:
                     grad_norm_window.append(grad_norm)
@@ -179,6 +199,16 @@ class Trainer:
             checkpointing_callback(self.training_progress)
 
     # ------------------------------------------------------------------
+    @staticmethod
+    def _cp_info_of(model):
+        """(group, cp_rank, cp_size) when the model (or the module inside a
+        sharded wrapper) was CP-patched; None otherwise."""
+        for m in (model, getattr(model, "module", None)):
+            info = getattr(m, "_cp_info", None) if m is not None else None
+            if info is not None and info[2] > 1:
+                return info
+        return None
+
     @staticmethod
     def _reset_loss_tracker() -> torch.Tensor:
         # [sum of last-interval losses, last loss, num batches]
@@ -229,8 +259,9 @@ class Trainer:
             dataloader_tag="train",
             num_train_steps_done=self.training_progress.num_seen_steps_total,
             losses={f"{loss_fun.tag} average": ResultItem(avg_loss, 4),
-                    f"{loss_fun.tag} last": ResultItem(cumulated_losses[0]
-                                                       / max(cumulated_losses[-1], 1), 4)},
+                    # last-BATCH loss, rank-averaged (reference publishes the
+                    # true last loss, not the window average again)
+                    f"{loss_fun.tag} last": ResultItem(reduced[1] / world, 4)},
             metrics=metrics, throughput_metrics=throughput)
         self.evaluation_result_publisher.publish_message(
             result, MessageTypes.EVALUATION_RESULT)
