@@ -24,6 +24,7 @@ SOURCES = [
     "adamw.hip",
     "attention_fwd.hip",
     "attention_bwd.hip",
+    "attention_v2.hip",
     "bindings.cpp",
 ]
 

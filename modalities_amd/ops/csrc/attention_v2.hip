@@ -1,0 +1,1114 @@
+// K1 v2: flash attention fwd + bwd for CDNA4/gfx950, GQA-aware.
+//
+// Same math and fragment maps as attention_fwd.hip / attention_bwd.hip
+// (v1), restructured for occupancy and stall behavior after r1 profiling
+// showed both backward kernels allocate the whole 512-register file
+// (1 wave/SIMD: dq 256v+136a, dkdv 256v+256a) and the forward stages
+// synchronously (HBM latency exposed every tile):
+//
+//   - 8-wave (512-thread) workgroups everywhere: 2 waves/SIMD co-hide
+//     LDS/HBM latency (MI355X_MICROARCH "Two waves per SIMD"); register
+//     budget forced <= 256 via __launch_bounds__(512, 2).
+//   - T14 issue-early/write-late staging with ONE register set and ONE
+//     LDS buffer in all three kernels (HBM latency hides under MFMAs).
+//   - ds_read_b64_tr_b16 batches of 8 with a single lgkmcnt drain per
+//     d-block (v1 fwd drained per MFMA pair).
+//   - heavy-first block order for the causal triangle (fwd/dq blocks with
+//     the most kv tiles dispatch first).
+//   - per-wave fully-masked-tile skip (waves outside the causal band for
+//     a tile skip its MFMAs but keep the barrier schedule).
+//   - head_dim 80 (the reference 2.7B shape, gpt2_model.py:446-461)
+//     supported natively: contraction runs D/16 = 5 k-steps; LDS images
+//     keep the 128-column layout (stride DS) with the d >= 80 panels
+//     zeroed once; output stores masked to d < 80.
+//
+// Reference parity: replaces flash-attn 2.8.3 / SDPA
+// (src/modalities/models/gpt2/gpt2_model.py:595-658).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "attn_common.h"
+
+namespace attn2 {
+
+using attnc::c_layout_to_frags;
+using attnc::crow;
+using attnc::lds_addr;
+using attnc::swz;
+using attnc::tr_read_b64;
+
+constexpr int KVBLK = 64;   // fwd/dq kv tile rows
+constexpr int QBLK = 32;    // q rows per wave (fwd/dq) / per LDS tile (dkdv)
+constexpr int NW = 8;       // waves per workgroup
+constexpr int NT = NW * 64; // threads per workgroup
+
+template <int D> struct Geom {
+  static constexpr int DS = (D == 64) ? 64 : 128;  // LDS column stride
+  static constexpr int NDSTEP = D / 16;            // contraction k-steps
+  static constexpr int NDBLK = (D + 31) / 32;      // output 32-col blocks
+  static_assert(D % 16 == 0 && D <= 128, "head_dim must be mult of 16, <=128");
+};
+
+// ---------------------------------------------------------------------------
+// Forward: 8 waves x 32 q rows = 256 q rows per WG; KV tiles of 64 in LDS
+// (K row-major XOR-swizzled; V as [DS/16][64][16] panels for tr reads).
+// Swapped QK^T (S^T = K Q^T) keeps softmax lane-local (guide T12);
+// defer-max rescale skip (T13, THR=8); exp2-domain state.
+template <int D_>
+struct FwdV2 {
+  using G = Geom<D_>;
+  static constexpr int D = D_, DS = G::DS;
+  static constexpr int NDSTEP = G::NDSTEP, NDBLK = G::NDBLK;
+  static constexpr int WG_Q = QBLK * NW;
+
+  struct Smem {
+    unsigned short k[KVBLK * DS];
+    unsigned short vt[(DS / 16) * KVBLK * 16];
+  };
+  static constexpr int NCHS = KVBLK * D / 8;       // 16B chunks per tensor
+  static constexpr int NCH = (NCHS + NT - 1) / NT; // chunks per thread
+
+  struct Stage {
+    shortx8 kk[NCH], vv[NCH];
+  };
+
+  static __device__ void stage_load(Stage& r, const unsigned short* k,
+                                    const unsigned short* v, long kv_base,
+                                    int kv0, int Tkv, int Hkv) {
+    const int tid = threadIdx.x;
+#pragma unroll
+    for (int it = 0; it < NCH; ++it) {
+      const int i = tid + it * NT;
+      if (NCHS % NT != 0 && i >= NCHS) continue;
+      const int r_ = (i * 8) / D, c = (i * 8) % D;
+      if (kv0 + r_ < Tkv) {
+        r.kk[it] = *reinterpret_cast<const shortx8*>(
+            k + kv_base + (long)(kv0 + r_) * Hkv * D + c);
+        r.vv[it] = *reinterpret_cast<const shortx8*>(
+            v + kv_base + (long)(kv0 + r_) * Hkv * D + c);
+      } else {
+#pragma unroll
+        for (int m = 0; m < 8; ++m) { r.kk[it][m] = 0; r.vv[it][m] = 0; }
+      }
+    }
+  }
+
+  static __device__ void stage_write(const Stage& r, Smem* sm) {
+    const int tid = threadIdx.x;
+#pragma unroll
+    for (int it = 0; it < NCH; ++it) {
+      const int i = tid + it * NT;
+      if (NCHS % NT != 0 && i >= NCHS) continue;
+      const int r_ = (i * 8) / D, c = (i * 8) % D;
+      *reinterpret_cast<shortx8*>(&sm->k[r_ * DS + swz(r_, c)]) = r.kk[it];
+      *reinterpret_cast<shortx8*>(
+          &sm->vt[(c >> 4) * (KVBLK * 16) + r_ * 16 + (c & 15)]) = r.vv[it];
+    }
+  }
+
+  static __device__ void run(const unsigned short* __restrict__ q,
+                             const unsigned short* __restrict__ k,
+                             const unsigned short* __restrict__ v,
+                             unsigned short* __restrict__ o,
+                             float* __restrict__ lse, int B, int Tq, int Tkv,
+                             int q_off, int Hq, int Hkv, float scale,
+                             char* smem_raw) {
+    Smem* sm = reinterpret_cast<Smem*>(smem_raw);
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int hi = lane >> 5, ln31 = lane & 31;
+
+    // heavy-first: high q blocks have the most kv tiles under causal
+    const int qi = (int)gridDim.x - 1 - (int)blockIdx.x;
+    const int qblk0 = qi * WG_Q;
+    const int h = blockIdx.y, b = blockIdx.z;
+    const int hkv = h / (Hq / Hkv);
+
+    const long q_base = (((long)b * Tq) * Hq + h) * D;
+    const long kv_base = (((long)b * Tkv) * Hkv + hkv) * D;
+    const int qg = qblk0 + wid * QBLK + ln31;  // this lane's LOCAL q row
+    const int qgl = qg + q_off;                // global vs keys
+    const float scale2 = scale * 1.44269504f;  // exp2 fold
+
+    // zero the d >= D panels of the V image once (D=80: panel 5 feeds the
+    // dblk-2 MFMAs and must read as 0; panels 6,7 cleared too, cheap)
+    if constexpr (D != DS) {
+      for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16; i += NT)
+        sm->vt[(D / 16) * KVBLK * 16 + i] = 0;
+      // first stage_write below is followed by __syncthreads
+    }
+
+    bf16x8 qfrag[NDSTEP];
+    {
+      const unsigned short* qr = q + q_base + (long)qg * Hq * D;
+#pragma unroll
+      for (int s = 0; s < NDSTEP; ++s) {
+        if (qg < Tq) {
+          qfrag[s] = *reinterpret_cast<const bf16x8*>(qr + hi * 8 + 16 * s);
+        } else {
+#pragma unroll
+          for (int m = 0; m < 8; ++m) qfrag[s][m] = (__bf16)0.f;
+        }
+      }
+    }
+
+    float m_run = -INFINITY, l_run = 0.f;
+    floatx16 acc_o[NDBLK];
+#pragma unroll
+    for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc_o[dblk][r] = 0.f;
+
+    const int q_hi_wg = min(qblk0 + WG_Q - 1, Tq - 1) + q_off;
+    const int n_tiles = (min(q_hi_wg, Tkv - 1)) / KVBLK + 1;
+    // this wave's own causal bound (skip fully-masked tiles' compute)
+    const int q_hi_wave = min(qblk0 + wid * QBLK + QBLK - 1, Tq - 1) + q_off;
+    const int q_lo_wave = qblk0 + wid * QBLK + q_off;
+
+    Stage st;
+    stage_load(st, k, v, kv_base, 0, Tkv, Hkv);
+    stage_write(st, sm);
+    __syncthreads();
+    if (n_tiles > 1) stage_load(st, k, v, kv_base, KVBLK, Tkv, Hkv);
+
+    for (int tile = 0; tile < n_tiles; ++tile) {
+      const int kv0 = tile * KVBLK;
+      const bool wave_active = (kv0 <= q_hi_wave);
+
+      if (wave_active) {
+        // ---- S^T = K Q^T ------------------------------------------------
+        floatx16 s0, s1;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) { s0[r] = 0.f; s1[r] = 0.f; }
+#pragma unroll
+        for (int s = 0; s < NDSTEP; ++s) {
+          const int col = hi * 8 + 16 * s;
+          const int r0 = ln31, r1 = ln31 + 32;
+          bf16x8 ka = *reinterpret_cast<const bf16x8*>(&sm->k[r0 * DS + swz(r0, col)]);
+          bf16x8 kb = *reinterpret_cast<const bf16x8*>(&sm->k[r1 * DS + swz(r1, col)]);
+          s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[s], s0, 0, 0, 0);
+          s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb, qfrag[s], s1, 0, 0, 0);
+        }
+
+        // ---- scale + causal mask; per-lane P rows (q = ln31) -----------
+        float p[32];
+        float tmax = -INFINITY;
+        if (kv0 + KVBLK - 1 <= q_lo_wave && kv0 + KVBLK <= Tkv && qg < Tq) {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            float v0 = s0[r] * scale2;
+            float v1 = s1[r] * scale2;
+            p[r] = v0;
+            p[16 + r] = v1;
+            tmax = fmaxf(tmax, fmaxf(v0, v1));
+          }
+        } else {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kg0 = kv0 + crow(r, hi);
+            const int kg1 = kv0 + 32 + crow(r, hi);
+            float v0 = (kg0 <= qgl && kg0 < Tkv) ? s0[r] * scale2 : -INFINITY;
+            float v1 = (kg1 <= qgl && kg1 < Tkv) ? s1[r] * scale2 : -INFINITY;
+            p[r] = v0;
+            p[16 + r] = v1;
+            tmax = fmaxf(tmax, fmaxf(v0, v1));
+          }
+        }
+        tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+
+        // ---- online softmax with defer-max (T13) ------------------------
+        constexpr float RESCALE_THR = 8.0f;
+        const bool defer = __all(tmax - m_run <= RESCALE_THR);
+        if (!defer) {
+          const float m_new = fmaxf(m_run, tmax);
+          const float m_safe = (m_new == -INFINITY) ? 0.f : m_new;
+          const float alpha = (m_run == -INFINITY)
+              ? 0.f : __builtin_amdgcn_exp2f(m_run - m_safe);
+          l_run *= alpha;
+#pragma unroll
+          for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) acc_o[dblk][r] *= alpha;
+          m_run = m_new;
+        }
+        const float m_safe2 = (m_run == -INFINITY) ? 0.f : m_run;
+        float psum = 0.f;
+#pragma unroll
+        for (int r = 0; r < 32; ++r) {
+          p[r] = __builtin_amdgcn_exp2f(p[r] - m_safe2);
+          psum += p[r];
+        }
+        psum += __shfl_xor(psum, 32, 64);
+        l_run += psum;
+
+        // ---- P -> bf16 PV fragments (T12) -------------------------------
+        bf16x8 pfrag[KVBLK / 16];
+        c_layout_to_frags(p, pfrag, KVBLK / 16);
+
+        // ---- O^T += V^T P^T: batched tr reads, one drain per d-block ----
+        {
+          const unsigned vt_base = lds_addr(sm->vt);
+          const int lam = ln31 & 15;
+          const unsigned lane_off = ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
+#pragma unroll
+          for (int dblk = 0; dblk < NDBLK; ++dblk) {
+            const unsigned pan = (unsigned)((dblk * 2 + (ln31 >> 4))
+                                            * (KVBLK * 16)) * 2;
+            uintx2 rv[2 * (KVBLK / 16)];
+#pragma unroll
+            for (int s = 0; s < KVBLK / 16; ++s) {
+              const unsigned a = pan + (16 * s + 8 * hi) * 32 + lane_off;
+              rv[2 * s] = tr_read_b64(vt_base + a);
+              rv[2 * s + 1] = tr_read_b64(vt_base + a + 128);
+            }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+            for (int s = 0; s < KVBLK / 16; ++s) {
+              unsigned w[4] = {rv[2 * s][0], rv[2 * s][1],
+                               rv[2 * s + 1][0], rv[2 * s + 1][1]};
+              bf16x8 va = *reinterpret_cast<bf16x8*>(w);
+              acc_o[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  va, pfrag[s], acc_o[dblk], 0, 0, 0);
+            }
+          }
+        }
+      }  // wave_active
+
+      __syncthreads();  // all waves done reading tile `tile`
+      if (tile + 1 < n_tiles) {
+        stage_write(st, sm);  // tile+1 regs -> LDS
+        if (tile + 2 < n_tiles)
+          stage_load(st, k, v, kv_base, (tile + 2) * KVBLK, Tkv, Hkv);
+        __syncthreads();      // tile+1 ready
+      }
+    }
+
+    // ---- epilogue -------------------------------------------------------
+    const float l_safe = (l_run > 0.f) ? l_run : 1.f;
+    const float inv_l = 1.f / l_safe;
+    if (qg < Tq) {
+      unsigned short* orow = o + q_base + (long)qg * Hq * D;
+#pragma unroll
+      for (int dblk = 0; dblk < NDBLK; ++dblk) {
+#pragma unroll
+        for (int r4 = 0; r4 < 4; ++r4) {
+          const int d = dblk * 32 + crow(r4 * 4, hi);
+          if (D != DS && d >= D) continue;  // head_dim 80 tail mask
+          unsigned int w[2];
+#pragma unroll
+          for (int i = 0; i < 2; ++i) {
+            const int r = r4 * 4 + 2 * i;
+            w[i] = (unsigned int)f32_to_bf16(acc_o[dblk][r] * inv_l) |
+                   ((unsigned int)f32_to_bf16(acc_o[dblk][r + 1] * inv_l) << 16);
+          }
+          *reinterpret_cast<uint2*>(orow + d) = make_uint2(w[0], w[1]);
+        }
+      }
+      if (hi == 0) {
+        lse[(((long)b * Hq) + h) * Tq + qg] =
+            (m_run == -INFINITY) ? 0.f : m_run * 0.6931471806f + __logf(l_safe);
+      }
+    }
+  }
+};
+
+// ---------------------------------------------------------------------------
+// dQ: 8 waves x 32 q rows = 256 q rows per WG; kv tiles of 64 in LDS
+// (K row + V row as S^T/dP^T A-operands; Kt panels as the dQ^T A-operand).
+// The kv tile is processed in two 32-row halves so only one (s, dp)
+// accumulator pair is live at a time (register budget).
+template <int D_>
+struct DQV2 {
+  using G = Geom<D_>;
+  static constexpr int D = D_, DS = G::DS;
+  static constexpr int NDSTEP = G::NDSTEP, NDBLK = G::NDBLK;
+  static constexpr int WG_Q = QBLK * NW;
+
+  struct Smem {
+    unsigned short k[KVBLK * DS];
+    unsigned short v[KVBLK * DS];
+    unsigned short kt[(DS / 16) * KVBLK * 16];
+  };
+  static constexpr int NCHS = KVBLK * D / 8;
+  static constexpr int NCH = (NCHS + NT - 1) / NT;
+
+  struct Stage {
+    shortx8 kk[NCH], vv[NCH];
+  };
+
+  static __device__ void stage_load(Stage& r, const unsigned short* k,
+                                    const unsigned short* v, long kv_base,
+                                    int kv0, int Tkv, int Hkv) {
+    const int tid = threadIdx.x;
+#pragma unroll
+    for (int it = 0; it < NCH; ++it) {
+      const int i = tid + it * NT;
+      if (NCHS % NT != 0 && i >= NCHS) continue;
+      const int r_ = (i * 8) / D, c = (i * 8) % D;
+      if (kv0 + r_ < Tkv) {
+        r.kk[it] = *reinterpret_cast<const shortx8*>(
+            k + kv_base + (long)(kv0 + r_) * Hkv * D + c);
+        r.vv[it] = *reinterpret_cast<const shortx8*>(
+            v + kv_base + (long)(kv0 + r_) * Hkv * D + c);
+      } else {
+#pragma unroll
+        for (int m = 0; m < 8; ++m) { r.kk[it][m] = 0; r.vv[it][m] = 0; }
+      }
+    }
+  }
+
+  static __device__ void stage_write(const Stage& r, Smem* sm) {
+    const int tid = threadIdx.x;
+#pragma unroll
+    for (int it = 0; it < NCH; ++it) {
+      const int i = tid + it * NT;
+      if (NCHS % NT != 0 && i >= NCHS) continue;
+      const int r_ = (i * 8) / D, c = (i * 8) % D;
+      *reinterpret_cast<shortx8*>(&sm->k[r_ * DS + swz(r_, c)]) = r.kk[it];
+      *reinterpret_cast<shortx8*>(&sm->v[r_ * DS + swz(r_, c)]) = r.vv[it];
+      *reinterpret_cast<shortx8*>(
+          &sm->kt[(c >> 4) * (KVBLK * 16) + r_ * 16 + (c & 15)]) = r.kk[it];
+    }
+  }
+
+  static __device__ void run(const unsigned short* q, const unsigned short* k,
+                             const unsigned short* v, const unsigned short* dout,
+                             const float* lse, const float* delta,
+                             unsigned short* dq, int B, int Tq, int Tkv,
+                             int q_off, int Hq, int Hkv, float scale,
+                             char* smem_raw) {
+    Smem* sm = reinterpret_cast<Smem*>(smem_raw);
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int hi = lane >> 5, ln31 = lane & 31;
+    const int h = blockIdx.y, b = blockIdx.z;
+    const int hkv = h / (Hq / Hkv);
+    const int qi = (int)gridDim.x - 1 - (int)blockIdx.x;  // heavy-first
+    const int qblk0 = qi * WG_Q;
+    const int qg = qblk0 + wid * QBLK + ln31;  // LOCAL q row
+    const int qgl = qg + q_off;
+
+    const long q_base = (((long)b * Tq) * Hq + h) * D;
+    const long kv_base = (((long)b * Tkv) * Hkv + hkv) * D;
+
+    if constexpr (D != DS) {
+      for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16; i += NT)
+        sm->kt[(D / 16) * KVBLK * 16 + i] = 0;
+    }
+
+    bf16x8 qfrag[NDSTEP], dofrag[NDSTEP];
+    {
+      const unsigned short* qr = q + q_base + (long)qg * Hq * D;
+      const unsigned short* dor = dout + q_base + (long)qg * Hq * D;
+#pragma unroll
+      for (int s = 0; s < NDSTEP; ++s) {
+        if (qg < Tq) {
+          qfrag[s] = *reinterpret_cast<const bf16x8*>(qr + hi * 8 + 16 * s);
+          dofrag[s] = *reinterpret_cast<const bf16x8*>(dor + hi * 8 + 16 * s);
+        } else {
+#pragma unroll
+          for (int m = 0; m < 8; ++m) { qfrag[s][m] = (__bf16)0.f;
+                                        dofrag[s][m] = (__bf16)0.f; }
+        }
+      }
+    }
+    const float my_lse2 = ((qg < Tq) ? lse[(((long)b * Hq) + h) * Tq + qg]
+                                      : 0.f) * 1.44269504f;
+    const float scale2 = scale * 1.44269504f;
+    const float my_delta = (qg < Tq)
+        ? delta[(((long)b * Hq) + h) * Tq + qg] : 0.f;
+
+    floatx16 acc_dq[NDBLK];
+#pragma unroll
+    for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc_dq[dblk][r] = 0.f;
+
+    const int q_hi_wg = min(qblk0 + WG_Q - 1, Tq - 1) + q_off;
+    const int n_tiles = min(q_hi_wg, Tkv - 1) / KVBLK + 1;
+    const int q_hi_wave = min(qblk0 + wid * QBLK + QBLK - 1, Tq - 1) + q_off;
+    const int q_lo_wave = qblk0 + wid * QBLK + q_off;
+
+    Stage st;
+    stage_load(st, k, v, kv_base, 0, Tkv, Hkv);
+    stage_write(st, sm);
+    __syncthreads();
+    if (n_tiles > 1) stage_load(st, k, v, kv_base, KVBLK, Tkv, Hkv);
+
+    const unsigned kt_base = lds_addr(sm->kt);
+    const int lam = ln31 & 15;
+    const unsigned lane_off = ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
+
+    for (int tile = 0; tile < n_tiles; ++tile) {
+      const int kv0 = tile * KVBLK;
+      const bool wave_active = (kv0 <= q_hi_wave);
+
+      if (wave_active) {
+#pragma unroll
+        for (int half = 0; half < 2; ++half) {
+          const int kv0h = kv0 + 32 * half;
+          // S^T = K Q^T ; dP^T = V dO^T  (col = q = ln31, rows = kv half)
+          floatx16 s_h, dp_h;
+#pragma unroll
+          for (int r = 0; r < 16; ++r) { s_h[r] = 0.f; dp_h[r] = 0.f; }
+#pragma unroll
+          for (int s = 0; s < NDSTEP; ++s) {
+            const int col = hi * 8 + 16 * s;
+            const int r_ = 32 * half + ln31;
+            bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+                &sm->k[r_ * DS + swz(r_, col)]);
+            bf16x8 va = *reinterpret_cast<const bf16x8*>(
+                &sm->v[r_ * DS + swz(r_, col)]);
+            s_h = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[s], s_h,
+                                                          0, 0, 0);
+            dp_h = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[s], dp_h,
+                                                           0, 0, 0);
+          }
+
+          // dS'^T = scale * P .* (dP - delta), P = exp2(scale2*S' - lse2);
+          // written back into dp_h (register-frugal)
+          if (kv0h + 31 <= q_lo_wave && kv0h + 32 <= Tkv && qg < Tq) {
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+              const float p = __builtin_amdgcn_exp2f(s_h[r] * scale2 - my_lse2);
+              dp_h[r] = scale * p * (dp_h[r] - my_delta);
+            }
+          } else {
+#pragma unroll
+            for (int r = 0; r < 16; ++r) {
+              const int kg = kv0h + crow(r, hi);
+              const float p = (kg <= qgl && kg < Tkv && qg < Tq)
+                  ? __builtin_amdgcn_exp2f(s_h[r] * scale2 - my_lse2) : 0.f;
+              dp_h[r] = scale * p * (dp_h[r] - my_delta);
+            }
+          }
+          bf16x8 dsfrag[2];
+          attnc::c16_to_frags(dp_h, dsfrag);
+
+          // dQ^T += Kt dS'^T (contraction over this kv half: k-steps
+          // 2*half, 2*half+1 of the Kt panel image)
+#pragma unroll
+          for (int dblk = 0; dblk < NDBLK; ++dblk) {
+            const unsigned pan = (unsigned)((dblk * 2 + (ln31 >> 4))
+                                            * (KVBLK * 16)) * 2;
+            uintx2 rk[4];
+#pragma unroll
+            for (int s2 = 0; s2 < 2; ++s2) {
+              const unsigned a = pan + (16 * (2 * half + s2) + 8 * hi) * 32
+                  + lane_off;
+              rk[2 * s2] = tr_read_b64(kt_base + a);
+              rk[2 * s2 + 1] = tr_read_b64(kt_base + a + 128);
+            }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+            for (int s2 = 0; s2 < 2; ++s2) {
+              unsigned w[4] = {rk[2 * s2][0], rk[2 * s2][1],
+                               rk[2 * s2 + 1][0], rk[2 * s2 + 1][1]};
+              bf16x8 kta = *reinterpret_cast<bf16x8*>(w);
+              acc_dq[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  kta, dsfrag[s2], acc_dq[dblk], 0, 0, 0);
+            }
+          }
+        }  // half
+      }  // wave_active
+
+      __syncthreads();
+      if (tile + 1 < n_tiles) {
+        stage_write(st, sm);
+        if (tile + 2 < n_tiles)
+          stage_load(st, k, v, kv_base, (tile + 2) * KVBLK, Tkv, Hkv);
+        __syncthreads();
+      }
+    }
+
+    if (qg < Tq) {
+      unsigned short* dqr = dq + q_base + (long)qg * Hq * D;
+#pragma unroll
+      for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+        for (int r4 = 0; r4 < 4; ++r4) {
+          const int d = dblk * 32 + crow(r4 * 4, hi);
+          if (D != DS && d >= D) continue;
+          unsigned int w[2];
+#pragma unroll
+          for (int i = 0; i < 2; ++i) {
+            const int r = r4 * 4 + 2 * i;
+            w[i] = (unsigned int)f32_to_bf16(acc_dq[dblk][r]) |
+                   ((unsigned int)f32_to_bf16(acc_dq[dblk][r + 1]) << 16);
+          }
+          *reinterpret_cast<uint2*>(dqr + d) = make_uint2(w[0], w[1]);
+        }
+    }
+  }
+};
+
+// ---------------------------------------------------------------------------
+// dKdV: MODE-parametrized.
+//   MODE 0 (fused, D=64/80): 8 waves x 32 kv rows = 256 kv rows per WG;
+//     computes dK and dV in one pass (fits the 256-register budget at
+//     D<=80).
+//   MODE 1 (dV-only) / MODE 2 (dK-only), D=128: 4 waves x 32 kv rows =
+//     128 kv rows per WG. The fused D=128 kernel needs ~290 registers
+//     (acc 128 + K/V fragments 64 + softmax/frags/staging) and spilled
+//     ~116 VGPRs at 1 wave/SIMD-equivalent cost; the split recomputes S
+//     (+25% MFMA) but each kernel is register-clean at 2-3 waves/SIMD and
+//     both run concurrently on separate streams.
+// K,V live in REGISTERS (lane owns kv row wave_kv0 + ln31); q tiles of 32
+// in LDS, single buffer, T14 issue-early/write-late. Iteration space =
+// (GQA q-head) x (q tiles).
+//   S [q][kv]  = mfma(A=Q_lds,  B=K_regs)    col = kv = ln31, rows q = crow
+//   dP[q][kv]  = mfma(A=dO_lds, B=V_regs)
+//   dV[kv][d] += mfma(A=frags(P),   B=dOt)   col = d = ln31, rows kv = crow
+//   dK[kv][d] += mfma(A=frags(dS'), B=Qt)
+template <int D_, int MODE>
+struct DKDVV2 {
+  using G = Geom<D_>;
+  static constexpr int D = D_, DS = G::DS;
+  static constexpr int NDSTEP = G::NDSTEP, NDBLK = G::NDBLK;
+  static constexpr bool DO_DV = (MODE != 2), DO_DK = (MODE != 1);
+  static constexpr int NWV = (MODE == 0) ? 8 : 4;   // waves per WG
+  static constexpr int NTV = NWV * 64;
+  static constexpr int WG_KV = QBLK * NWV;
+
+  // manual LDS carve (only the images this mode reads)
+  static constexpr int PANEL = (DS / 16) * QBLK * 16;  // elements
+  static constexpr int O_QROW = 0;
+  static constexpr int O_DOROW = O_QROW + QBLK * DS;
+  static constexpr int O_QT = O_DOROW + (DO_DK ? QBLK * DS : 0);
+  static constexpr int O_DOT = O_QT + (DO_DK ? PANEL : 0);
+  static constexpr int O_END = O_DOT + (DO_DV ? PANEL : 0);
+  static constexpr int O_STATS = (O_END * 2 + 15) / 16 * 16;  // bytes, 16-al
+  static constexpr size_t SMEM_BYTES = O_STATS + (DO_DK ? 64 : 32) * 4;
+
+  static constexpr int NCHS = QBLK * D / 8;
+  static constexpr int NCH = (NCHS + NTV - 1) / NTV;
+
+  struct Stage {
+    shortx8 qq[NCH], dd[NCH];
+    float stat;
+  };
+
+  static __device__ void stage_load(Stage& r, const unsigned short* q,
+                                    const unsigned short* dout, long q_base,
+                                    const float* lse_h, const float* delta_h,
+                                    int q0, int Tq, int Hq) {
+    const int tid = threadIdx.x;
+#pragma unroll
+    for (int it = 0; it < NCH; ++it) {
+      const int i = tid + it * NTV;
+      if (NCHS % NTV != 0 && i >= NCHS) continue;
+      const int row = (i * 8) / D, c = (i * 8) % D;
+      if (q0 + row < Tq) {
+        r.qq[it] = *reinterpret_cast<const shortx8*>(
+            q + q_base + (long)(q0 + row) * Hq * D + c);
+        r.dd[it] = *reinterpret_cast<const shortx8*>(
+            dout + q_base + (long)(q0 + row) * Hq * D + c);
+      } else {
+#pragma unroll
+        for (int m = 0; m < 8; ++m) { r.qq[it][m] = 0; r.dd[it][m] = 0; }
+      }
+    }
+    if (tid < 32) {
+      r.stat = (q0 + tid < Tq) ? lse_h[q0 + tid] * 1.44269504f : 0.f;
+    } else if (DO_DK && tid < 64) {
+      r.stat = (q0 + tid - 32 < Tq) ? delta_h[q0 + tid - 32] : 0.f;
+    }
+  }
+
+  static __device__ void stage_write(const Stage& r, char* smem_raw) {
+    unsigned short* base = reinterpret_cast<unsigned short*>(smem_raw);
+    float* stats = reinterpret_cast<float*>(smem_raw + O_STATS);
+    const int tid = threadIdx.x;
+#pragma unroll
+    for (int it = 0; it < NCH; ++it) {
+      const int i = tid + it * NTV;
+      if (NCHS % NTV != 0 && i >= NCHS) continue;
+      const int row = (i * 8) / D, c = (i * 8) % D;
+      *reinterpret_cast<shortx8*>(
+          &base[O_QROW + row * DS + swz(row, c)]) = r.qq[it];
+      const int pan = (c >> 4) * (QBLK * 16) + row * 16 + (c & 15);
+      if constexpr (DO_DK) {
+        *reinterpret_cast<shortx8*>(
+            &base[O_DOROW + row * DS + swz(row, c)]) = r.dd[it];
+        *reinterpret_cast<shortx8*>(&base[O_QT + pan]) = r.qq[it];
+      }
+      if constexpr (DO_DV) {
+        *reinterpret_cast<shortx8*>(&base[O_DOT + pan]) = r.dd[it];
+      }
+    }
+    if (tid < 32) stats[tid] = r.stat;
+    else if (DO_DK && tid < 64) stats[tid] = r.stat;
+  }
+
+  static __device__ void run(const unsigned short* q, const unsigned short* k,
+                             const unsigned short* v, const unsigned short* dout,
+                             const float* lse, const float* delta,
+                             unsigned short* dk, unsigned short* dv,
+                             int B, int Tq, int Tkv, int q_off, int Hq,
+                             int Hkv, float scale, long dvp, long dvc,
+                             char* smem_raw) {
+    unsigned short* lds = reinterpret_cast<unsigned short*>(smem_raw);
+    float* lse_s = reinterpret_cast<float*>(smem_raw + O_STATS);
+    float* delta_s = lse_s + 32;
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int hi = lane >> 5, ln31 = lane & 31;
+    const int hkv = blockIdx.y, b = blockIdx.z;
+    const int rep = Hq / Hkv;
+    const int kvblk0 = blockIdx.x * WG_KV;  // heavy-first is natural here
+    const int wv_kv0 = kvblk0 + wid * QBLK;
+    const int kvg = wv_kv0 + ln31;          // this lane's kv row
+    const float scale2k = scale * 1.44269504f;
+
+    const long kv_base = (((long)b * Tkv) * Hkv + hkv) * D;
+
+    if constexpr (D != DS) {
+      for (int i = threadIdx.x; i < (DS / 16 - D / 16) * QBLK * 16; i += NTV) {
+        if constexpr (DO_DK) lds[O_QT + (D / 16) * QBLK * 16 + i] = 0;
+        if constexpr (DO_DV) lds[O_DOT + (D / 16) * QBLK * 16 + i] = 0;
+      }
+    }
+
+    bf16x8 kfrag[NDSTEP], vfrag[DO_DK ? NDSTEP : 1];
+    {
+      const unsigned short* kr = k + kv_base + (long)kvg * Hkv * D;
+      const unsigned short* vr = v + kv_base + (long)kvg * Hkv * D;
+#pragma unroll
+      for (int s = 0; s < NDSTEP; ++s) {
+        if (kvg < Tkv) {
+          kfrag[s] = *reinterpret_cast<const bf16x8*>(kr + hi * 8 + 16 * s);
+          if constexpr (DO_DK)
+            vfrag[s] = *reinterpret_cast<const bf16x8*>(vr + hi * 8 + 16 * s);
+        } else {
+#pragma unroll
+          for (int m = 0; m < 8; ++m) {
+            kfrag[s][m] = (__bf16)0.f;
+            if constexpr (DO_DK) vfrag[s][m] = (__bf16)0.f;
+          }
+        }
+      }
+    }
+
+    floatx16 acc_dk[DO_DK ? NDBLK : 1], acc_dv[DO_DV ? NDBLK : 1];
+#pragma unroll
+    for (int dblk = 0; dblk < NDBLK; ++dblk)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        if constexpr (DO_DK) acc_dk[dblk][r] = 0.f;
+        if constexpr (DO_DV) acc_dv[dblk][r] = 0.f;
+      }
+
+    // flattened (GQA q-head) x (q tile) iteration space with incremental
+    // cursors (div/mod-free; see v1)
+    const int first_qtile = max(0, kvblk0 - q_off) / QBLK;
+    const int n_qtiles = (Tq + QBLK - 1) / QBLK;
+    const int tiles_per_head = n_qtiles - first_qtile;
+    const int n_iter = rep * tiles_per_head;
+    const int q0_first = first_qtile * QBLK;
+
+    struct Cursor {
+      int q0;
+      long q_base;
+      const float *lse_h, *delta_h;
+    };
+    auto cursor_init = [&](Cursor& c) {
+      const int hq = hkv * rep;
+      c.q0 = q0_first;
+      c.q_base = (((long)b * Tq) * Hq + hq) * D;
+      c.lse_h = lse + (((long)b * Hq) + hq) * Tq;
+      c.delta_h = delta + (((long)b * Hq) + hq) * Tq;
+    };
+    auto cursor_next = [&](Cursor& c) {
+      c.q0 += QBLK;
+      if (c.q0 >= n_qtiles * QBLK) {
+        c.q0 = q0_first;
+        c.q_base += (long)D;
+        c.lse_h += Tq;
+        c.delta_h += Tq;
+      }
+    };
+
+    Stage st;
+    Cursor cc, cp;
+    cursor_init(cc);
+    cursor_init(cp);
+    stage_load(st, q, dout, cc.q_base, cc.lse_h, cc.delta_h, cc.q0, Tq, Hq);
+    stage_write(st, smem_raw);
+    __syncthreads();
+    if (n_iter > 1) {
+      cursor_next(cp);
+      stage_load(st, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0, Tq, Hq);
+    }
+
+    const unsigned qt_base = lds_addr(lds + O_QT);
+    const unsigned dot_base = lds_addr(lds + O_DOT);
+    const int lam = ln31 & 15;
+    const unsigned lane_off = ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
+
+    for (int t = 0; t < n_iter; ++t) {
+      const int q0 = cc.q0;
+      cursor_next(cc);
+      // a tile strictly below this wave's kv rows is fully masked for it
+      const bool wave_active = (q0 + QBLK - 1 + q_off >= wv_kv0);
+
+      if (wave_active) {
+        // S[q][kv], dP[q][kv] (col = kv = ln31, rows q = crow)
+        floatx16 s_acc, dp_acc;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) { s_acc[r] = 0.f; dp_acc[r] = 0.f; }
+#pragma unroll
+        for (int s = 0; s < NDSTEP; ++s) {
+          const int col = hi * 8 + 16 * s;
+          bf16x8 qa = *reinterpret_cast<const bf16x8*>(
+              &lds[O_QROW + ln31 * DS + swz(ln31, col)]);
+          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[s], s_acc,
+                                                          0, 0, 0);
+          if constexpr (DO_DK) {
+            bf16x8 da = *reinterpret_cast<const bf16x8*>(
+                &lds[O_DOROW + ln31 * DS + swz(ln31, col)]);
+            dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vfrag[s],
+                                                             dp_acc, 0, 0, 0);
+          }
+        }
+
+        // softmax back into the accumulator vectors (register-frugal:
+        // s_acc becomes P; dp_acc becomes dS' when dK is computed)
+        if (q0 + q_off >= wv_kv0 + QBLK && q0 + QBLK <= Tq
+            && wv_kv0 + QBLK <= Tkv) {
+          // interior: every q row of this tile covers every kv row here
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const float lq = lse_s[crow(r, hi)];
+            const float pv = __builtin_amdgcn_exp2f(s_acc[r] * scale2k - lq);
+            s_acc[r] = pv;
+            if constexpr (DO_DK) {
+              const float dl = delta_s[crow(r, hi)];
+              dp_acc[r] = scale * pv * (dp_acc[r] - dl);
+            }
+          }
+        } else {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int qg = q0 + crow(r, hi);  // local
+            const bool ok = (qg < Tq) && (qg + q_off >= kvg) && (kvg < Tkv);
+            const float lq = lse_s[crow(r, hi)];
+            const float pv = ok
+                ? __builtin_amdgcn_exp2f(s_acc[r] * scale2k - lq) : 0.f;
+            s_acc[r] = pv;
+            if constexpr (DO_DK) {
+              const float dl = delta_s[crow(r, hi)];
+              dp_acc[r] = scale * pv * (dp_acc[r] - dl);
+            }
+          }
+        }
+        bf16x8 pfrag[2], dsfrag[2];
+        if constexpr (DO_DV) attnc::c16_to_frags(s_acc, pfrag);
+        if constexpr (DO_DK) attnc::c16_to_frags(dp_acc, dsfrag);
+
+        // dV += P^T dO ; dK += dS'^T Q  (B-operands via tr reads)
+#pragma unroll
+        for (int dblk = 0; dblk < NDBLK; ++dblk) {
+          const unsigned pan = (unsigned)((dblk * 2 + (ln31 >> 4))
+                                          * (QBLK * 16)) * 2;
+          uintx2 rd[4], rq[4];
+#pragma unroll
+          for (int s = 0; s < 2; ++s) {
+            const unsigned a = pan + (16 * s + 8 * hi) * 32 + lane_off;
+            if constexpr (DO_DV) {
+              rd[2 * s] = tr_read_b64(dot_base + a);
+              rd[2 * s + 1] = tr_read_b64(dot_base + a + 128);
+            }
+            if constexpr (DO_DK) {
+              rq[2 * s] = tr_read_b64(qt_base + a);
+              rq[2 * s + 1] = tr_read_b64(qt_base + a + 128);
+            }
+          }
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+          for (int s = 0; s < 2; ++s) {
+            if constexpr (DO_DV) {
+              unsigned wd[4] = {rd[2 * s][0], rd[2 * s][1],
+                                rd[2 * s + 1][0], rd[2 * s + 1][1]};
+              bf16x8 dob = *reinterpret_cast<bf16x8*>(wd);
+              acc_dv[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  pfrag[s], dob, acc_dv[dblk], 0, 0, 0);
+            }
+            if constexpr (DO_DK) {
+              unsigned wq[4] = {rq[2 * s][0], rq[2 * s][1],
+                                rq[2 * s + 1][0], rq[2 * s + 1][1]};
+              bf16x8 qb = *reinterpret_cast<bf16x8*>(wq);
+              acc_dk[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  dsfrag[s], qb, acc_dk[dblk], 0, 0, 0);
+            }
+          }
+        }
+      }  // wave_active
+
+      __syncthreads();
+      if (t + 1 < n_iter) {
+        stage_write(st, smem_raw);
+        if (t + 2 < n_iter) {
+          cursor_next(cp);
+          stage_load(st, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0,
+                     Tq, Hq);
+        }
+        __syncthreads();
+      }
+    }
+
+    // stores: col = ln31 = d within block, rows crow = local kv row
+    const bool all_in = (kvblk0 + WG_KV <= Tkv);
+#pragma unroll
+    for (int dblk = 0; dblk < NDBLK; ++dblk) {
+      const int d = dblk * 32 + ln31;
+      if (D != DS && d >= D) continue;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kvrow = wv_kv0 + crow(r, hi);
+        if (all_in || kvrow < Tkv) {
+          if constexpr (DO_DK) {
+            const long off = kv_base + (long)kvrow * Hkv * D + d;
+            dk[off] = f32_to_bf16(acc_dk[dblk][r]);
+          }
+          if constexpr (DO_DV) {
+            const long dvoff = ((long)b * Tkv + kvrow) * dvp + dvc
+                + (long)hkv * D + d;
+            dv[dvoff] = f32_to_bf16(acc_dv[dblk][r]);
+          }
+        }
+      }
+    }
+  }
+};
+
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(NT, 2) void fwd2_kernel(
+    const unsigned short* q, const unsigned short* k, const unsigned short* v,
+    unsigned short* o, float* lse, int B, int Tq, int Tkv, int q_off, int Hq,
+    int Hkv, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  FwdV2<D>::run(q, k, v, o, lse, B, Tq, Tkv, q_off, Hq, Hkv, scale, smem_raw);
+}
+
+template <int D>
+__global__ __launch_bounds__(NT, 2) void dq2_kernel(
+    const unsigned short* q, const unsigned short* k, const unsigned short* v,
+    const unsigned short* dout, const float* lse, const float* delta,
+    unsigned short* dq, int B, int Tq, int Tkv, int q_off, int Hq, int Hkv,
+    float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  DQV2<D>::run(q, k, v, dout, lse, delta, dq, B, Tq, Tkv, q_off, Hq, Hkv,
+               scale, smem_raw);
+}
+
+template <int D, int MODE>
+__global__ __launch_bounds__((DKDVV2<D, MODE>::NTV), 2)
+void dkdv2_kernel(
+    const unsigned short* q, const unsigned short* k, const unsigned short* v,
+    const unsigned short* dout, const float* lse, const float* delta,
+    unsigned short* dk, unsigned short* dv, int B, int Tq, int Tkv, int q_off,
+    int Hq, int Hkv, float scale, long dvp, long dvc) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  DKDVV2<D, MODE>::run(q, k, v, dout, lse, delta, dk, dv, B, Tq, Tkv, q_off,
+                       Hq, Hkv, scale, dvp, dvc, smem_raw);
+}
+
+// delta preprocess: delta[b,h,t] = rowsum(dO * O) (local copy — device
+// code does not link across TUs without rdc)
+__global__ void delta2_kernel(const unsigned short* __restrict__ dout,
+                              const unsigned short* __restrict__ o,
+                              float* __restrict__ delta,
+                              long total_rows, int T, int H, int D) {
+  const long row = (blockIdx.x * (long)blockDim.x + threadIdx.x) / WAVE_SIZE;
+  const int lane = threadIdx.x & 63;
+  if (row >= total_rows) return;  // row = ((b*T)+t)*H + h
+  const long base = row * D;
+  float acc = 0.f;
+  for (int d = lane * 2; d < D; d += WAVE_SIZE * 2) {
+    unsigned int du = *reinterpret_cast<const unsigned int*>(dout + base + d);
+    unsigned int ou = *reinterpret_cast<const unsigned int*>(o + base + d);
+    acc += bf16_to_f32((unsigned short)(du & 0xffff)) *
+           bf16_to_f32((unsigned short)(ou & 0xffff));
+    acc += bf16_to_f32((unsigned short)(du >> 16)) *
+           bf16_to_f32((unsigned short)(ou >> 16));
+  }
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) {
+    const int h = (int)(row % H);
+    const long bt = row / H;
+    const int t = (int)(bt % T);
+    const long b = bt / T;
+    delta[((b * H) + h) * T + t] = acc;
+  }
+}
+
+}  // namespace attn2
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> attn_fwd2(torch::Tensor q, torch::Tensor k,
+                                     torch::Tensor v, bool causal,
+                                     long q_offset) {
+  TORCH_CHECK(causal, "attn_fwd2: only causal attention is implemented");
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 4
+              && q.is_contiguous(), "q must be contiguous bf16 [B,T,Hq,D]");
+  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  const int B = q.size(0), Tq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Tkv = k.size(1), Hkv = k.size(2);
+  TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  TORCH_CHECK(D == 64 || D == 80 || D == 128,
+              "attn_fwd2: head_dim must be 64, 80 or 128");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, Hq, Tq}, q.options().dtype(torch::kFloat32));
+  const float scale = 1.0f / sqrtf((float)D);
+  const int n_qblk = (Tq + attn2::QBLK * attn2::NW - 1) / (attn2::QBLK * attn2::NW);
+  dim3 grid(n_qblk, Hq, B);
+  auto stream = at::cuda::getCurrentHIPStream();
+  auto launch = [&](auto kfn, size_t smem) {
+    hipLaunchKernelGGL(kfn, grid, dim3(attn2::NT), smem, stream,
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
+                       B, Tq, Tkv, (int)q_offset, Hq, Hkv, scale);
+  };
+  if (D == 128) launch(attn2::fwd2_kernel<128>, sizeof(typename attn2::FwdV2<128>::Smem));
+  else if (D == 80) launch(attn2::fwd2_kernel<80>, sizeof(typename attn2::FwdV2<80>::Smem));
+  else launch(attn2::fwd2_kernel<64>, sizeof(typename attn2::FwdV2<64>::Smem));
+  HIP_CHECK_KERNEL();
+  return {o, lse};
+}
+
+static std::vector<torch::Tensor> attn_bwd2_impl(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor o, torch::Tensor lse, bool causal, long q_offset,
+    torch::Tensor dv, long dvp, long dvc) {
+  TORCH_CHECK(causal, "attn_bwd2: only causal attention is implemented");
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
+  const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Tkv = k.size(1), Hkv = k.size(2);
+  const int q_off = (int)q_offset;
+  const float scale = 1.0f / sqrtf((float)D);
+  TORCH_CHECK(D == 64 || D == 80 || D == 128,
+              "attn_bwd2: head_dim must be 64, 80 or 128");
+  auto stream = at::cuda::getCurrentHIPStream();
+
+  auto delta = torch::empty({B, Hq, T}, q.options().dtype(torch::kFloat32));
+  {
+    long total_rows = (long)B * T * Hq;
+    long threads = total_rows * WAVE_SIZE;
+    int block = 256;
+    long grid = (threads + block - 1) / block;
+    hipLaunchKernelGGL(attn2::delta2_kernel, dim3((unsigned)grid), dim3(block),
+                       0, stream, (const unsigned short*)dout.data_ptr(),
+                       (const unsigned short*)o.data_ptr(),
+                       delta.data_ptr<float>(), total_rows, T, Hq, D);
+    HIP_CHECK_KERNEL();
+  }
+
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+
+  // dq, dv and dk are independent given delta: run on separate streams so
+  // their tail waves overlap (raw fork/join events keep this
+  // hipGraph-capturable)
+  static hipStream_t side = nullptr, side2 = nullptr;
+  static hipEvent_t ev_fork = nullptr, ev_join = nullptr, ev_join2 = nullptr;
+  if (side == nullptr) {
+    hipStreamCreateWithFlags(&side, hipStreamNonBlocking);
+    hipStreamCreateWithFlags(&side2, hipStreamNonBlocking);
+    hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming);
+    hipEventCreateWithFlags(&ev_join, hipEventDisableTiming);
+    hipEventCreateWithFlags(&ev_join2, hipEventDisableTiming);
+  }
+  hipEventRecord(ev_fork, stream);
+  hipStreamWaitEvent(side, ev_fork, 0);
+
+  const int WG = attn2::QBLK * attn2::NW;  // 256 q rows per dq workgroup
+  const int n_qblk = (T + WG - 1) / WG;
+  auto launch_dq = [&](auto dqk, size_t smem_dq) {
+    hipLaunchKernelGGL(dqk, dim3(n_qblk, Hq, B), dim3(attn2::NT), smem_dq,
+                       stream, (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (unsigned short*)dq.data_ptr(), B, T, Tkv, q_off, Hq,
+                       Hkv, scale);
+    HIP_CHECK_KERNEL();
+  };
+  auto launch_kv = [&](auto kfn, int wg_kv, int nthreads, size_t smem,
+                       hipStream_t st) {
+    const int n_kvblk = (Tkv + wg_kv - 1) / wg_kv;
+    hipLaunchKernelGGL(kfn, dim3(n_kvblk, Hkv, B), dim3(nthreads), smem,
+                       st, (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       (unsigned short*)dk.data_ptr(),
+                       (unsigned short*)dv.data_ptr(), B, T, Tkv, q_off, Hq,
+                       Hkv, scale, dvp, dvc);
+    HIP_CHECK_KERNEL();
+  };
+  bool used_side2 = false;
+  if (D == 128) {
+    // fused dkdv needs ~290 registers at D=128: run the register-clean
+    // dV-only / dK-only kernels concurrently instead
+    using DV = attn2::DKDVV2<128, 1>;
+    using DK = attn2::DKDVV2<128, 2>;
+    launch_dq(attn2::dq2_kernel<128>, sizeof(typename attn2::DQV2<128>::Smem));
+    hipStreamWaitEvent(side2, ev_fork, 0);
+    used_side2 = true;
+    launch_kv(attn2::dkdv2_kernel<128, 1>, DV::WG_KV, DV::NTV, DV::SMEM_BYTES,
+              side);
+    launch_kv(attn2::dkdv2_kernel<128, 2>, DK::WG_KV, DK::NTV, DK::SMEM_BYTES,
+              side2);
+  } else if (D == 80) {
+    using F = attn2::DKDVV2<80, 0>;
+    launch_dq(attn2::dq2_kernel<80>, sizeof(typename attn2::DQV2<80>::Smem));
+    launch_kv(attn2::dkdv2_kernel<80, 0>, F::WG_KV, F::NTV, F::SMEM_BYTES,
+              side);
+  } else {
+    using F = attn2::DKDVV2<64, 0>;
+    launch_dq(attn2::dq2_kernel<64>, sizeof(typename attn2::DQV2<64>::Smem));
+    launch_kv(attn2::dkdv2_kernel<64, 0>, F::WG_KV, F::NTV, F::SMEM_BYTES,
+              side);
+  }
+  hipEventRecord(ev_join, side);
+  hipStreamWaitEvent(stream, ev_join, 0);
+  if (used_side2) {
+    hipEventRecord(ev_join2, side2);
+    hipStreamWaitEvent(stream, ev_join2, 0);
+  }
+  return {dq, dk, dv};
+}
+
+std::vector<torch::Tensor> attn_bwd2(torch::Tensor dout, torch::Tensor q,
+                                     torch::Tensor k, torch::Tensor v,
+                                     torch::Tensor o, torch::Tensor lse,
+                                     bool causal, long q_offset) {
+  auto dv = torch::empty_like(v);
+  const long Hkv = v.size(2), D = v.size(3);
+  auto r = attn_bwd2_impl(dout, q, k, v, o, lse, causal, q_offset, dv,
+                          Hkv * D, 0);
+  return {r[0], r[1], dv};
+}
+
+std::vector<torch::Tensor> attn_bwd2_qkvjoint(torch::Tensor dout,
+                                              torch::Tensor q, torch::Tensor k,
+                                              torch::Tensor v, torch::Tensor o,
+                                              torch::Tensor lse, bool causal,
+                                              long q_offset, torch::Tensor dqkv,
+                                              long dv_col_off) {
+  TORCH_CHECK(dqkv.is_cuda() && dqkv.dtype() == torch::kBFloat16
+              && dqkv.dim() == 3 && dqkv.is_contiguous());
+  const long Ctot = dqkv.size(2), Hkv = v.size(2), D = v.size(3);
+  TORCH_CHECK(dqkv.size(0) == v.size(0) && dqkv.size(1) == v.size(1)
+              && dv_col_off + Hkv * D <= Ctot);
+  auto r = attn_bwd2_impl(dout, q, k, v, o, lse, causal, q_offset, dqkv,
+                          Ctot, dv_col_off);
+  return {r[0], r[1]};
+}

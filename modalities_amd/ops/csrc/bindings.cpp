@@ -60,6 +60,26 @@ std::vector<torch::Tensor> attn_bwd_dkdv_ablate(torch::Tensor dout,
                                                 torch::Tensor v,
                                                 torch::Tensor lse,
                                                 torch::Tensor delta, long mode);
+std::vector<torch::Tensor> attn_fwd2(torch::Tensor q, torch::Tensor k,
+                                     torch::Tensor v, bool causal,
+                                     long q_offset);
+std::vector<torch::Tensor> attn_bwd2(torch::Tensor dout, torch::Tensor q,
+                                     torch::Tensor k, torch::Tensor v,
+                                     torch::Tensor o, torch::Tensor lse,
+                                     bool causal, long q_offset);
+std::vector<torch::Tensor> attn_bwd2_qkvjoint(torch::Tensor dout,
+                                              torch::Tensor q, torch::Tensor k,
+                                              torch::Tensor v, torch::Tensor o,
+                                              torch::Tensor lse, bool causal,
+                                              long q_offset, torch::Tensor dqkv,
+                                              long dv_col_off);
+
+// v1/v2 dispatch: v2 (8-wave, 2 waves/SIMD) is the perf path; v1 kept for
+// A/B and as a fallback switch. Flip with set_attn_impl / MA_ATTN_IMPL.
+static int g_attn_impl = [] {
+  const char* e = getenv("MA_ATTN_IMPL");
+  return e ? atoi(e) : 2;
+}();
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (K5)");
@@ -80,19 +100,51 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "AdamW with device-resident step counter (hipGraph-safe)");
   m.def("multi_tensor_sqsum", &multi_tensor_sqsum, "sum of squares (K10)");
   m.def("multi_tensor_scale", &multi_tensor_scale, "in-place scale (K10)");
-  m.def("attn_fwd", &attn_fwd, "flash attention forward (K1)",
-        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"),
-        py::arg("q_offset") = 0);
-  m.def("attn_bwd", &attn_bwd, "flash attention backward (K1)",
-        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
-        py::arg("o"), py::arg("lse"), py::arg("causal"),
-        py::arg("q_offset") = 0);
+  m.def("attn_fwd",
+        [](torch::Tensor q, torch::Tensor k, torch::Tensor v, bool causal,
+           long q_offset) {
+          const long D = q.size(3);
+          return (g_attn_impl == 2 || D == 80)
+              ? attn_fwd2(q, k, v, causal, q_offset)
+              : attn_fwd(q, k, v, causal, q_offset);
+        },
+        "flash attention forward (K1)", py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("causal"), py::arg("q_offset") = 0);
+  m.def("attn_bwd",
+        [](torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+           torch::Tensor v, torch::Tensor o, torch::Tensor lse, bool causal,
+           long q_offset) {
+          const long D = q.size(3);
+          return (g_attn_impl == 2 || D == 80)
+              ? attn_bwd2(dout, q, k, v, o, lse, causal, q_offset)
+              : attn_bwd(dout, q, k, v, o, lse, causal, q_offset);
+        },
+        "flash attention backward (K1)", py::arg("dout"), py::arg("q"),
+        py::arg("k"), py::arg("v"), py::arg("o"), py::arg("lse"),
+        py::arg("causal"), py::arg("q_offset") = 0);
+  m.def("set_attn_impl", [](long v) { g_attn_impl = (int)v; },
+        "select the K1 kernel generation (1 or 2) for A/B");
+  m.def("get_attn_impl", []() { return (long)g_attn_impl; });
+  m.def("attn_fwd_v1", &attn_fwd, "K1 v1 forward (direct, for A/B)");
+  m.def("attn_bwd_v1", &attn_bwd, "K1 v1 backward (direct, for A/B)");
+  m.def("attn_fwd_v2", &attn_fwd2, "K1 v2 forward (direct, for A/B)");
+  m.def("attn_bwd_v2", &attn_bwd2, "K1 v2 backward (direct, for A/B)");
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16,
         "MFMA fragment-layout probe (verification)");
   m.def("attn_fwd_ablate", &attn_fwd_ablate,
         "attention fwd cost-attribution ablation (timing only)");
-  m.def("attn_bwd_qkvjoint", &attn_bwd_qkvjoint,
+  m.def("attn_bwd_qkvjoint",
+        [](torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+           torch::Tensor v, torch::Tensor o, torch::Tensor lse, bool causal,
+           long q_offset, torch::Tensor dqkv, long dv_col_off) {
+          const long D = q.size(3);
+          return (g_attn_impl == 2 || D == 80)
+              ? attn_bwd2_qkvjoint(dout, q, k, v, o, lse, causal, q_offset,
+                                   dqkv, dv_col_off)
+              : attn_bwd_qkvjoint(dout, q, k, v, o, lse, causal, q_offset,
+                                  dqkv, dv_col_off);
+        },
         "attention backward writing dv into a fused-QKV grad buffer");
   m.def("attn_bwd_dkdv_ablate", &attn_bwd_dkdv_ablate,
         "attention bwd dkdv cost-attribution ablation (timing only)");

@@ -126,19 +126,25 @@ def ref_attention(q, k, v, causal=True):
     return (att @ vf).permute(0, 2, 1, 3)
 
 
+@pytest.mark.parametrize("impl", [1, 2])
 @pytest.mark.parametrize("B,T,Hq,Hkv,D", [
     (2, 128, 4, 4, 128),
     (2, 256, 4, 2, 128),
     (1, 200, 4, 1, 128),   # T not a multiple of 128
     (2, 256, 4, 2, 64),
     (1, 4096, 2, 2, 128),  # long sequence
+    (2, 256, 4, 2, 80),    # the reference 2.7B head_dim (v2 only)
+    (1, 300, 4, 4, 80),
 ])
-def test_attn_fwd(B, T, Hq, Hkv, D):
+def test_attn_fwd(B, T, Hq, Hkv, D, impl):
+    if impl == 1 and D == 80:
+        pytest.skip("head_dim 80 is v2-only")
     torch.manual_seed(5)
     q = bf(torch.randn(B, T, Hq, D, device=DEV))
     k = bf(torch.randn(B, T, Hkv, D, device=DEV))
     v = bf(torch.randn(B, T, Hkv, D, device=DEV))
-    o, lse = EXT.attn_fwd(q, k, v, True)
+    fwd = EXT.attn_fwd_v1 if impl == 1 else EXT.attn_fwd_v2
+    o, lse = fwd(q, k, v, True, 0)
     ref = ref_attention(q, k, v)
     torch.testing.assert_close(o.float(), ref, rtol=3e-2, atol=3e-2)
     # lse check
@@ -151,21 +157,29 @@ def test_attn_fwd(B, T, Hq, Hkv, D):
     torch.testing.assert_close(lse, ref_lse, rtol=2e-2, atol=2e-2)
 
 
+@pytest.mark.parametrize("impl", [1, 2])
 @pytest.mark.parametrize("B,T,Hq,Hkv,D", [
     (2, 128, 4, 4, 128),
     (2, 256, 4, 2, 128),
     (1, 200, 4, 1, 128),
     (2, 256, 4, 2, 64),
+    (2, 256, 4, 2, 80),
+    (1, 300, 4, 4, 80),
+    (1, 1024, 2, 2, 128),
 ])
-def test_attn_bwd(B, T, Hq, Hkv, D):
+def test_attn_bwd(B, T, Hq, Hkv, D, impl):
+    if impl == 1 and D == 80:
+        pytest.skip("head_dim 80 is v2-only")
     torch.manual_seed(6)
     q = bf(torch.randn(B, T, Hq, D, device=DEV))
     k = bf(torch.randn(B, T, Hkv, D, device=DEV))
     v = bf(torch.randn(B, T, Hkv, D, device=DEV))
     do = bf(torch.randn(B, T, Hq, D, device=DEV))
 
-    o, lse = EXT.attn_fwd(q, k, v, True)
-    dq, dk, dv = EXT.attn_bwd(do, q, k, v, o, lse, True)
+    fwd = EXT.attn_fwd_v1 if impl == 1 else EXT.attn_fwd_v2
+    bwd = EXT.attn_bwd_v1 if impl == 1 else EXT.attn_bwd_v2
+    o, lse = fwd(q, k, v, True, 0)
+    dq, dk, dv = bwd(do, q, k, v, o, lse, True, 0)
 
     qf = q.float().requires_grad_(True)
     kf = k.float().requires_grad_(True)
