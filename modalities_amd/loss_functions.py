@@ -29,6 +29,9 @@ class CLMCrossEntropyLoss(Loss):
         super().__init__(tag)
         self.target_key = target_key
         self.prediction_key = prediction_key
+        # set by the trainer/evaluator when the model's lm_head is
+        # vocab-sharded (TP): (group, tp_rank, tp_size, vocab)
+        self.tp_vocab_info = None
 
     def forward(self, forward_batch) -> torch.Tensor:
         # Also directly callable as (logits, targets) for pipeline schedules
@@ -38,6 +41,12 @@ class CLMCrossEntropyLoss(Loss):
             logits = forward_batch.get_predictions(self.prediction_key)
         else:
             logits, labels = forward_batch
+        if self.tp_vocab_info is not None:
+            from modalities_amd.parallel.tp import vocab_parallel_cross_entropy
+            group, tp_rank, tp_size, vocab = self.tp_vocab_info
+            return vocab_parallel_cross_entropy(
+                logits.view(-1, logits.shape[-1]), labels.reshape(-1),
+                group, tp_rank, tp_size, vocab, ignore_index=-100)
         return fused_cross_entropy(logits.view(-1, logits.shape[-1]),
                                    labels.reshape(-1), ignore_index=-100)
 

@@ -42,7 +42,12 @@ constexpr int NW = 8;       // waves per workgroup
 constexpr int NT = NW * 64; // threads per workgroup
 
 template <int D> struct Geom {
-  static constexpr int DS = (D == 64) ? 64 : 128;  // LDS column stride
+  static constexpr int DS = (D == 64) ? 64 : 128;  // logical column count
+  // Row pitch padded by one b128 access width (guide G4): row-major reads
+  // at (row = lane, col fixed) land on distinct banks with AFFINE
+  // addresses — the compiler folds the whole k-loop into ds offset
+  // immediates (the XOR swizzle costs ~5 address VALU per read).
+  static constexpr int DP = DS + 8;                // padded LDS row pitch
   static constexpr int NDSTEP = D / 16;            // contraction k-steps
   static constexpr int NDBLK = (D + 31) / 32;      // output 32-col blocks
   static_assert(D % 16 == 0 && D <= 128, "head_dim must be mult of 16, <=128");
@@ -56,12 +61,12 @@ template <int D> struct Geom {
 template <int D_>
 struct FwdV2 {
   using G = Geom<D_>;
-  static constexpr int D = D_, DS = G::DS;
+  static constexpr int D = D_, DS = G::DS, DP = G::DP;
   static constexpr int NDSTEP = G::NDSTEP, NDBLK = G::NDBLK;
   static constexpr int WG_Q = QBLK * NW;
 
   struct Smem {
-    unsigned short k[KVBLK * DS];
+    unsigned short k[KVBLK * DP];
     unsigned short vt[(DS / 16) * KVBLK * 16];
   };
   static constexpr int NCHS = KVBLK * D / 8;       // 16B chunks per tensor
@@ -99,7 +104,7 @@ struct FwdV2 {
       const int i = tid + it * NT;
       if (NCHS % NT != 0 && i >= NCHS) continue;
       const int r_ = (i * 8) / D, c = (i * 8) % D;
-      *reinterpret_cast<shortx8*>(&sm->k[r_ * DS + swz(r_, c)]) = r.kk[it];
+      *reinterpret_cast<shortx8*>(&sm->k[r_ * DP + c]) = r.kk[it];
       *reinterpret_cast<shortx8*>(
           &sm->vt[(c >> 4) * (KVBLK * 16) + r_ * 16 + (c & 15)]) = r.vv[it];
     }
@@ -170,6 +175,12 @@ struct FwdV2 {
     __syncthreads();
     if (n_tiles > 1) stage_load(st, k, v, kv_base, KVBLK, Tkv, Hkv);
 
+    // tile-invariant transpose-read base (single LDS buffer)
+    const int lam = ln31 & 15;
+    const unsigned trb = lds_addr(sm->vt)
+        + (unsigned)(ln31 >> 4) * (KVBLK * 16 * 2) + (unsigned)hi * 256
+        + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
+
     for (int tile = 0; tile < n_tiles; ++tile) {
       const int kv0 = tile * KVBLK;
       const bool wave_active = (kv0 <= q_hi_wave);
@@ -183,8 +194,8 @@ struct FwdV2 {
         for (int s = 0; s < NDSTEP; ++s) {
           const int col = hi * 8 + 16 * s;
           const int r0 = ln31, r1 = ln31 + 32;
-          bf16x8 ka = *reinterpret_cast<const bf16x8*>(&sm->k[r0 * DS + swz(r0, col)]);
-          bf16x8 kb = *reinterpret_cast<const bf16x8*>(&sm->k[r1 * DS + swz(r1, col)]);
+          bf16x8 ka = *reinterpret_cast<const bf16x8*>(&sm->k[r0 * DP + col]);
+          bf16x8 kb = *reinterpret_cast<const bf16x8*>(&sm->k[r1 * DP + col]);
           s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[s], s0, 0, 0, 0);
           s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb, qfrag[s], s1, 0, 0, 0);
         }
@@ -244,34 +255,28 @@ struct FwdV2 {
         bf16x8 pfrag[KVBLK / 16];
         c_layout_to_frags(p, pfrag, KVBLK / 16);
 
-        // ---- O^T += V^T P^T: batched tr reads, one drain per d-block ----
-        {
-          const unsigned vt_base = lds_addr(sm->vt);
-          const int lam = ln31 & 15;
-          const unsigned lane_off = ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
+        // ---- O^T += V^T P^T: batched tr reads, one drain per d-block;
+        // one tile-invariant base VGPR + ds offset immediates ----
+        attnc::static_for<NDBLK>([&](auto dblk_) {
+          constexpr int dblk = decltype(dblk_)::value;
+          uintx2 rv[2 * (KVBLK / 16)];
+          attnc::static_for<KVBLK / 16>([&](auto s_) {
+            constexpr int s = decltype(s_)::value;
+            constexpr unsigned a = dblk * 2 * (KVBLK * 16) * 2 + s * 512;
+            rv[2 * s] = attnc::tr_read_b64_off<a>(trb);
+            rv[2 * s + 1] = attnc::tr_read_b64_off<a + 128>(trb);
+          });
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-          for (int dblk = 0; dblk < NDBLK; ++dblk) {
-            const unsigned pan = (unsigned)((dblk * 2 + (ln31 >> 4))
-                                            * (KVBLK * 16)) * 2;
-            uintx2 rv[2 * (KVBLK / 16)];
-#pragma unroll
-            for (int s = 0; s < KVBLK / 16; ++s) {
-              const unsigned a = pan + (16 * s + 8 * hi) * 32 + lane_off;
-              rv[2 * s] = tr_read_b64(vt_base + a);
-              rv[2 * s + 1] = tr_read_b64(vt_base + a + 128);
-            }
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-            __builtin_amdgcn_sched_barrier(0);
-#pragma unroll
-            for (int s = 0; s < KVBLK / 16; ++s) {
-              unsigned w[4] = {rv[2 * s][0], rv[2 * s][1],
-                               rv[2 * s + 1][0], rv[2 * s + 1][1]};
-              bf16x8 va = *reinterpret_cast<bf16x8*>(w);
-              acc_o[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                  va, pfrag[s], acc_o[dblk], 0, 0, 0);
-            }
+          for (int s = 0; s < KVBLK / 16; ++s) {
+            unsigned w[4] = {rv[2 * s][0], rv[2 * s][1],
+                             rv[2 * s + 1][0], rv[2 * s + 1][1]};
+            bf16x8 va = *reinterpret_cast<bf16x8*>(w);
+            acc_o[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                va, pfrag[s], acc_o[dblk], 0, 0, 0);
           }
-        }
+        });
       }  // wave_active
 
       __syncthreads();  // all waves done reading tile `tile`
@@ -298,8 +303,8 @@ struct FwdV2 {
 #pragma unroll
           for (int i = 0; i < 2; ++i) {
             const int r = r4 * 4 + 2 * i;
-            w[i] = (unsigned int)f32_to_bf16(acc_o[dblk][r] * inv_l) |
-                   ((unsigned int)f32_to_bf16(acc_o[dblk][r + 1] * inv_l) << 16);
+            w[i] = attnc::cvt_pk_bf16(acc_o[dblk][r] * inv_l,
+                                      acc_o[dblk][r + 1] * inv_l);
           }
           *reinterpret_cast<uint2*>(orow + d) = make_uint2(w[0], w[1]);
         }
@@ -320,13 +325,13 @@ struct FwdV2 {
 template <int D_>
 struct DQV2 {
   using G = Geom<D_>;
-  static constexpr int D = D_, DS = G::DS;
+  static constexpr int D = D_, DS = G::DS, DP = G::DP;
   static constexpr int NDSTEP = G::NDSTEP, NDBLK = G::NDBLK;
   static constexpr int WG_Q = QBLK * NW;
 
   struct Smem {
-    unsigned short k[KVBLK * DS];
-    unsigned short v[KVBLK * DS];
+    unsigned short k[KVBLK * DP];
+    unsigned short v[KVBLK * DP];
     unsigned short kt[(DS / 16) * KVBLK * 16];
   };
   static constexpr int NCHS = KVBLK * D / 8;
@@ -364,8 +369,8 @@ struct DQV2 {
       const int i = tid + it * NT;
       if (NCHS % NT != 0 && i >= NCHS) continue;
       const int r_ = (i * 8) / D, c = (i * 8) % D;
-      *reinterpret_cast<shortx8*>(&sm->k[r_ * DS + swz(r_, c)]) = r.kk[it];
-      *reinterpret_cast<shortx8*>(&sm->v[r_ * DS + swz(r_, c)]) = r.vv[it];
+      *reinterpret_cast<shortx8*>(&sm->k[r_ * DP + c]) = r.kk[it];
+      *reinterpret_cast<shortx8*>(&sm->v[r_ * DP + c]) = r.vv[it];
       *reinterpret_cast<shortx8*>(
           &sm->kt[(c >> 4) * (KVBLK * 16) + r_ * 16 + (c & 15)]) = r.kk[it];
     }
@@ -435,17 +440,18 @@ struct DQV2 {
     __syncthreads();
     if (n_tiles > 1) stage_load(st, k, v, kv_base, KVBLK, Tkv, Hkv);
 
-    const unsigned kt_base = lds_addr(sm->kt);
     const int lam = ln31 & 15;
-    const unsigned lane_off = ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
+    const unsigned trb = lds_addr(sm->kt)
+        + (unsigned)(ln31 >> 4) * (KVBLK * 16 * 2) + (unsigned)hi * 256
+        + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 
     for (int tile = 0; tile < n_tiles; ++tile) {
       const int kv0 = tile * KVBLK;
       const bool wave_active = (kv0 <= q_hi_wave);
 
       if (wave_active) {
-#pragma unroll
-        for (int half = 0; half < 2; ++half) {
+        attnc::static_for<2>([&](auto half_) {
+          constexpr int half = decltype(half_)::value;
           const int kv0h = kv0 + 32 * half;
           // S^T = K Q^T ; dP^T = V dO^T  (col = q = ln31, rows = kv half)
           floatx16 s_h, dp_h;
@@ -455,10 +461,8 @@ struct DQV2 {
           for (int s = 0; s < NDSTEP; ++s) {
             const int col = hi * 8 + 16 * s;
             const int r_ = 32 * half + ln31;
-            bf16x8 ka = *reinterpret_cast<const bf16x8*>(
-                &sm->k[r_ * DS + swz(r_, col)]);
-            bf16x8 va = *reinterpret_cast<const bf16x8*>(
-                &sm->v[r_ * DS + swz(r_, col)]);
+            bf16x8 ka = *reinterpret_cast<const bf16x8*>(&sm->k[r_ * DP + col]);
+            bf16x8 va = *reinterpret_cast<const bf16x8*>(&sm->v[r_ * DP + col]);
             s_h = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[s], s_h,
                                                           0, 0, 0);
             dp_h = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[s], dp_h,
@@ -486,19 +490,18 @@ struct DQV2 {
           attnc::c16_to_frags(dp_h, dsfrag);
 
           // dQ^T += Kt dS'^T (contraction over this kv half: k-steps
-          // 2*half, 2*half+1 of the Kt panel image)
-#pragma unroll
-          for (int dblk = 0; dblk < NDBLK; ++dblk) {
-            const unsigned pan = (unsigned)((dblk * 2 + (ln31 >> 4))
-                                            * (KVBLK * 16)) * 2;
+          // 2*half, 2*half+1 of the Kt panel image; base VGPR + offset
+          // immediates)
+          attnc::static_for<NDBLK>([&](auto dblk_) {
+            constexpr int dblk = decltype(dblk_)::value;
             uintx2 rk[4];
-#pragma unroll
-            for (int s2 = 0; s2 < 2; ++s2) {
-              const unsigned a = pan + (16 * (2 * half + s2) + 8 * hi) * 32
-                  + lane_off;
-              rk[2 * s2] = tr_read_b64(kt_base + a);
-              rk[2 * s2 + 1] = tr_read_b64(kt_base + a + 128);
-            }
+            attnc::static_for<2>([&](auto s2_) {
+              constexpr int s2 = decltype(s2_)::value;
+              constexpr unsigned a = dblk * 2 * (KVBLK * 16) * 2
+                  + (2 * half + s2) * 512;
+              rk[2 * s2] = attnc::tr_read_b64_off<a>(trb);
+              rk[2 * s2 + 1] = attnc::tr_read_b64_off<a + 128>(trb);
+            });
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
@@ -509,8 +512,8 @@ struct DQV2 {
               acc_dq[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                   kta, dsfrag[s2], acc_dq[dblk], 0, 0, 0);
             }
-          }
-        }  // half
+          });
+        });  // half
       }  // wave_active
 
       __syncthreads();
@@ -534,8 +537,7 @@ struct DQV2 {
 #pragma unroll
           for (int i = 0; i < 2; ++i) {
             const int r = r4 * 4 + 2 * i;
-            w[i] = (unsigned int)f32_to_bf16(acc_dq[dblk][r]) |
-                   ((unsigned int)f32_to_bf16(acc_dq[dblk][r + 1]) << 16);
+            w[i] = attnc::cvt_pk_bf16(acc_dq[dblk][r], acc_dq[dblk][r + 1]);
           }
           *reinterpret_cast<uint2*>(dqr + d) = make_uint2(w[0], w[1]);
         }
@@ -572,10 +574,11 @@ struct DKDVV2 {
   static constexpr int WG_KV = QBLK * NWV;
 
   // manual LDS carve (only the images this mode reads)
+  static constexpr int DP = G::DP;                     // padded row pitch
   static constexpr int PANEL = (DS / 16) * QBLK * 16;  // elements
   static constexpr int O_QROW = 0;
-  static constexpr int O_DOROW = O_QROW + QBLK * DS;
-  static constexpr int O_QT = O_DOROW + (DO_DK ? QBLK * DS : 0);
+  static constexpr int O_DOROW = O_QROW + QBLK * DP;
+  static constexpr int O_QT = O_DOROW + (DO_DK ? QBLK * DP : 0);
   static constexpr int O_DOT = O_QT + (DO_DK ? PANEL : 0);
   static constexpr int O_END = O_DOT + (DO_DV ? PANEL : 0);
   static constexpr int O_STATS = (O_END * 2 + 15) / 16 * 16;  // bytes, 16-al
@@ -625,12 +628,10 @@ struct DKDVV2 {
       const int i = tid + it * NTV;
       if (NCHS % NTV != 0 && i >= NCHS) continue;
       const int row = (i * 8) / D, c = (i * 8) % D;
-      *reinterpret_cast<shortx8*>(
-          &base[O_QROW + row * DS + swz(row, c)]) = r.qq[it];
+      *reinterpret_cast<shortx8*>(&base[O_QROW + row * DP + c]) = r.qq[it];
       const int pan = (c >> 4) * (QBLK * 16) + row * 16 + (c & 15);
       if constexpr (DO_DK) {
-        *reinterpret_cast<shortx8*>(
-            &base[O_DOROW + row * DS + swz(row, c)]) = r.dd[it];
+        *reinterpret_cast<shortx8*>(&base[O_DOROW + row * DP + c]) = r.dd[it];
         *reinterpret_cast<shortx8*>(&base[O_QT + pan]) = r.qq[it];
       }
       if constexpr (DO_DV) {
@@ -741,10 +742,11 @@ struct DKDVV2 {
       stage_load(st, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0, Tq, Hq);
     }
 
-    const unsigned qt_base = lds_addr(lds + O_QT);
-    const unsigned dot_base = lds_addr(lds + O_DOT);
     const int lam = ln31 & 15;
-    const unsigned lane_off = ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
+    const unsigned tr_lane = (unsigned)(ln31 >> 4) * (QBLK * 16 * 2)
+        + (unsigned)hi * 256 + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
+    const unsigned qt_trb = lds_addr(lds + O_QT) + tr_lane;
+    const unsigned dot_trb = lds_addr(lds + O_DOT) + tr_lane;
 
     for (int t = 0; t < n_iter; ++t) {
       const int q0 = cc.q0;
@@ -761,12 +763,12 @@ struct DKDVV2 {
         for (int s = 0; s < NDSTEP; ++s) {
           const int col = hi * 8 + 16 * s;
           bf16x8 qa = *reinterpret_cast<const bf16x8*>(
-              &lds[O_QROW + ln31 * DS + swz(ln31, col)]);
+              &lds[O_QROW + ln31 * DP + col]);
           s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[s], s_acc,
                                                           0, 0, 0);
           if constexpr (DO_DK) {
             bf16x8 da = *reinterpret_cast<const bf16x8*>(
-                &lds[O_DOROW + ln31 * DS + swz(ln31, col)]);
+                &lds[O_DOROW + ln31 * DP + col]);
             dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vfrag[s],
                                                              dp_acc, 0, 0, 0);
           }
@@ -806,24 +808,23 @@ struct DKDVV2 {
         if constexpr (DO_DV) attnc::c16_to_frags(s_acc, pfrag);
         if constexpr (DO_DK) attnc::c16_to_frags(dp_acc, dsfrag);
 
-        // dV += P^T dO ; dK += dS'^T Q  (B-operands via tr reads)
-#pragma unroll
-        for (int dblk = 0; dblk < NDBLK; ++dblk) {
-          const unsigned pan = (unsigned)((dblk * 2 + (ln31 >> 4))
-                                          * (QBLK * 16)) * 2;
+        // dV += P^T dO ; dK += dS'^T Q  (B-operands via tr reads; base
+        // VGPR + offset immediates)
+        attnc::static_for<NDBLK>([&](auto dblk_) {
+          constexpr int dblk = decltype(dblk_)::value;
           uintx2 rd[4], rq[4];
-#pragma unroll
-          for (int s = 0; s < 2; ++s) {
-            const unsigned a = pan + (16 * s + 8 * hi) * 32 + lane_off;
+          attnc::static_for<2>([&](auto s_) {
+            constexpr int s = decltype(s_)::value;
+            constexpr unsigned a = dblk * 2 * (QBLK * 16) * 2 + s * 512;
             if constexpr (DO_DV) {
-              rd[2 * s] = tr_read_b64(dot_base + a);
-              rd[2 * s + 1] = tr_read_b64(dot_base + a + 128);
+              rd[2 * s] = attnc::tr_read_b64_off<a>(dot_trb);
+              rd[2 * s + 1] = attnc::tr_read_b64_off<a + 128>(dot_trb);
             }
             if constexpr (DO_DK) {
-              rq[2 * s] = tr_read_b64(qt_base + a);
-              rq[2 * s + 1] = tr_read_b64(qt_base + a + 128);
+              rq[2 * s] = attnc::tr_read_b64_off<a>(qt_trb);
+              rq[2 * s + 1] = attnc::tr_read_b64_off<a + 128>(qt_trb);
             }
-          }
+          });
           asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
           __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
@@ -843,7 +844,7 @@ struct DKDVV2 {
                   dsfrag[s], qb, acc_dk[dblk], 0, 0, 0);
             }
           }
-        }
+        });
       }  // wave_active
 
       __syncthreads();
@@ -870,12 +871,16 @@ struct DKDVV2 {
         if (all_in || kvrow < Tkv) {
           if constexpr (DO_DK) {
             const long off = kv_base + (long)kvrow * Hkv * D + d;
-            dk[off] = f32_to_bf16(acc_dk[dblk][r]);
+            union { __bf16 h; unsigned short u; } ck;
+            ck.h = (__bf16)acc_dk[dblk][r];
+            dk[off] = ck.u;
           }
           if constexpr (DO_DV) {
             const long dvoff = ((long)b * Tkv + kvrow) * dvp + dvc
                 + (long)hkv * D + d;
-            dv[dvoff] = f32_to_bf16(acc_dv[dblk][r]);
+            union { __bf16 h; unsigned short u; } cv;
+            cv.h = (__bf16)acc_dv[dblk][r];
+            dv[dvoff] = cv.u;
           }
         }
       }

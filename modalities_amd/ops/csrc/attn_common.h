@@ -2,6 +2,8 @@
 // All maps were verified on gfx950 by the probes in attention_fwd.hip
 // (mfma_probe_32x32x16 / tr16_probe).
 #pragma once
+#include <type_traits>
+
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
@@ -21,6 +23,32 @@ __device__ __forceinline__ uintx2 tr_read_b64(unsigned addr_bytes) {
   return r;
 }
 
+// Immediate-offset variant: ONE base VGPR per kernel + compile-time
+// offsets keep the per-tile transpose reads free of address VALU (the
+// base is tile-invariant with a single LDS buffer).
+template <int OFF>
+__device__ __forceinline__ uintx2 tr_read_b64_off(unsigned base_bytes) {
+  static_assert(OFF >= 0 && OFF < 65536, "ds offset is 16-bit");
+  uintx2 r;
+  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:%c2"
+               : "=v"(r) : "v"(base_bytes), "i"(OFF));
+  return r;
+}
+
+// compile-time-unrolled loop whose induction variable is a constant
+// expression (usable as a template argument, e.g. for ds offset:N).
+template <int I, int N, typename F>
+__device__ __forceinline__ void static_for_impl(F&& f) {
+  if constexpr (I < N) {
+    f(std::integral_constant<int, I>{});
+    static_for_impl<I + 1, N>(f);
+  }
+}
+template <int N, typename F>
+__device__ __forceinline__ void static_for(F&& f) {
+  static_for_impl<0, N>(f);
+}
+
 __device__ __forceinline__ unsigned lds_addr(const void* p) {
   return (unsigned)(unsigned long long)p;  // LDS aperture: low 32 = offset
 }
@@ -37,6 +65,16 @@ __device__ __forceinline__ int crow(int reg, int hi) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
 }
 
+// One v_cvt_pk_bf16_f32 per f32 pair (hipcc fuses adjacent __bf16 casts;
+// the manual round-to-nearest-even bit-twiddle of common.h costs ~6 VALU
+// per element and dominated the v2 kernels' issue budget).
+__device__ __forceinline__ unsigned int cvt_pk_bf16(float lo, float hi) {
+  union { __bf16 h[2]; unsigned int u; } c;
+  c.h[0] = (__bf16)lo;
+  c.h[1] = (__bf16)hi;
+  return c.u;
+}
+
 // Re-layout f32 values from the MFMA C layout (col = lane&31 = j fixed,
 // rows i = crow(reg,hi)) into bf16 A/B fragments with per-lane element
 // k = hi*8+m (+16 per frag): in-register transpose via cvt_pk +
@@ -46,10 +84,8 @@ __device__ __forceinline__ void c_layout_to_frags(const float* p, bf16x8* frag,
   for (int s = 0; s < nfrag; ++s) {
     unsigned int w[4];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      w[i] = ((unsigned int)f32_to_bf16(p[8 * s + 2 * i])) |
-             ((unsigned int)f32_to_bf16(p[8 * s + 2 * i + 1]) << 16);
-    }
+    for (int i = 0; i < 4; ++i)
+      w[i] = cvt_pk_bf16(p[8 * s + 2 * i], p[8 * s + 2 * i + 1]);
     auto r02 = __builtin_amdgcn_permlane32_swap(w[0], w[2], false, false);
     auto r13 = __builtin_amdgcn_permlane32_swap(w[1], w[3], false, false);
     unsigned int fw[4] = {(unsigned int)r02[0], (unsigned int)r13[0],
@@ -66,10 +102,8 @@ __device__ __forceinline__ void c16_to_frags(const floatx16& c, bf16x8* frag) {
   for (int s = 0; s < 2; ++s) {
     unsigned int w[4];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      w[i] = ((unsigned int)f32_to_bf16(c[8 * s + 2 * i])) |
-             ((unsigned int)f32_to_bf16(c[8 * s + 2 * i + 1]) << 16);
-    }
+    for (int i = 0; i < 4; ++i)
+      w[i] = cvt_pk_bf16(c[8 * s + 2 * i], c[8 * s + 2 * i + 1]);
     auto r02 = __builtin_amdgcn_permlane32_swap(w[0], w[2], false, false);
     auto r13 = __builtin_amdgcn_permlane32_swap(w[1], w[3], false, false);
     unsigned int fw[4] = {(unsigned int)r02[0], (unsigned int)r13[0],

@@ -153,6 +153,76 @@ def scatter_seq(x, group, dim=1):
 
 # ---------------------------------------------------------------------------
 
+class _TPVocabEmbeddingForward:
+    """Vocab(row)-sharded embedding: each rank holds vocab/tp rows; lookup
+    masks out-of-range ids to zero and all-reduces the partial embeddings
+    (reference analog: embedding RowwiseParallel, model_factory.py:657-766).
+    Bound as the instance `forward` of the nn.Embedding whose weight was
+    sharded (functional lookup avoids recursing into itself)."""
+
+    def __init__(self, emb, group, tp_rank: int, tp_size: int, vocab: int):
+        self.emb = emb
+        self.group = group
+        self.voff = tp_rank * (vocab // tp_size)
+        self.vloc = vocab // tp_size
+
+    def __call__(self, ids: torch.Tensor) -> torch.Tensor:
+        in_range = (ids >= self.voff) & (ids < self.voff + self.vloc)
+        local = (ids - self.voff).clamp(0, self.vloc - 1)
+        w = self.emb.weight
+        y = torch.nn.functional.embedding(local, w)
+        y = y * in_range.unsqueeze(-1).to(w.dtype)
+        return reduce_from_tp(y, self.group)
+
+
+class _VocabParallelCE(torch.autograd.Function):
+    """Cross entropy over vocab-sharded logits [N, V/tp] without gathering
+    the full logits (the reference gathers: lm_head ColwiseParallel ->
+    Replicate; the gather-free loss removes the single largest activation
+    at tp>1). Max/log-sum-exp/target-logit combine with three scalar-ish
+    collectives."""
+
+    @staticmethod
+    def forward(ctx, logits, targets, group, voff, ignore_index):
+        lf = logits.float()
+        N, Vl = lf.shape
+        gmax = lf.max(dim=-1).values
+        dist.all_reduce(gmax, op=dist.ReduceOp.MAX, group=group)
+        ex = torch.exp(lf - gmax.unsqueeze(-1))
+        z = ex.sum(dim=-1)
+        dist.all_reduce(z, group=group)
+        valid = targets != ignore_index
+        in_range = (targets >= voff) & (targets < voff + Vl) & valid
+        tloc = (targets - voff).clamp(0, Vl - 1)
+        tl = lf.gather(-1, tloc.unsqueeze(-1)).squeeze(-1)
+        tl = tl * in_range.to(tl.dtype)
+        dist.all_reduce(tl, group=group)
+        n_valid = valid.sum().clamp(min=1)
+        loss = ((torch.log(z) + gmax - tl) * valid).sum() / n_valid
+        ctx.save_for_backward(ex, z, tloc, in_range, valid)
+        ctx.meta = (logits.dtype, n_valid)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        ex, z, tloc, in_range, valid = ctx.saved_tensors
+        dtype, n_valid = ctx.meta
+        scale = (dloss * valid.to(ex.dtype) / n_valid).unsqueeze(-1)
+        dlogits = ex / z.unsqueeze(-1) * scale
+        dlogits.scatter_add_(-1, tloc.unsqueeze(-1),
+                             -(scale * in_range.unsqueeze(-1).to(ex.dtype)))
+        return dlogits.to(dtype), None, None, None, None
+
+
+def vocab_parallel_cross_entropy(logits, targets, group, tp_rank: int,
+                                 tp_size: int, vocab: int,
+                                 ignore_index: int = -100) -> torch.Tensor:
+    """Mean CE over vocab-sharded logits [N, vocab/tp]; targets are GLOBAL
+    token ids [N]."""
+    voff = tp_rank * (vocab // tp_size)
+    return _VocabParallelCE.apply(logits, targets, group, voff, ignore_index)
+
+
 @torch.no_grad()
 def _shard_linear_(lin: nn.Linear, tp_rank: int, tp_size: int, dim: int) -> None:
     """Shard a Linear's weight (and bias for dim=0) in place.
@@ -177,6 +247,25 @@ def _shard_linear_(lin: nn.Linear, tp_rank: int, tp_size: int, dim: int) -> None
         lin.in_features = in_per
 
 
+@torch.no_grad()
+def _shard_fused_qkv_(lin, tp_rank: int, tp_size: int, Cq: int, Ckv: int):
+    """Shard the joint QKV weight [Cq+2*Ckv, h] per head group: each rank
+    keeps its q rows, k rows and v rows re-packed as a contiguous
+    [ (Cq+2*Ckv)/tp, h ] weight so the fused path stays a single GEMM."""
+    w = lin.weight
+    qp, kvp = Cq // tp_size, Ckv // tp_size
+    rows = []
+    for base, per in ((0, qp), (Cq, kvp), (Cq + Ckv, kvp)):
+        rows.append(w[base + tp_rank * per: base + (tp_rank + 1) * per])
+    lin.weight = nn.Parameter(torch.cat(rows, dim=0).clone())
+    if lin.bias is not None:
+        b = lin.bias
+        bs = [b[base + tp_rank * per: base + (tp_rank + 1) * per]
+              for base, per in ((0, qp), (Cq, kvp), (Cq + Ckv, kvp))]
+        lin.bias = nn.Parameter(torch.cat(bs, dim=0).clone())
+    lin.out_features = qp + 2 * kvp
+
+
 class _TPAttentionForward:
     """Replacement forward for CausalSelfAttention under TP(+SP)."""
 
@@ -192,9 +281,35 @@ class _TPAttentionForward:
         else:
             x = copy_to_tp(x, self.group)
         B, T, C = x.shape
-        q = attn.q_attn(x).view(B, T, attn.n_head_q, attn.head_dim)
-        k = attn.k_attn(x).view(B, T, attn.n_head_kv, attn.head_dim)
-        v = attn.v_attn(x).view(B, T, attn.n_head_kv, attn.head_dim)
+        c_loc = attn.n_head_q * attn.head_dim       # local (sharded) widths
+        kv_loc = attn.n_head_kv * attn.head_dim
+        if getattr(attn, "fused_qkv", False):
+            qkv = attn.qkv_attn(x)
+            if (attn.q_norm is None and rope_cos is not None
+                    and qkv.is_cuda):
+                from modalities_amd.ops.backend import use_hip
+                if use_hip(qkv):
+                    from modalities_amd.ops.attention import (
+                        fused_qkv_rope_attention)
+                    y = fused_qkv_rope_attention(qkv, rope_cos, rope_sin,
+                                                 attn.n_head_q,
+                                                 attn.n_head_kv,
+                                                 attn.head_dim)
+                    y = y.reshape(B, T, c_loc)
+                    y = attn.c_proj(y)
+                    if self.sp:
+                        y = scatter_seq(y, self.group, dim=1)
+                    else:
+                        y = reduce_from_tp(y, self.group)
+                    return attn.resid_dropout(y)
+            q, k, v = qkv.split([c_loc, kv_loc, kv_loc], dim=-1)
+            q = q.contiguous().view(B, T, attn.n_head_q, attn.head_dim)
+            k = k.contiguous().view(B, T, attn.n_head_kv, attn.head_dim)
+            v = v.contiguous().view(B, T, attn.n_head_kv, attn.head_dim)
+        else:
+            q = attn.q_attn(x).view(B, T, attn.n_head_q, attn.head_dim)
+            k = attn.k_attn(x).view(B, T, attn.n_head_kv, attn.head_dim)
+            v = attn.v_attn(x).view(B, T, attn.n_head_kv, attn.head_dim)
         if attn.q_norm is not None:
             q = attn.q_norm(q)
             k = attn.k_norm(k)
@@ -252,7 +367,8 @@ class _TPGeluMLPForward:
 def get_gpt2_tensor_parallelized_model(model, device_mesh=None, group=None,
                                        tp_rank: Optional[int] = None,
                                        tp_size: Optional[int] = None,
-                                       sequence_parallel: bool = False):
+                                       sequence_parallel: bool = False,
+                                       shard_vocab: bool = False):
     """Shard a GPT2LLM in place for TP (reference analog:
     model_factory.py:657-766).
 
@@ -280,12 +396,13 @@ def get_gpt2_tensor_parallelized_model(model, device_mesh=None, group=None,
     for block in model.blocks:
         attn = block.attn
         if getattr(attn, "fused_qkv", False):
-            raise NotImplementedError(
-                "tensor parallelism requires the unfused q/k/v layout "
-                "(build the model with fused_qkv=False)")
-        _shard_linear_(attn.q_attn, tp_rank, tp_size, dim=0)
-        _shard_linear_(attn.k_attn, tp_rank, tp_size, dim=0)
-        _shard_linear_(attn.v_attn, tp_rank, tp_size, dim=0)
+            _shard_fused_qkv_(attn.qkv_attn, tp_rank, tp_size,
+                              cfg.n_embd,
+                              attn.head_dim * attn.n_head_kv)
+        else:
+            _shard_linear_(attn.q_attn, tp_rank, tp_size, dim=0)
+            _shard_linear_(attn.k_attn, tp_rank, tp_size, dim=0)
+            _shard_linear_(attn.v_attn, tp_rank, tp_size, dim=0)
         _shard_linear_(attn.c_proj, tp_rank, tp_size, dim=1)
         attn.n_head_q //= tp_size
         attn.n_head_kv //= tp_size
@@ -313,4 +430,42 @@ def get_gpt2_tensor_parallelized_model(model, device_mesh=None, group=None,
             lambda mod, args, out: _SliceSeq.apply(out, group, 1))
         model.lm_head_norm.register_forward_pre_hook(
             lambda mod, args: (_GatherSeqToReplicated.apply(args[0], group, 1),))
+
+    if shard_vocab:
+        # Vocab sharding (reference shards embedding Rowwise and lm_head
+        # Colwise, model_factory.py:657-766; we additionally keep the loss
+        # gather-free via vocab_parallel_cross_entropy):
+        #  - wte: rows (vocab dim) sharded, masked lookup + all-reduce
+        #  - lm_head: out-features (vocab) sharded -> logits [B,T,V/tp]
+        #  - weight tying survives (both shard the same [V, h] rows)
+        vocab = cfg.vocab_size
+        if vocab % tp_size:
+            raise ValueError(f"vocab_size {vocab} not divisible by tp={tp_size}")
+        vloc = vocab // tp_size
+        tied = model.lm_head.weight is model.wte.weight
+        with torch.no_grad():
+            wte_shard = nn.Parameter(
+                model.wte.weight[tp_rank * vloc:(tp_rank + 1) * vloc].clone())
+            model.wte.weight = wte_shard
+            model.wte.num_embeddings = vloc
+            if tied:
+                model.lm_head.weight = wte_shard
+            else:
+                model.lm_head.weight = nn.Parameter(
+                    model.lm_head.weight[tp_rank * vloc:
+                                         (tp_rank + 1) * vloc].clone())
+            model.lm_head.out_features = vloc
+            if getattr(model.lm_head, "bias", None) is not None:
+                model.lm_head.bias = nn.Parameter(
+                    model.lm_head.bias[tp_rank * vloc:
+                                       (tp_rank + 1) * vloc].clone())
+        model.wte.forward = _TPVocabEmbeddingForward(
+            model.wte, group, tp_rank, tp_size, vocab)
+        # column-parallel entry conjugate: the sharded lm_head's input grad
+        # (dlogits_local @ W_local) is a PARTIAL sum — all-reduce it in
+        # backward so the residual stream sees the full gradient.
+        model.lm_head.register_forward_pre_hook(
+            lambda mod, args: (copy_to_tp(args[0], group),))
+        # loss wiring: the trainer/evaluator propagate this to the loss fn
+        model._tp_vocab_info = (group, tp_rank, tp_size, vocab)
     return model

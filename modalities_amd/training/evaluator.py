@@ -22,6 +22,8 @@ class Evaluator:
 
     @torch.no_grad()
     def evaluate_batch(self, batch, model, loss_fun: Loss) -> torch.Tensor:
+        from modalities_amd.training.trainer import Trainer
+        Trainer._propagate_tp_vocab_info(model, loss_fun)
         result_batch = model_predict_batch(model, batch.to(self.device))
         return loss_fun(result_batch)
 

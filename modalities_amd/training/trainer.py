@@ -94,6 +94,7 @@ class Trainer:
                 return True, loss, grad_norm
             return False, loss, None
         sharded = isinstance(model, XGMIShardedModel)
+        self._propagate_tp_vocab_info(model, loss_fun)
         cp_info = self._cp_info_of(model)
         if cp_info is not None:
             # CP: logits come back seq-sharded [B, T/cp, V]; slice the
@@ -199,6 +200,19 @@ class Trainer:
             checkpointing_callback(self.training_progress)
 
     # ------------------------------------------------------------------
+    @staticmethod
+    def _propagate_tp_vocab_info(model, loss_fun) -> None:
+        """When the model's lm_head is vocab-sharded (TP shard_vocab), the
+        loss must combine across the tp group (vocab-parallel CE). The TP
+        transform leaves `_tp_vocab_info` on the model; hand it to the
+        loss once."""
+        if getattr(loss_fun, "tp_vocab_info", "no") is None:
+            for m in (model, getattr(model, "module", None)):
+                info = getattr(m, "_tp_vocab_info", None) if m is not None else None
+                if info is not None:
+                    loss_fun.tp_vocab_info = info
+                    return
+
     @staticmethod
     def _cp_info_of(model):
         """(group, cp_rank, cp_size) when the model (or the module inside a

@@ -188,3 +188,87 @@ def test_tp2_dp2_matches_single_process():
     for r in range(4):
         assert results[r] == pytest.approx(ref_losses, rel=2e-4), \
             (results[r], ref_losses)
+
+
+# ---------------------------------------------------------------------------
+# Vocab sharding: embedding rows + lm_head columns sharded
+# over the tp group, loss via gather-free vocab-parallel cross-entropy
+# (reference shards both but gathers logits, model_factory.py:657-766).
+
+def _tp_vocab_worker(rank, world, fused_qkv, tying):
+    import torch.distributed as dist
+
+    from modalities_amd.parallel.tp import (get_gpt2_tensor_parallelized_model,
+                                            vocab_parallel_cross_entropy)
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg(fused_qkv=fused_qkv, use_weight_tying=tying))
+    model = get_gpt2_tensor_parallelized_model(
+        model, group=dist.group.WORLD, tp_rank=rank, tp_size=world,
+        shard_vocab=True)
+    assert model.lm_head.weight.shape[0] == VOCAB // world
+    assert model.wte.weight.shape[0] == VOCAB // world
+    if tying:
+        assert model.lm_head.weight is model.wte.weight
+    x, y = make_batch()
+    out = model({"input_ids": x})["logits"]      # [B, T, V/world]
+    assert out.shape[-1] == VOCAB // world
+    info = model._tp_vocab_info
+    loss = vocab_parallel_cross_entropy(out.reshape(-1, out.shape[-1]),
+                                        y.reshape(-1), info[0], info[1],
+                                        info[2], info[3])
+    loss.backward()
+    g_wte = model.wte.weight.grad.clone()
+    g_qw = (model.blocks[0].attn.qkv_attn.weight.grad.clone() if fused_qkv
+            else model.blocks[0].attn.q_attn.weight.grad.clone())
+    return (out.detach().numpy(), loss.item(), g_wte.numpy(), g_qw.numpy())
+
+
+@pytest.mark.parametrize("fused_qkv,tying", [(False, False), (True, False),
+                                             (False, True)])
+def test_tp2_vocab_sharded_matches_unsharded(fused_qkv, tying):
+    torch.manual_seed(0)
+    ref_model = GPT2LLM(tiny_cfg(fused_qkv=fused_qkv, use_weight_tying=tying))
+    x, y = make_batch()
+    out = ref_model({"input_ids": x})["logits"]
+    loss = torch.nn.functional.cross_entropy(out.view(-1, VOCAB).float(),
+                                             y.reshape(-1))
+    loss.backward()
+
+    results = run_distributed(_tp_vocab_worker, world_size=2,
+                              port=find_free_port(), args=(fused_qkv, tying))
+    # logits: rank r holds vocab slice r
+    for r in range(2):
+        got = torch.from_numpy(results[r][0])
+        ref_slice = out.detach()[..., r * VOCAB // 2:(r + 1) * VOCAB // 2]
+        torch.testing.assert_close(got, ref_slice, rtol=2e-4, atol=2e-4)
+        assert results[r][1] == pytest.approx(loss.item(), rel=1e-5)
+    # wte grad: rank r holds rows slice r (colwise-replicated input grads
+    # are identical across ranks)
+    name = "wte.weight"
+    ref_wte = dict(ref_model.named_parameters())[name].grad
+    for r in range(2):
+        got = torch.from_numpy(results[r][2])
+        torch.testing.assert_close(got, ref_wte[r * VOCAB // 2:
+                                                (r + 1) * VOCAB // 2],
+                                   rtol=2e-4, atol=2e-4)
+    # q (or fused qkv) projection grad: column-parallel rows per rank
+    if fused_qkv:
+        wname = "blocks.0.attn.qkv_attn.weight"
+        ref_g = dict(ref_model.named_parameters())[wname].grad
+        C, KV = 64, 32 * 2 // 2 * 2  # n_embd, kv_dim (n_head_kv*hd = 32)
+        KV = 32
+        for r in range(2):
+            got = torch.from_numpy(results[r][3])
+            exp = torch.cat([
+                ref_g[r * C // 2:(r + 1) * C // 2],
+                ref_g[C + r * KV // 2:C + (r + 1) * KV // 2],
+                ref_g[C + KV + r * KV // 2:C + KV + (r + 1) * KV // 2]])
+            torch.testing.assert_close(got, exp, rtol=2e-4, atol=2e-4)
+    else:
+        wname = "blocks.0.attn.q_attn.weight"
+        ref_g = dict(ref_model.named_parameters())[wname].grad
+        per = ref_g.shape[0] // 2
+        for r in range(2):
+            got = torch.from_numpy(results[r][3])
+            torch.testing.assert_close(got, ref_g[r * per:(r + 1) * per],
+                                       rtol=2e-4, atol=2e-4)
