@@ -122,6 +122,7 @@ class TrainingComponentsInstantiationModel(BaseModel):
     device_mesh: Any = None
     mfu_calculator: Any = None
     app_state: Any = None
+    pp_schedule: Any = None  # PipelineSchedule when PP is active
 
 
 class TextGenerationSettings(BaseModel):

@@ -115,8 +115,10 @@ class Main:
             evaluation_interval_in_steps=settings.intervals.evaluation_interval_in_steps,
             checkpointing_interval_in_steps=settings.intervals.checkpointing_interval_in_steps,
             training_log_interval_in_steps=settings.intervals.training_log_interval_in_steps,
-            device=device)
-        evaluator = Evaluator(progress_publisher, results_publisher, device=device)
+            device=device,
+            pp_schedule=components.pp_schedule)
+        evaluator = Evaluator(progress_publisher, results_publisher, device=device,
+                              pp_schedule=components.pp_schedule)
         gym = Gym(trainer, evaluator, components.loss_fn,
                   num_ranks=settings.cuda_env.world_size)
 

@@ -213,6 +213,24 @@ class PipelineSchedule:
              loss_fn: Optional[Callable], losses_out: Optional[list] = None):
         raise NotImplementedError
 
+    @torch.no_grad()
+    def eval_step(self, inputs: torch.Tensor, targets: Optional[torch.Tensor],
+                  loss_fn: Optional[Callable],
+                  losses_out: Optional[list] = None) -> list:
+        """Forward-only schedule pass for evaluation (reference
+        evaluator.py:88-180 dispatches the PP schedule in eval): every
+        microbatch flows through the stages, the last stage computes the
+        loss, no backward and no gradient traffic."""
+        self.loss_fn = loss_fn
+        losses_out = losses_out if losses_out is not None else []
+        mb_inputs = inputs.chunk(self.n_microbatches, dim=0)
+        mb_targets = targets.chunk(self.n_microbatches, dim=0) \
+            if targets is not None else [None] * self.n_microbatches
+        for mb_x, mb_y in zip(mb_inputs, mb_targets):
+            self._forward_mb(mb_x, mb_y, losses_out)
+        self._drain_sends()
+        return losses_out
+
     def broadcast_mean_loss(self, losses: list) -> torch.Tensor:
         """Mean micro-batch loss, broadcast from the last stage so every PP
         rank logs the true value (the trainer's loss reduction is then
@@ -280,7 +298,190 @@ class Schedule1F1B(PipelineSchedule):
         return losses_out
 
 
-SCHEDULES = {"gpipe": ScheduleGPipe, "1f1b": Schedule1F1B}
+class ScheduleInterleaved1F1B:
+    """Interleaved 1F1B over VIRTUAL stages (reference exposes torch
+    pipelining's Interleaved1F1B, pipeline_parallelism.py:14-20): the model
+    is split into pp_size * num_chunks stages; rank r holds chunks
+    c = 0..v-1 as global stages c*pp + r ("loop" placement). Warmup depth
+    (pp - r - 1)*2 + (v-1)*pp forwards, then 1F1B steady state with the
+    Megatron chunk rotation — smaller bubble than 1F1B at equal
+    microbatches ((pp-1)/(m*v) vs (pp-1)/m).
+
+    All sends are isend (ordering per channel matches both endpoints'
+    schedules); activations flow r -> r+1 within a chunk and wrap
+    pp-1 -> 0 between chunks; gradients mirror."""
+
+    def __init__(self, stages: list, pp_rank: int, pp_size: int,
+                 n_microbatches: int, group=None, device=None,
+                 activation_shape_fn: Optional[Callable] = None,
+                 sharded_engines: Optional[list] = None):
+        if n_microbatches % pp_size:
+            raise ValueError(
+                f"interleaved 1F1B needs n_microbatches ({n_microbatches}) "
+                f"divisible by pp_size ({pp_size})")
+        self.stages = stages                      # this rank's chunks
+        self.v = len(stages)
+        self.pp_rank = pp_rank
+        self.pp_size = pp_size
+        self.num_stages = pp_size                 # for broadcast_mean_loss
+        self.n_microbatches = n_microbatches
+        self.group = group
+        self.device = device or torch.device("cpu")
+        self.activation_shape_fn = activation_shape_fn
+        self.sharded_engines = sharded_engines
+        self.is_first = pp_rank == 0              # owns global stage 0
+        self.is_last = pp_rank == pp_size - 1     # owns the final stage
+        self._pending: list = []
+        ranks = dist.get_process_group_ranks(group) if group is not None else             list(range(dist.get_world_size())) if dist.is_initialized() else [0]
+        self._ranks = ranks
+        self._prev_rank = ranks[pp_rank - 1] if pp_rank > 0 else ranks[-1]
+        self._next_rank = ranks[pp_rank + 1] if pp_rank < pp_size - 1 else ranks[0]
+
+    # channel helpers --------------------------------------------------------
+    def _send(self, tensor, dst):
+        t = tensor.detach().contiguous()
+        work = dist.isend(t, dst=dst, group=self.group)
+        self._pending.append((work, t))
+
+    def _drain_sends(self):
+        for work, _ in self._pending:
+            work.wait()
+        self._pending = []
+
+    def _recv_from(self, src, shape, dtype):
+        buf = torch.empty(shape, dtype=dtype, device=self.device)
+        dist.recv(buf, src=src, group=self.group)
+        return buf
+
+    def _act_shape(self, mb_size, seq_len):
+        if self.activation_shape_fn is not None:
+            return self.activation_shape_fn(mb_size, seq_len)
+        return (mb_size, seq_len, self.stages[0].config.n_embd)
+
+    def _act_dtype(self):
+        return next(self.stages[0].parameters()).dtype
+
+    # stage-boundary roles ---------------------------------------------------
+    def _chunk_is_first(self, c):   # global stage c*pp + r == 0
+        return c == 0 and self.pp_rank == 0
+
+    def _chunk_is_last(self, c):
+        return c == self.v - 1 and self.pp_rank == self.pp_size - 1
+
+    def _fwd_chunk_at(self, k):
+        return (k // self.pp_size) % self.v
+
+    def _bwd_chunk_at(self, k):
+        return self.v - 1 - ((k // self.pp_size) % self.v)
+
+    # one fwd / bwd op -------------------------------------------------------
+    def _forward_op(self, c, mb_idx, mb_inputs, mb_targets, losses_out,
+                    saved):
+        if self._chunk_is_first(c):
+            x = mb_inputs[mb_idx]
+        else:
+            mb = mb_inputs[mb_idx]
+            x = self._recv_from(self._prev_rank,
+                                self._act_shape(mb.shape[0], mb.shape[1]),
+                                self._act_dtype())
+            x.requires_grad_(True)
+        eng = self.sharded_engines[c] if self.sharded_engines else None
+        out = eng(x) if eng is not None else self.stages[c](x)
+        loss = None
+        if self._chunk_is_last(c):
+            if self.loss_fn is not None:
+                loss = self.loss_fn((out, mb_targets[mb_idx]))
+                losses_out.append(loss.detach())
+        else:
+            self._send(out, self._next_rank)
+        saved[c].append((x, out, loss))
+
+    def _backward_op(self, c, saved):
+        x, out, loss = saved[c].pop(0)
+        if self._chunk_is_last(c):
+            (loss / self.n_microbatches).backward()
+        else:
+            dgrad = torch.empty_like(out)
+            dist.recv(dgrad, src=self._next_rank, group=self.group)
+            torch.autograd.backward(out, grad_tensors=dgrad)
+        if self.sharded_engines:
+            self.sharded_engines[c].backward_epilogue()
+        if not self._chunk_is_first(c):
+            self._send(x.grad, self._prev_rank)
+
+    # the schedule -----------------------------------------------------------
+    def step(self, inputs, targets, loss_fn, losses_out=None):
+        self.loss_fn = loss_fn
+        losses_out = losses_out if losses_out is not None else []
+        mb_inputs = list(inputs.chunk(self.n_microbatches, dim=0))
+        mb_targets = list(targets.chunk(self.n_microbatches, dim=0)) \
+            if targets is not None else [None] * self.n_microbatches
+
+        total = self.n_microbatches * self.v
+        n_warmup = min(total, (self.pp_size - self.pp_rank - 1) * 2
+                       + (self.v - 1) * self.pp_size)
+        saved = [[] for _ in range(self.v)]
+        fwd_count = [0] * self.v   # per-chunk microbatch cursor
+        kf = kb = 0
+
+        def do_fwd():
+            nonlocal kf
+            c = self._fwd_chunk_at(kf)
+            self._forward_op(c, fwd_count[c], mb_inputs, mb_targets,
+                             losses_out, saved)
+            fwd_count[c] += 1
+            kf += 1
+
+        def do_bwd():
+            nonlocal kb
+            self._backward_op(self._bwd_chunk_at(kb), saved)
+            kb += 1
+
+        for _ in range(n_warmup):
+            do_fwd()
+        for _ in range(total - n_warmup):   # steady 1F1B
+            do_fwd()
+            do_bwd()
+        while kb < total:                   # cooldown
+            do_bwd()
+        self._drain_sends()
+        return losses_out
+
+    @torch.no_grad()
+    def eval_step(self, inputs, targets, loss_fn, losses_out=None):
+        self.loss_fn = loss_fn
+        losses_out = losses_out if losses_out is not None else []
+        mb_inputs = list(inputs.chunk(self.n_microbatches, dim=0))
+        mb_targets = list(targets.chunk(self.n_microbatches, dim=0)) \
+            if targets is not None else [None] * self.n_microbatches
+        saved = [[] for _ in range(self.v)]
+        fwd_count = [0] * self.v
+        for k in range(self.n_microbatches * self.v):
+            c = self._fwd_chunk_at(k)
+            self._forward_op(c, fwd_count[c], mb_inputs, mb_targets,
+                             losses_out, saved)
+            fwd_count[c] += 1
+        self._drain_sends()
+        return losses_out
+
+    def broadcast_mean_loss(self, losses: list) -> torch.Tensor:
+        src = self._ranks[self.pp_size - 1]
+        if self.is_last and losses:
+            val = torch.stack([l.detach().float().cpu() for l in losses]).mean()
+        else:
+            val = torch.zeros((), dtype=torch.float32)
+        buf = val.to(self.device) if self.device.type == "cuda" else val
+        dist.broadcast(buf, src=src, group=self.group)
+        return buf.cpu()
+
+
+def interleaved_stage_ids(pp_rank: int, pp_size: int, num_chunks: int) -> list[int]:
+    """Global stage ids held by one rank under loop placement."""
+    return [c * pp_size + pp_rank for c in range(num_chunks)]
+
+
+SCHEDULES = {"gpipe": ScheduleGPipe, "1f1b": Schedule1F1B,
+             "interleaved": ScheduleInterleaved1F1B}
 
 
 def get_pipeline_schedule(variant: str, **kwargs) -> PipelineSchedule:

@@ -221,3 +221,91 @@ def test_pp2_dp2_matches_single_process():
                                        rtol=1e-4, atol=1e-6)
             found += 1
     assert found >= 2  # wte on stage-0 ranks, lm_head on stage-1 ranks
+
+
+# ---------------------------------------------------------------------------
+# Interleaved 1F1B: pp=2 with 2 virtual chunks per rank
+# (4 global stages), vs single-process reference.
+
+def _pp_interleaved_worker(rank, world):
+    import torch.distributed as dist
+
+    from modalities_amd.loss_functions import CLMCrossEntropyLoss
+    from modalities_amd.parallel.pp import (ScheduleInterleaved1F1B,
+                                            interleaved_stage_ids)
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    v = 2
+    stages_all = split_model_into_stages(model, world * v)
+    my = [stages_all[i] for i in interleaved_stage_ids(rank, world, v)]
+    sched = ScheduleInterleaved1F1B(my, pp_rank=rank, pp_size=world,
+                                    n_microbatches=N_MB,
+                                    group=dist.group.WORLD)
+    x, y = make_batch()
+    loss_fn = CLMCrossEntropyLoss("target_ids", "logits")
+    losses = sched.step(x, y, loss_fn)
+    mean = sched.broadcast_mean_loss(losses)
+    grads = {}
+    for c, st in enumerate(my):
+        for n, p in st.named_parameters():
+            if p.grad is not None:
+                grads[f"chunk{c}.{n}"] = p.grad.clone().numpy()
+    # eval_step: forward-only losses must equal the training-step losses
+    # (same params, no optimizer step in between)
+    eval_losses = sched.eval_step(x, y, loss_fn)
+    return ([l.item() for l in losses], grads, mean.item(),
+            [l.item() for l in eval_losses])
+
+
+def test_pp2_interleaved_matches_single_process():
+    ref_losses, ref_grads = reference_loss_and_grads()
+    results = run_distributed(_pp_interleaved_worker, world_size=2,
+                              port=find_free_port())
+    # rank 1 holds the last stage (chunk 1 = global stage 3)
+    pp_losses = results[1][0]
+    assert sorted(pp_losses) == pytest.approx(sorted(ref_losses), rel=1e-5)
+    assert results[0][2] == pytest.approx(sum(ref_losses) / N_MB, rel=1e-5)
+    assert results[1][3] == pytest.approx(pp_losses, rel=1e-5)  # eval_step
+    # 4 blocks over 4 stages: rank0 chunk0 = {wte, block0}, chunk1 = block2;
+    # rank1 chunk0 = block1, chunk1 = {block3, head}
+    g0, g1 = results[0][1], results[1][1]
+    torch.testing.assert_close(torch.from_numpy(g0["chunk0.wte.weight"]),
+                               torch.from_numpy(ref_grads["wte.weight"]),
+                               rtol=1e-4, atol=1e-6)
+    torch.testing.assert_close(
+        torch.from_numpy(g0["chunk1.blocks.0.attn.q_attn.weight"]),
+        torch.from_numpy(ref_grads["blocks.2.attn.q_attn.weight"]),
+        rtol=1e-4, atol=1e-6)
+    torch.testing.assert_close(
+        torch.from_numpy(g1["chunk0.blocks.0.attn.q_attn.weight"]),
+        torch.from_numpy(ref_grads["blocks.1.attn.q_attn.weight"]),
+        rtol=1e-4, atol=1e-6)
+    torch.testing.assert_close(torch.from_numpy(g1["chunk1.lm_head.weight"]),
+                               torch.from_numpy(ref_grads["lm_head.weight"]),
+                               rtol=1e-4, atol=1e-6)
+
+
+def _pp_eval_worker(rank, world, variant):
+    import torch.distributed as dist
+
+    from modalities_amd.loss_functions import CLMCrossEntropyLoss
+    from modalities_amd.parallel.pp import get_pipeline_schedule
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    stages = split_model_into_stages(model, world)
+    sched = get_pipeline_schedule(
+        variant, stage=stages[rank], stage_idx=rank, num_stages=world,
+        n_microbatches=N_MB, group=dist.group.WORLD)
+    x, y = make_batch()
+    loss_fn = CLMCrossEntropyLoss("target_ids", "logits")
+    losses = sched.eval_step(x, y, loss_fn)
+    return sched.broadcast_mean_loss(losses).item()
+
+
+@pytest.mark.parametrize("variant", ["gpipe", "1f1b"])
+def test_pp2_eval_step_forward_only(variant):
+    ref_losses, _ = reference_loss_and_grads()
+    results = run_distributed(_pp_eval_worker, world_size=2,
+                              port=find_free_port(), args=(variant,))
+    for r in range(2):
+        assert results[r] == pytest.approx(sum(ref_losses) / N_MB, rel=1e-5)
