@@ -163,3 +163,34 @@ class SyntheticLMDataset(Dataset):
         g = torch.Generator().manual_seed(self.seed + idx)
         return {self.sample_key: torch.randint(0, self.vocab_size,
                                                (self.sequence_length + 1,), generator=g)}
+
+
+class MemMapDataset(Dataset):
+    """Tokenize-on-the-fly dataset over a JSONL file via the byte-offset
+    index (capability parity with reference dataloader/dataset.py:134-188):
+    each __getitem__ mmap-reads one line, extracts the text field
+    (jq_pattern, simple ".field[.sub]" selectors) and tokenizes it."""
+
+    def __init__(self, raw_data_path: Path, tokenizer, sample_key: str,
+                 index_path: Path = None, jq_pattern: str = ".text"):
+        import json
+
+        from modalities_amd.dataloader.create_index import LargeFileLinesReader
+        self.sample_key = sample_key
+        self.reader = LargeFileLinesReader(Path(raw_data_path),
+                                           index_path=index_path)
+        self.tokenizer = tokenizer
+        self._fields = [f for f in jq_pattern.split(".") if f]
+        self._json = json
+
+    def __len__(self) -> int:
+        return len(self.reader)
+
+    def __getitem__(self, idx: int) -> dict[str, torch.Tensor]:
+        if idx >= len(self.reader):
+            raise IndexError("Index out of bounds")
+        obj = self._json.loads(self.reader[idx])
+        for f in self._fields:
+            obj = obj[f]
+        ids = self.tokenizer.tokenize(obj)
+        return {self.sample_key: torch.tensor(ids, dtype=torch.long)}

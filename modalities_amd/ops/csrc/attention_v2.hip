@@ -200,18 +200,22 @@ struct FwdV2 {
           s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb, qfrag[s], s1, 0, 0, 0);
         }
 
-        // ---- scale + causal mask; per-lane P rows (q = ln31) -----------
+        // ---- scale + causal mask; per-lane P rows (q = ln31).
+        // Interior fast path: row max taken on RAW scores (one scale
+        // multiply at the end instead of 32), P computed later as
+        // exp2(fma(s, scale2, -m)) — saves ~64 VALU/tile.
         float p[32];
         float tmax = -INFINITY;
+        bool raw_scores = false;
         if (kv0 + KVBLK - 1 <= q_lo_wave && kv0 + KVBLK <= Tkv && qg < Tq) {
+          raw_scores = true;
 #pragma unroll
           for (int r = 0; r < 16; ++r) {
-            float v0 = s0[r] * scale2;
-            float v1 = s1[r] * scale2;
-            p[r] = v0;
-            p[16 + r] = v1;
-            tmax = fmaxf(tmax, fmaxf(v0, v1));
+            p[r] = s0[r];
+            p[16 + r] = s1[r];
+            tmax = fmaxf(tmax, fmaxf(s0[r], s1[r]));
           }
+          tmax *= scale2;
         } else {
 #pragma unroll
           for (int r = 0; r < 16; ++r) {
@@ -243,10 +247,18 @@ struct FwdV2 {
         }
         const float m_safe2 = (m_run == -INFINITY) ? 0.f : m_run;
         float psum = 0.f;
+        if (raw_scores) {
 #pragma unroll
-        for (int r = 0; r < 32; ++r) {
-          p[r] = __builtin_amdgcn_exp2f(p[r] - m_safe2);
-          psum += p[r];
+          for (int r = 0; r < 32; ++r) {
+            p[r] = __builtin_amdgcn_exp2f(fmaf(p[r], scale2, -m_safe2));
+            psum += p[r];
+          }
+        } else {
+#pragma unroll
+          for (int r = 0; r < 32; ++r) {
+            p[r] = __builtin_amdgcn_exp2f(p[r] - m_safe2);
+            psum += p[r];
+          }
         }
         psum += __shfl_xor(psum, 32, 64);
         l_run += psum;
@@ -474,7 +486,8 @@ struct DQV2 {
           if (kv0h + 31 <= q_lo_wave && kv0h + 32 <= Tkv && qg < Tq) {
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
-              const float p = __builtin_amdgcn_exp2f(s_h[r] * scale2 - my_lse2);
+              const float p = __builtin_amdgcn_exp2f(
+                  fmaf(s_h[r], scale2, -my_lse2));
               dp_h[r] = scale * p * (dp_h[r] - my_delta);
             }
           } else {
@@ -482,7 +495,8 @@ struct DQV2 {
             for (int r = 0; r < 16; ++r) {
               const int kg = kv0h + crow(r, hi);
               const float p = (kg <= qgl && kg < Tkv && qg < Tq)
-                  ? __builtin_amdgcn_exp2f(s_h[r] * scale2 - my_lse2) : 0.f;
+                  ? __builtin_amdgcn_exp2f(fmaf(s_h[r], scale2, -my_lse2))
+                  : 0.f;
               dp_h[r] = scale * p * (dp_h[r] - my_delta);
             }
           }
@@ -782,7 +796,8 @@ struct DKDVV2 {
 #pragma unroll
           for (int r = 0; r < 16; ++r) {
             const float lq = lse_s[crow(r, hi)];
-            const float pv = __builtin_amdgcn_exp2f(s_acc[r] * scale2k - lq);
+            const float pv = __builtin_amdgcn_exp2f(
+                fmaf(s_acc[r], scale2k, -lq));
             s_acc[r] = pv;
             if constexpr (DO_DK) {
               const float dl = delta_s[crow(r, hi)];
@@ -796,7 +811,7 @@ struct DKDVV2 {
             const bool ok = (qg < Tq) && (qg + q_off >= kvg) && (kvg < Tkv);
             const float lq = lse_s[crow(r, hi)];
             const float pv = ok
-                ? __builtin_amdgcn_exp2f(s_acc[r] * scale2k - lq) : 0.f;
+                ? __builtin_amdgcn_exp2f(fmaf(s_acc[r], scale2k, -lq)) : 0.f;
             s_acc[r] = pv;
             if constexpr (DO_DK) {
               const float dl = delta_s[crow(r, hi)];
@@ -910,7 +925,7 @@ __global__ __launch_bounds__(NT, 2) void dq2_kernel(
 }
 
 template <int D, int MODE>
-__global__ __launch_bounds__((DKDVV2<D, MODE>::NTV), 2)
+__global__ __launch_bounds__((DKDVV2<D, MODE>::NTV), (MODE == 1 ? 3 : 2))
 void dkdv2_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
     const unsigned short* dout, const float* lse, const float* delta,

@@ -19,6 +19,7 @@ from modalities_amd.checkpointing import (AppState,
 from modalities_amd.dataloader.dataloader import (GPT2LLMCollateFn, LLMDataLoader,
                                                   LossMaskingCollateFnWrapper)
 from modalities_amd.dataloader.dataset import (CombinedDataset, DummyDataset,
+                                               MemMapDataset,
                                                PackedMemMapDatasetContinuous,
                                                PackedMemMapDatasetMegatron,
                                                SyntheticLMDataset)
@@ -169,6 +170,7 @@ COMPONENTS: list[ComponentEntity] = [
                     PackedMemMapDatasetContinuous, None),
     ComponentEntity("dataset", "packed_mem_map_dataset_megatron",
                     PackedMemMapDatasetMegatron, None),
+    ComponentEntity("dataset", "mem_map_dataset", MemMapDataset, None),
     ComponentEntity("dataset", "dummy_dataset", DummyDataset, None),
     ComponentEntity("dataset", "combined", CombinedDataset, None),
     ComponentEntity("dataset", "synthetic_lm", SyntheticLMDataset, None),

@@ -171,3 +171,51 @@ def test_pbin_byte_compat_with_reference_layout(tmp_path):
     assert raw[12:18] == np.asarray([1, 2, 3], dtype="<u2").tobytes()
     index = pickle.loads(raw[18:])
     assert index == [(0, 6)]
+
+
+def test_memmap_dataset_tokenize_on_the_fly(tmp_path):
+    """MemMapDataset (reference dataset.py:134-188): jsonl -> index ->
+    per-item tokenize must match direct tokenization."""
+    from modalities_amd.api import create_raw_data_index
+    from modalities_amd.dataloader.dataset import MemMapDataset
+    from modalities_amd.tokenization.tokenizer_wrapper import CharTokenizer
+
+    src = tmp_path / "corpus.jsonl"
+    texts = ["hello world", "lorem ipsum dolor", "a"]
+    src.write_text("\n".join(json.dumps({"text": t}) for t in texts) + "\n")
+    idx = tmp_path / "corpus.idx"
+    create_raw_data_index(src, idx)
+    tok = CharTokenizer()
+    ds = MemMapDataset(src, tok, sample_key="input_ids", index_path=idx)
+    assert len(ds) == 3
+    for i, t in enumerate(texts):
+        assert ds[i]["input_ids"].tolist() == list(tok.tokenize(t))
+    with pytest.raises(IndexError):
+        ds[3]
+
+
+def test_verify_tokenization_consistency(tmp_path):
+    from modalities_amd.utils.verify_tokenization_consistency import (
+        verify_index, verify_tokenization_consistency)
+
+    src = tmp_path / "corpus.jsonl"
+    texts = ["hello world", "lorem ipsum dolor sit amet", "xyz"]
+    src.write_text("\n".join(json.dumps({"text": t}) for t in texts) + "\n")
+    n = verify_tokenization_consistency(
+        src, tokenizer_config={"component_key": "tokenizer",
+                               "variant_key": "char", "config": {}})
+    assert n == 3
+
+    # corrupt index must be detected
+    import pickle
+    from modalities_amd.api import create_raw_data_index
+    idx = tmp_path / "corpus.idx"
+    create_raw_data_index(src, idx)
+    with open(idx, "rb") as f:
+        index = pickle.load(f)
+    index[1] = (index[1][0] + 2, index[1][1])
+    bad = tmp_path / "bad.idx"
+    with open(bad, "wb") as f:
+        pickle.dump(index, f)
+    with pytest.raises(AssertionError):
+        verify_index(src, bad)
