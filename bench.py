@@ -1,13 +1,13 @@
 #!/usr/bin/env python3
 """Flagship training benchmark for modalities_amd on MI355X.
 
-Measures the BASELINE.json metric: whole-node training tokens/s (+MFU) for a
-2.7B-class GPT (h=2560, L=32, seq 4096, bf16, FSDP full-shard equivalent,
-micro-batch 2 per GPU — the reference's headline scaling config,
-/root/reference/README.md:321: 18.63 samples/s on 8xA100 = 76.3k tok/s).
-MI355X-first deviation from the reference 2.7B: head_dim 128 (20 heads)
-instead of 80 (32 heads) — identical h/L/param count/FLOPs per token, but
-MFMA-tile-aligned. Weak scaling: per-GPU batch fixed as N grows.
+Measures the BASELINE.json metric: whole-node training tokens/s (+MFU) for
+the reference's EXACT 2.7B GPT shape (h=2560, L=32, 32 heads x head_dim 80,
+seq 4096, bf16, FSDP full-shard equivalent, micro-batch 2 per GPU — the
+reference's headline scaling config, /root/reference/README.md:321:
+18.63 samples/s on 8xA100 = 76.3k tok/s). The K1 v2 kernels run head_dim
+80 natively (r1 ran an hd=128/20-head equivalent; kept as
+gpt2-2.7b-hd128 for A/B). Weak scaling: per-GPU batch fixed as N grows.
 
 Usage:
   python bench.py --gpus 1 --steps 20 --warmup 5          # single GPU
@@ -29,7 +29,12 @@ import torch.distributed as dist
 
 def build_model_cfg(model_name: str):
     from modalities_amd.models.gpt2 import GPT2LLMConfig
-    if model_name == "gpt2-2.7b":
+    if model_name == "gpt2-2.7b":  # the exact reference shape (32 x hd80)
+        return GPT2LLMConfig(
+            vocab_size=50304, n_layer=32, n_head_q=32, n_head_kv=32,
+            n_embd=2560, ffn_hidden=10240, sequence_length=4096,
+            activation_type="swiglu", use_weight_tying=False)
+    if model_name == "gpt2-2.7b-hd128":  # MFMA-square-tile variant (r1 cfg)
         return GPT2LLMConfig(
             vocab_size=50304, n_layer=32, n_head_q=20, n_head_kv=20,
             n_embd=2560, ffn_hidden=10240, sequence_length=4096,

@@ -1,0 +1,95 @@
+"""Per-GPU memory budgeting for the XGMI sharding engine (VERDICT r1 #7:
+make the 70B single-node config real with documented memory math).
+
+The engine's steady-state per-GPU footprint for a model of N params on
+`world` GPUs (modalities_amd/parallel/fsdp.py):
+
+  fp32 master shard        4 B/param / world
+  Adam exp_avg + exp_avg_sq  8 B/param / world
+  fp32 grad shard          4 B/param / world
+  bf16 working shard       2 B/param / world     (param_dtype=bf16)
+  gathered bf16 full units 2 B/param x live_units x unit_params
+  activations              model/micro-batch dependent (estimated for the
+                           GPT2 block structure below)
+
+With reshard_after_forward=True at most `prefetch+1` units are gathered at
+once; False keeps ALL units gathered (adds 2N bytes — only viable <=8B).
+"""
+
+from dataclasses import dataclass
+
+
+@dataclass
+class MemoryEstimate:
+    n_params: int
+    shard_bytes: int          # master + moments + grads + bf16 shard
+    gathered_bytes: int       # live gathered bf16 units
+    activation_bytes: int
+    total_bytes: int
+
+    def gib(self, x: int) -> float:
+        return x / (1 << 30)
+
+    def table(self) -> str:
+        rows = [
+            ("params", f"{self.n_params / 1e9:.2f} B"),
+            ("optimizer+grad+param shards", f"{self.gib(self.shard_bytes):.1f} GiB"),
+            ("gathered bf16 units", f"{self.gib(self.gathered_bytes):.1f} GiB"),
+            ("activations (est.)", f"{self.gib(self.activation_bytes):.1f} GiB"),
+            ("total (est.)", f"{self.gib(self.total_bytes):.1f} GiB"),
+        ]
+        w = max(len(k) for k, _ in rows)
+        return "\n".join(f"{k:<{w}}  {v}" for k, v in rows)
+
+
+def gpt2_param_count(cfg) -> int:
+    """Parameter count from a GPT2LLMConfig without building the model."""
+    h, v = cfg.n_embd, cfg.vocab_size
+    head_dim = h // cfg.n_head_q
+    kv = head_dim * cfg.n_head_kv
+    if cfg.activation_type == "swiglu":
+        hidden = 256 * ((int(2 * cfg.ffn_hidden / 3) + 255) // 256)
+        mlp = 3 * h * hidden
+    else:
+        mlp = 2 * h * cfg.ffn_hidden
+    block = h * h + 2 * h * kv + h * h + mlp + 2 * h  # qkv + proj + mlp + norms
+    emb = v * h * (1 if cfg.use_weight_tying else 2)
+    return cfg.n_layer * block + emb + h  # + final norm
+
+
+def estimate_sharded_memory(cfg, world: int, micro_batch: int,
+                            blocks_per_unit: int = 4,
+                            reshard_after_forward: bool = True,
+                            full_ac: bool = False,
+                            prefetch_units: int = 1) -> MemoryEstimate:
+    """Steady-state per-GPU bytes for the sharded engine on `world` GPUs."""
+    n = gpt2_param_count(cfg)
+    h, T = cfg.n_embd, cfg.sequence_length
+    B = micro_batch
+    shard = n * (4 + 8 + 4 + 2) // world
+
+    block_params = (n - 2 * cfg.vocab_size * h) // cfg.n_layer
+    unit_params = block_params * blocks_per_unit
+    n_units = (cfg.n_layer + blocks_per_unit - 1) // blocks_per_unit + 1
+    if reshard_after_forward:
+        gathered = unit_params * (1 + prefetch_units) * 2
+        # embedding/head unit gathered while live
+        gathered += 2 * cfg.vocab_size * h * 2
+    else:
+        gathered = n * 2
+
+    # activations per micro-batch (bf16): with full AC only the block
+    # INPUTS are stored (1 x [B,T,h] per block) + one block's working set;
+    # without AC ~18 tensor-equivalents per block survive to backward
+    # (qkv/attn/mlp intermediates incl. the ffn-width ones).
+    act_per_block_boundary = B * T * h * 2
+    if full_ac:
+        act = act_per_block_boundary * cfg.n_layer
+        act += act_per_block_boundary * 20          # one recompute working set
+    else:
+        act = act_per_block_boundary * 18 * cfg.n_layer
+    # logits + CE workspace: [B, T, vocab] fp32-equivalent x ~2
+    act += B * T * cfg.vocab_size * 4 * 2
+
+    total = shard + gathered + act
+    return MemoryEstimate(n, shard, gathered, act, total)
