@@ -9,6 +9,7 @@ import torch
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: test requires an MI355X GPU (run via gpurun)")
+    config.addinivalue_line("markers", "slow: multi-process / long-running CPU test")
 
 
 def pytest_collection_modifyitems(config, items):
