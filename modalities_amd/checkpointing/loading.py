@@ -48,6 +48,23 @@ class ShardedCheckpointLoading:
         model = app_state.model
         new_world, new_rank = model.world, model.rank
 
+        # model/checkpoint compatibility: every engine unit must exist in
+        # the saved layout with the same LOGICAL length (resharding changes
+        # only the per-rank split; a different model is a hard error).
+        for u in model.units:
+            if u.name not in saved_layout:
+                raise ValueError(
+                    f"Checkpoint {folder} does not match the model: unit "
+                    f"{u.name!r} missing from the saved shard layout "
+                    f"(saved units: {sorted(saved_layout)})")
+            logical_here = (u.offsets[-1] + u.numels[-1]) if u.numels else 0
+            if saved_layout[u.name]["logical_numel"] != logical_here:
+                raise ValueError(
+                    f"Checkpoint {folder} does not match the model: unit "
+                    f"{u.name!r} has {saved_layout[u.name]['logical_numel']} "
+                    f"saved parameters vs {logical_here} in the model "
+                    f"(different architecture or parallel layout)")
+
         # target CPU staging buffers for this rank's new shards
         targets: dict[str, torch.Tensor] = {}
         for u in model.units:

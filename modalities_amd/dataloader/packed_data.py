@@ -51,13 +51,29 @@ class EmbeddedStreamData:
         if not self._data_path.is_file():
             raise FileNotFoundError(f"Packed data not found at {self._data_path.absolute()}. "
                                     f"Create it with `modalities-amd data pack_encoded_data`.")
+        file_size = self._data_path.stat().st_size
         with self._data_path.open("rb") as f:
             self.data_len = int.from_bytes(f.read(DATA_SECTION_LENGTH_IN_BYTES), "little")
             self.token_size_in_bytes = int.from_bytes(
                 f.read(TOKEN_SIZE_DESCRIPTOR_LENGTH_IN_BYTES), "little", signed=False)
+            if self.token_size_in_bytes not in (1, 2, 4):
+                raise ValueError(
+                    f"Corrupt .pbin header in {self._data_path}: token size "
+                    f"{self.token_size_in_bytes} not in (1, 2, 4)")
+            if HEADER_SIZE_IN_BYTES + self.data_len > file_size:
+                raise ValueError(
+                    f"Truncated .pbin file {self._data_path}: header claims "
+                    f"{self.data_len} data bytes but the file has only "
+                    f"{file_size - HEADER_SIZE_IN_BYTES} after the header")
             if load_index:
                 f.seek(HEADER_SIZE_IN_BYTES + self.data_len)
-                self._index_base: Optional[list[tuple[int, int]]] = pickle.loads(f.read())
+                try:
+                    self._index_base: Optional[list[tuple[int, int]]] = \
+                        pickle.loads(f.read())
+                except Exception as e:
+                    raise ValueError(
+                        f"Corrupt .pbin document index in {self._data_path} "
+                        f"(truncated file or bad header?): {e}") from e
             else:
                 self._index_base = None
         self._data = np.memmap(self._data_path, mode="r", offset=HEADER_SIZE_IN_BYTES,
