@@ -26,14 +26,16 @@ def read_last_checkpoint_info(experiment_checkpoint_root: Path) -> Path:
         return Path(json.load(f)["checkpoint_folder_path"])
 
 
-def read_checkpoint_meta(folder: Path) -> dict:
-    with open(Path(folder) / "meta.json", encoding="utf-8") as f:
+def read_checkpoint_meta(folder: Path, partition: str = "") -> dict:
+    sfx = f"_{partition}" if partition else ""
+    with open(Path(folder) / f"meta{sfx}.json", encoding="utf-8") as f:
         return json.load(f)
 
 
 class ShardedCheckpointLoading:
-    def __init__(self, global_rank: int):
+    def __init__(self, global_rank: int, partition: str = ""):
         self.global_rank = global_rank
+        self.partition = partition  # PP/TP model partition (see saving.py)
 
     @torch.no_grad()
     def load_checkpoint_(self, app_state: AppState, folder: Path) -> dict:
@@ -41,7 +43,7 @@ class ShardedCheckpointLoading:
         the saved world size differs. Returns the saved training-progress
         metadata dict."""
         folder = Path(folder)
-        meta = read_checkpoint_meta(folder)
+        meta = read_checkpoint_meta(folder, self.partition)
         saved_world = meta["world_size"]
         saved_layout = meta["shard_layout"]
 
@@ -79,7 +81,8 @@ class ShardedCheckpointLoading:
             if not self._rank_file_overlaps(model, saved_layout, saved_rank,
                                             saved_world, new_rank):
                 continue
-            shards = torch.load(folder / f"shards_rank_{saved_rank}.pt",
+            sfx = f"_{self.partition}" if self.partition else ""
+            shards = torch.load(folder / f"shards{sfx}_rank_{saved_rank}.pt",
                                 map_location="cpu", weights_only=True)
             for u in model.units:
                 lay = saved_layout[u.name]

@@ -42,28 +42,48 @@ def checkpoint_folder_name(experiment_id: str, progress: TrainingProgress) -> st
 
 
 class ShardedCheckpointSaving:
-    """Execution half: writes/deletes checkpoint folders."""
+    """Execution half: writes/deletes checkpoint folders.
+
+    ``partition`` namespaces a MODEL partition under PP/TP composition
+    (e.g. "pp0_tp1"): the engine's DP-shard resharding applies within one
+    partition; each partition carries its own shard layout and meta file
+    (reference analog: DCP's global FQN space handles this implicitly;
+    tests/end2end_tests/test_fsdp2_warmstart_pp_tp.py). The default ""
+    keeps the flat single-partition layout (meta.json / shards_rank_N.pt).
+
+    ``dp_rank``/``dp_world`` override the file-naming rank/world when the
+    engine's DP group is a subgroup of WORLD (PP/TP composition); the
+    default uses the global rank/world (pure-DP runs).
+    """
 
     def __init__(self, checkpoint_path: Path, experiment_id: str,
-                 global_rank: int):
+                 global_rank: int, partition: str = "",
+                 dp_rank: int = None, dp_world: int = None):
         self.checkpoint_path = Path(checkpoint_path)
         self.experiment_id = experiment_id
         self.global_rank = global_rank
+        self.partition = partition
+        self.dp_rank = dp_rank
+        self.dp_world = dp_world
 
     def _folder(self, progress: TrainingProgress) -> Path:
         return (self.checkpoint_path / self.experiment_id
                 / checkpoint_folder_name(self.experiment_id, progress))
 
+    def _suffix(self) -> str:
+        return f"_{self.partition}" if self.partition else ""
+
     @torch.no_grad()
     def save_checkpoint(self, app_state: AppState, progress: TrainingProgress):
         folder = self._folder(progress)
         folder.mkdir(parents=True, exist_ok=True)
-        world = dist.get_world_size() if is_dist() else 1
+        world = self.dp_world if self.dp_world is not None else             (dist.get_world_size() if is_dist() else 1)
+        rank = self.dp_rank if self.dp_rank is not None else self.global_rank
 
         shards = {k: v.detach().to("cpu") for k, v in app_state.shard_state().items()}
-        torch.save(shards, folder / f"shards_rank_{self.global_rank}.pt")
+        torch.save(shards, folder / f"shards{self._suffix()}_rank_{rank}.pt")
 
-        if self.global_rank == 0:
+        if rank == 0:
             meta = {
                 "world_size": world,
                 "shard_layout": app_state.shard_layout(),
@@ -75,8 +95,10 @@ class ShardedCheckpointSaving:
                     "num_target_tokens": progress.num_target_tokens,
                 },
             }
-            with open(folder / "meta.json", "w", encoding="utf-8") as f:
+            with open(folder / f"meta{self._suffix()}.json", "w",
+                      encoding="utf-8") as f:
                 json.dump(meta, f, indent=1)
+        if self.global_rank == 0:
             info = {"checkpoint_folder_path": str(folder.absolute())}
             with open(folder.parent / "last_checkpoint_info.json", "w",
                       encoding="utf-8") as f:

@@ -408,8 +408,13 @@ class XGMIShardedModel(nn.Module):
             unit.alloc_grad_views()
 
     def _finish_unit_backward(self, idx):
-        if idx not in self._units_needing_reduce:
-            return
+        # Reduce whenever this unit has accumulated grads — NOT only when
+        # it is still in _units_needing_reduce. With several forwards
+        # queued before their backwards (pipeline schedules), the set is
+        # discharged by the FIRST micro-batch's backward; gating the
+        # reduce on membership silently dropped every later micro-batch's
+        # gradient (caught by the PP warmstart-equivalence test).
+        # reduce_scatter_grads no-ops when no grads are allocated.
         self._units_needing_reduce.discard(idx)
         unit = self.units[idx]
         unit.reduce_scatter_grads(self.streams)
@@ -437,6 +442,10 @@ class XGMIShardedModel(nn.Module):
             self._finish_unit_backward(idx)
         for u in self.units:
             u.wait_reduce()
+            # per-backward hook counters must not leak into the next
+            # backward (a unit with hook-less params would otherwise carry
+            # residue that shifts reduce timing after a warmstart)
+            u.grads_seen = 0
         if self._replicate_group is not None:
             ws = dist.get_world_size(self._replicate_group)
             for u in self.units:

@@ -6,6 +6,8 @@ from a checkpoint must exactly equal the uninterrupted run. Adds
 cross-world-size resharding coverage the reference delegates to DCP.
 """
 
+from pathlib import Path
+
 import pytest
 import torch
 
@@ -232,3 +234,86 @@ def test_full_state_reassembly_single_process(tmp_path):
     load_full_model_state_from_checkpoint(folder, fresh)
     for name, p in fresh.named_parameters():
         torch.testing.assert_close(p.data, ref_sd[name], rtol=1e-6, atol=1e-7)
+
+
+# ---------------------------------------------------------------------------
+# Warmstart equivalence under PP x TP composition (reference
+# tests/end2end_tests/test_fsdp2_warmstart_pp_tp.py): the partitioned
+# checkpoint layout (one shard layout per pp/tp model partition).
+
+def _pp_tp_build(rank, world):
+    import torch.distributed as dist
+
+    from modalities_amd.parallel.mesh import DeviceMesh, ParallelismDegrees
+    from modalities_amd.parallel.pp import (get_pipeline_schedule,
+                                            split_model_into_stages)
+    from modalities_amd.parallel.tp import get_gpt2_tensor_parallelized_model
+
+    mesh = DeviceMesh(world_size=world, rank=rank, pp=2, tp=2, dp_shard=1)
+    pp = mesh.dims[ParallelismDegrees.PP]
+    tp = mesh.dims[ParallelismDegrees.TP]
+    dp = mesh.dims[ParallelismDegrees.DP_SHARD]
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    model = get_gpt2_tensor_parallelized_model(
+        model, group=tp.group, tp_rank=tp.rank, tp_size=tp.size)
+    stages = split_model_into_stages(model, pp.size)
+    engine = XGMIShardedModel.from_transformer(
+        stages[pp.rank], torch.device("cpu"), process_group=dp.group,
+        param_dtype=torch.float32, rank=dp.rank, world_size=dp.size)
+    opt = get_adam_w(engine, lr=1e-3, weight_decay=0.0)
+    sched = get_pipeline_schedule(
+        "1f1b", stage=stages[pp.rank], stage_idx=pp.rank, num_stages=pp.size,
+        n_microbatches=2, group=pp.group, sharded_engine=engine)
+    return mesh, pp, tp, engine, opt, sched
+
+
+def _pp_tp_train(sched, opt, step_ids):
+    from modalities_amd.loss_functions import CLMCrossEntropyLoss
+    loss_fn = CLMCrossEntropyLoss("target_ids", "logits")
+    means = []
+    for i in step_ids:
+        x, y = make_batch(100 + i)
+        losses = sched.step(x, y, loss_fn)
+        opt.step()
+        opt.zero_grad()
+        means.append(sched.broadcast_mean_loss(losses).item())
+    return means
+
+
+def _warmstart_pp_tp(rank, world, tmp_dir):
+    mesh, pp, tp, engine, opt, sched = _pp_tp_build(rank, world)
+    part = f"pp{pp.rank}tp{tp.rank}"
+
+    # uninterrupted reference
+    ref = _pp_tp_train(sched, opt, range(6))
+
+    # run A: 3 steps, checkpoint (each partition saves its own shards+meta)
+    mesh, pp, tp, engine, opt, sched = _pp_tp_build(rank, world)
+    a = _pp_tp_train(sched, opt, range(3))
+    app = AppState(engine, opt)
+    saving = ShardedCheckpointSaving(Path(tmp_dir), "exp_pptp",
+                                     global_rank=rank, partition=part,
+                                     dp_rank=0, dp_world=1)
+    saving.save_checkpoint(app, progress_at(3))
+
+    # run B: fresh build, load this partition, continue
+    mesh, pp, tp, engine2, opt2, sched2 = _pp_tp_build(rank, world)
+    engine2.units[0].master_shard.add_(0.5)  # prove the load overwrites
+    app2 = AppState(engine2, opt2)
+    folder = read_last_checkpoint_info(Path(tmp_dir) / "exp_pptp")
+    meta = ShardedCheckpointLoading(rank, partition=part).load_checkpoint_(
+        app2, folder)
+    assert meta["num_seen_steps"] == 3
+    b = _pp_tp_train(sched2, opt2, range(3, 6))
+    return ref, a + b
+
+
+def test_warmstart_loss_equivalence_pp2_tp2(tmp_path):
+    from tests.utils_dist import run_distributed
+    results = run_distributed(_warmstart_pp_tp, world_size=4,
+                              port=find_free_port(), args=(str(tmp_path),),
+                              timeout_s=300)
+    for r in range(4):
+        ref, resumed = results[r]
+        assert ref == pytest.approx(resumed, rel=1e-6), (r, ref, resumed)

@@ -93,7 +93,7 @@ def test_missing_reference_target():
         other: object
 
     factory = ComponentFactory(get_default_registry())
-    cfg = {"loss_fn": {"component_key": "loss", "variant_key": "clm_cross_entropy",
+    cfg = {"loss_fn": {"component_key": "loss", "variant_key": "clm_cross_entropy_loss",
                        "config": {"target_key": "t", "prediction_key": "p"}},
            "other": {"instance_key": "nonexistent", "pass_type": "BY_REFERENCE"}}
     with pytest.raises(KeyError, match="nonexistent"):

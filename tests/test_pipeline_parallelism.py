@@ -309,3 +309,87 @@ def test_pp2_eval_step_forward_only(variant):
                               port=find_free_port(), args=(variant,))
     for r in range(2):
         assert results[r] == pytest.approx(sum(ref_losses) / N_MB, rel=1e-5)
+
+
+# ---------------------------------------------------------------------------
+# TP2 x PP2 x DP2 world-8 composition (VERDICT r1 #2b: stress the engine's
+# collective ordering at the full 8-GPU-node world size on CPU/gloo).
+
+def _tp2_pp2_dp2_worker(rank, world):
+    import torch.distributed as dist
+
+    from modalities_amd.loss_functions import CLMCrossEntropyLoss
+    from modalities_amd.parallel.mesh import DeviceMesh, ParallelismDegrees
+    from modalities_amd.parallel.pp import get_pipeline_schedule
+    from modalities_amd.parallel.tp import get_gpt2_tensor_parallelized_model
+
+    mesh = DeviceMesh(world_size=world, rank=rank, pp=2, dp_shard=2, tp=2)
+    pp = mesh.dims[ParallelismDegrees.PP]
+    dp = mesh.dims[ParallelismDegrees.DP_SHARD]
+    tp = mesh.dims[ParallelismDegrees.TP]
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    # TP-shard IN PLACE first (patches block forwards), then PP-split: the
+    # stages reference the already-sharded blocks (reference order:
+    # model_factory TP plan -> PipelineFactory)
+    model = get_gpt2_tensor_parallelized_model(
+        model, group=tp.group, tp_rank=tp.rank, tp_size=tp.size)
+    stages = split_model_into_stages(model, pp.size)
+    stage = stages[pp.rank]
+    sched = get_pipeline_schedule(
+        "1f1b", stage=stage, stage_idx=pp.rank, num_stages=pp.size,
+        n_microbatches=2, group=pp.group)
+    x, y = make_batch(batch=4)
+    rows = slice(dp.rank * 2, dp.rank * 2 + 2)
+    loss_fn = CLMCrossEntropyLoss("target_ids", "logits")
+    losses = sched.step(x[rows], y[rows], loss_fn)
+    mean = sched.broadcast_mean_loss(losses)
+    dist.all_reduce(mean, group=dp.group)  # global mean across dp replicas
+    mean /= dp.size
+
+    out = {}
+    for n, p in stage.named_parameters():
+        if p.grad is None:
+            continue
+        if n in ("wte.weight", "lm_head.weight",
+                 "blocks.0.attn.q_attn.weight"):
+            g = p.grad.clone()
+            dist.all_reduce(g, group=dp.group)  # dp-mean (replicated grads)
+            out[n] = (g / dp.size).numpy()
+    return out, mean.item(), tp.rank, pp.rank
+
+
+def test_tp2_pp2_dp2_world8_matches_single_process():
+    torch.manual_seed(0)
+    ref_model = GPT2LLM(tiny_cfg())
+    x, y = make_batch(batch=4)
+    losses = []
+    for mb_x, mb_y in zip(x.chunk(2), y.chunk(2)):
+        out = ref_model({"input_ids": mb_x})["logits"]
+        loss = torch.nn.functional.cross_entropy(
+            out.reshape(-1, VOCAB).float(), mb_y.reshape(-1))
+        (loss / 2).backward()
+        losses.append(loss.item())
+    ref = {n: p.grad for n, p in ref_model.named_parameters()}
+    ref_mean = sum(losses) / 2
+
+    results = run_distributed(_tp2_pp2_dp2_worker, world_size=8,
+                              port=find_free_port(), timeout_s=300)
+    checked = 0
+    for r in range(8):
+        grads, mean, tp_rank, pp_rank = results[r]
+        assert mean == pytest.approx(ref_mean, rel=1e-5)
+        for name, g in grads.items():
+            got = torch.from_numpy(g)
+            if name == "blocks.0.attn.q_attn.weight":
+                # column-parallel: this tp rank holds rows slice tp_rank;
+                # stage-1 blocks renumber, so only compare on stage 0
+                if pp_rank != 0:
+                    continue
+                per = ref[name].shape[0] // 2
+                exp = ref[name][tp_rank * per:(tp_rank + 1) * per]
+            else:
+                exp = ref[name]
+            torch.testing.assert_close(got, exp, rtol=1e-4, atol=1e-6)
+            checked += 1
+    assert checked >= 8
