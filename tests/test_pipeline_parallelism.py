@@ -393,3 +393,81 @@ def test_tp2_pp2_dp2_world8_matches_single_process():
             torch.testing.assert_close(got, exp, rtol=1e-4, atol=1e-6)
             checked += 1
     assert checked >= 8
+
+
+# ---------------------------------------------------------------------------
+# Generic FQN splitter (model-agnostic splitting; reference
+# pipeline_parallelism.py:131-277 + stages_generator.py:15-120)
+
+def test_fqn_split_gpt2_equals_full_model():
+    from modalities_amd.parallel.pp_split import split_model_into_stages_by_fqn
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    stages = split_model_into_stages_by_fqn(model, 2)
+    assert stages[0].is_first and stages[1].is_last
+    x, _ = make_batch()
+    with torch.no_grad():
+        full = model({"input_ids": x})["logits"]
+        chained = stages[1](stages[0](x))
+    torch.testing.assert_close(full, chained)
+    # every segment lives on exactly one stage; balanced by param count
+    all_fqns = stages[0].fqns + stages[1].fqns
+    assert all_fqns[0] == "embed" and all_fqns[-1] == "head"
+    assert len(set(all_fqns)) == len(all_fqns) == 4 + 2
+
+
+def test_fqn_split_explicit_fqns_and_errors():
+    from modalities_amd.parallel.pp_split import split_model_into_stages_by_fqn
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    stages = split_model_into_stages_by_fqn(
+        model, 2, stage_fqns=[["embed", "blocks.0"],
+                              ["blocks.1", "blocks.2", "blocks.3", "head"]])
+    x, _ = make_batch()
+    with torch.no_grad():
+        full = model({"input_ids": x})["logits"]
+        torch.testing.assert_close(stages[1](stages[0](x)), full)
+    with pytest.raises(KeyError, match="does_not_exist"):
+        split_model_into_stages_by_fqn(model, 2,
+                                       stage_fqns=[["does_not_exist"], ["head"]])
+
+
+def test_fqn_split_arbitrary_sequential():
+    """Any nn.Sequential splits without model-specific code."""
+    from modalities_amd.parallel.pp_split import split_model_into_stages_by_fqn
+    torch.manual_seed(1)
+    model = torch.nn.Sequential(
+        torch.nn.Linear(64, 64), torch.nn.GELU(), torch.nn.Linear(64, 64),
+        torch.nn.GELU(), torch.nn.Linear(64, 64), torch.nn.GELU(),
+        torch.nn.Linear(64, 64))
+    stages = split_model_into_stages_by_fqn(model, 2)
+    x = torch.randn(3, 64)
+    with torch.no_grad():
+        torch.testing.assert_close(stages[1](stages[0](x)), model(x))
+    # balance by params: 2 of the 4 equal linears per stage
+    w = [sum(p.numel() for p in s.parameters()) for s in stages]
+    assert max(w) / max(1, min(w)) < 1.5, w
+
+
+def _pp_fqn_worker(rank, world):
+    import torch.distributed as dist
+
+    from modalities_amd.loss_functions import CLMCrossEntropyLoss
+    from modalities_amd.parallel.pp import get_pipeline_schedule
+    from modalities_amd.parallel.pp_split import split_model_into_stages_by_fqn
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    stages = split_model_into_stages_by_fqn(model, world)
+    sched = get_pipeline_schedule(
+        "1f1b", stage=stages[rank], stage_idx=rank, num_stages=world,
+        n_microbatches=N_MB, group=dist.group.WORLD)
+    x, y = make_batch()
+    losses = sched.step(x, y, CLMCrossEntropyLoss("target_ids", "logits"))
+    return [l.item() for l in losses]
+
+
+def test_pp2_fqn_stages_match_single_process():
+    ref_losses, _ = reference_loss_and_grads()
+    results = run_distributed(_pp_fqn_worker, world_size=2,
+                              port=find_free_port())
+    assert results[1] == pytest.approx(ref_losses, rel=1e-5)
