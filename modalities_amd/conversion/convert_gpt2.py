@@ -82,8 +82,13 @@ def convert_model_checkpoint(modalities_model: GPT2LLM):
             sd[f"{hp}.self_attn.v_proj.weight"] = src[f"{p}.attn.v_attn.weight"]
         sd[f"{hp}.self_attn.o_proj.weight"] = src[f"{p}.attn.c_proj.weight"]
         sd[f"{hp}.post_attention_layernorm.weight"] = src[f"{p}.ffn_norm.weight"]
-        sd[f"{hp}.mlp.gate_proj.weight"] = src[f"{p}.mlp.W.weight"]
-        sd[f"{hp}.mlp.up_proj.weight"] = src[f"{p}.mlp.V.weight"]
+        if f"{p}.mlp.Wv.weight" in src:  # packed layout: gate|up halves
+            wv = src[f"{p}.mlp.Wv.weight"]
+            sd[f"{hp}.mlp.gate_proj.weight"] = wv[:wv.shape[0] // 2]
+            sd[f"{hp}.mlp.up_proj.weight"] = wv[wv.shape[0] // 2:]
+        else:
+            sd[f"{hp}.mlp.gate_proj.weight"] = src[f"{p}.mlp.W.weight"]
+            sd[f"{hp}.mlp.up_proj.weight"] = src[f"{p}.mlp.V.weight"]
         sd[f"{hp}.mlp.down_proj.weight"] = src[f"{p}.mlp.W_2.weight"]
         if cfg.bias:
             for ours, theirs in [("q_attn", "q_proj"), ("k_attn", "k_proj"),

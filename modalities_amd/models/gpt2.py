@@ -78,6 +78,8 @@ class GPT2LLMConfig(BaseModel):
     bias: bool = False
     attention_implementation: AttentionImplementation = AttentionImplementation.HIP_FLASH
     activation_type: ActivationType = ActivationType.SWIGLU
+    # MI355X-first: pack the SwiGLU W/V up-projections into one GEMM
+    packed_swiglu: bool = True
     qkv_transform: QueryKeyValueTransformType = QueryKeyValueTransformType.ROTARY
     rope_base: float = 10000.0
     use_weight_tying: bool = False
@@ -214,7 +216,8 @@ class GPT2Block(nn.Module):
             cfg.attention_norm_config, fused_qkv=cfg.fused_qkv)
         self.ffn_norm = make_norm(cfg.ffn_norm_config, cfg.n_embd)
         if cfg.activation_type == ActivationType.SWIGLU:
-            self.mlp = SwiGLU(cfg.n_embd, cfg.ffn_hidden, cfg.bias)
+            self.mlp = SwiGLU(cfg.n_embd, cfg.ffn_hidden, cfg.bias,
+                              packed=cfg.packed_swiglu)
         else:
             self.mlp = TransformerMLP(cfg.n_embd, cfg.ffn_hidden, cfg.bias, cfg.dropout)
 

@@ -8,7 +8,7 @@ import torch
 import torch.nn as nn
 
 from modalities_amd.batch import DatasetBatch, InferenceResultBatch
-from modalities_amd.ops import silu_mul
+from modalities_amd.ops import silu_mul, silu_mul_joint
 from modalities_amd.ops.linear import TwoStreamLinear
 
 WeightDecayGroups = dict[str, list[str]]
@@ -40,13 +40,23 @@ class NNModel(nn.Module):
 class SwiGLU(nn.Module):
     """SwiGLU MLP: W_2(silu(W x) * V x), hidden = 256-rounded 2/3 * 4 * d
     for even TP sharding (reference: src/modalities/models/model.py:116-142).
-    The silu*mul epilogue runs as a single HIP kernel on device (K6)."""
 
-    def __init__(self, n_embd: int, ffn_hidden: int, bias: bool = False):
+    MI355X-first layout: ``packed=True`` fuses the W and V up-projections
+    into ONE [n_embd, 2*hidden] GEMM (half the up-projection launches at a
+    better hipBLASLt shape); the silu*mul epilogue reads the joint output
+    and its backward writes one joint gradient buffer (K6 joint kernel).
+    ``packed=False`` keeps the reference's separate W/V parameters."""
+
+    def __init__(self, n_embd: int, ffn_hidden: int, bias: bool = False,
+                 packed: bool = True):
         super().__init__()
         self.hidden_dim = self._get_hidden_dim(ffn_hidden)
-        self.W = TwoStreamLinear(n_embd, self.hidden_dim, bias=bias)
-        self.V = TwoStreamLinear(n_embd, self.hidden_dim, bias=bias)
+        self.packed = packed
+        if packed:
+            self.Wv = TwoStreamLinear(n_embd, 2 * self.hidden_dim, bias=bias)
+        else:
+            self.W = TwoStreamLinear(n_embd, self.hidden_dim, bias=bias)
+            self.V = TwoStreamLinear(n_embd, self.hidden_dim, bias=bias)
         self.W_2 = TwoStreamLinear(self.hidden_dim, n_embd, bias=bias)
 
     @staticmethod
@@ -54,7 +64,19 @@ class SwiGLU(nn.Module):
         # 2/3 * ffn_hidden, rounded up to a multiple of 256.
         return 256 * ((int(2 * ffn_hidden / 3) + 256 - 1) // 256)
 
+    # gate/up views of the packed weight (conversion / tests / TP introspect
+    # the reference's separate-W/V naming through these)
+    @property
+    def W_weight(self) -> torch.Tensor:
+        return self.Wv.weight[:self.hidden_dim] if self.packed else self.W.weight
+
+    @property
+    def V_weight(self) -> torch.Tensor:
+        return self.Wv.weight[self.hidden_dim:] if self.packed else self.V.weight
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.packed:
+            return self.W_2(silu_mul_joint(self.Wv(x)))
         return self.W_2(silu_mul(self.W(x), self.V(x)))
 
 

@@ -89,7 +89,8 @@ GPT2_PLAIN_REGEXES = [
     r".*wte\.weight", r".*wpe\.weight",
     r".*q_attn\.weight", r".*k_attn\.weight", r".*v_attn\.weight",
     r".*qkv_attn\.weight",
-    r".*mlp\.W\.weight", r".*mlp\.V\.weight", r".*mlp\.c_fc\.weight",
+    r".*mlp\.W\.weight", r".*mlp\.V\.weight", r".*mlp\.Wv\.weight",
+    r".*mlp\.c_fc\.weight",
     r".*lm_head\.weight",
 ]
 GPT2_PROJECTION_REGEXES = [

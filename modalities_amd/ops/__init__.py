@@ -25,7 +25,7 @@ from modalities_amd.ops.backend import (  # noqa: F401
 )
 from modalities_amd.ops.rms_norm import rms_norm  # noqa: F401
 from modalities_amd.ops.rope import precompute_rope_cos_sin, rope_apply  # noqa: F401
-from modalities_amd.ops.swiglu import silu_mul  # noqa: F401
+from modalities_amd.ops.swiglu import silu_mul, silu_mul_joint  # noqa: F401
 from modalities_amd.ops.cross_entropy import fused_cross_entropy  # noqa: F401
 from modalities_amd.ops.attention import flash_attention  # noqa: F401
 from modalities_amd.ops.adamw import fused_adamw_step  # noqa: F401

@@ -14,6 +14,8 @@ void rope_bwd_slice(torch::Tensor dx, torch::Tensor cos_t, torch::Tensor sin_t,
 torch::Tensor silu_mul_fwd(torch::Tensor g, torch::Tensor u);
 std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dout, torch::Tensor g,
                                         torch::Tensor u);
+torch::Tensor silu_mul_joint_fwd(torch::Tensor wv);
+torch::Tensor silu_mul_joint_bwd(torch::Tensor dout, torch::Tensor wv);
 std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
                                              torch::Tensor targets,
                                              long ignore_index);
@@ -91,6 +93,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "inverse-RoPE a [B,T,H,D] grad into a fused-QKV grad buffer slice");
   m.def("silu_mul_fwd", &silu_mul_fwd, "silu(g)*u forward (K6)");
   m.def("silu_mul_bwd", &silu_mul_bwd, "silu(g)*u backward (K6)");
+  m.def("silu_mul_joint_fwd", &silu_mul_joint_fwd,
+        "silu/mul over a packed [.., 2H] gate|up tensor (K6 joint)");
+  m.def("silu_mul_joint_bwd", &silu_mul_joint_bwd,
+        "joint-layout silu/mul backward into one dwv buffer");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE forward (K8)");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE backward (K8)");
   m.def("fused_adamw_masked", &fused_adamw_masked,

@@ -112,6 +112,62 @@ __global__ void silu_mul_bwd_kernel(const unsigned short* __restrict__ dout,
   }
 }
 
+// Joint-layout variant: the gate/up halves live in ONE [rows, 2H] tensor
+// (the packed SwiGLU up-projection output: a single GEMM instead of two).
+// fwd reads both halves of a row; bwd writes the joint gradient buffer
+// directly (no cat before the packed GEMM backward).
+__global__ void silu_mul_joint_fwd_kernel(const unsigned short* __restrict__ wv,
+                                          unsigned short* __restrict__ out,
+                                          long n8, long h8) {
+  shortx8* ov = reinterpret_cast<shortx8*>(out);
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / h8, c8 = i % h8;
+    const shortx8 gg = *reinterpret_cast<const shortx8*>(
+        wv + (row * 2 * h8 + c8) * 8);
+    const shortx8 uu = *reinterpret_cast<const shortx8*>(
+        wv + (row * 2 * h8 + h8 + c8) * 8);
+    shortx8 oo;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf16_to_f32((unsigned short)gg[j]);
+      float uf = bf16_to_f32((unsigned short)uu[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      oo[j] = (short)f32_to_bf16(gf * sig * uf);
+    }
+    ov[i] = oo;
+  }
+}
+
+__global__ void silu_mul_joint_bwd_kernel(const unsigned short* __restrict__ dout,
+                                          const unsigned short* __restrict__ wv,
+                                          unsigned short* __restrict__ dwv,
+                                          long n8, long h8) {
+  const shortx8* dv = reinterpret_cast<const shortx8*>(dout);
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / h8, c8 = i % h8;
+    const long gbase = (row * 2 * h8 + c8) * 8;
+    const long ubase = (row * 2 * h8 + h8 + c8) * 8;
+    const shortx8 gg = *reinterpret_cast<const shortx8*>(wv + gbase);
+    const shortx8 uu = *reinterpret_cast<const shortx8*>(wv + ubase);
+    shortx8 dd = dv[i], og, ou;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float df = bf16_to_f32((unsigned short)dd[j]);
+      float gf = bf16_to_f32((unsigned short)gg[j]);
+      float uf = bf16_to_f32((unsigned short)uu[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      float silu = gf * sig;
+      float dsilu = sig * (1.f + gf * (1.f - sig));
+      og[j] = (short)f32_to_bf16(df * uf * dsilu);
+      ou[j] = (short)f32_to_bf16(df * silu);
+    }
+    *reinterpret_cast<shortx8*>(dwv + gbase) = og;
+    *reinterpret_cast<shortx8*>(dwv + ubase) = ou;
+  }
+}
+
 int grid_for(long work, int block) {
   long g = (work + block - 1) / block;
   return (int)min(g, (long)(256 * 8));  // cap + grid-stride (guide G11)
@@ -214,4 +270,40 @@ std::vector<torch::Tensor> silu_mul_bwd(torch::Tensor dout, torch::Tensor g,
                      (unsigned short*)du.data_ptr(), n8);
   HIP_CHECK_KERNEL();
   return {dg, du};
+}
+
+
+torch::Tensor silu_mul_joint_fwd(torch::Tensor wv) {
+  TORCH_CHECK(wv.is_cuda() && wv.dtype() == torch::kBFloat16 && wv.is_contiguous());
+  const long H = wv.size(-1) / 2;
+  TORCH_CHECK(wv.size(-1) % 2 == 0 && H % 8 == 0,
+              "silu_mul_joint: last dim must be 2*H with H %% 8 == 0");
+  auto sizes = wv.sizes().vec();
+  sizes.back() = H;
+  auto out = torch::empty(sizes, wv.options());
+  const long rows = wv.numel() / (2 * H);
+  const long n8 = rows * (H / 8);
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(silu_mul_joint_fwd_kernel, dim3(grid_for(n8, 256)),
+                     dim3(256), 0, stream,
+                     (const unsigned short*)wv.data_ptr(),
+                     (unsigned short*)out.data_ptr(), n8, H / 8);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
+torch::Tensor silu_mul_joint_bwd(torch::Tensor dout, torch::Tensor wv) {
+  TORCH_CHECK(dout.is_cuda() && dout.is_contiguous() && wv.is_contiguous());
+  const long H = wv.size(-1) / 2;
+  auto dwv = torch::empty_like(wv);
+  const long rows = wv.numel() / (2 * H);
+  const long n8 = rows * (H / 8);
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(silu_mul_joint_bwd_kernel, dim3(grid_for(n8, 256)),
+                     dim3(256), 0, stream,
+                     (const unsigned short*)dout.data_ptr(),
+                     (const unsigned short*)wv.data_ptr(),
+                     (unsigned short*)dwv.data_ptr(), n8, H / 8);
+  HIP_CHECK_KERNEL();
+  return dwv;
 }

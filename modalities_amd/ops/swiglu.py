@@ -28,3 +28,28 @@ def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
         return _SiluMulHip.apply(gate, up)
     gf = gate.float()
     return (torch.nn.functional.silu(gf) * up.float()).to(gate.dtype)
+
+
+class _SiluMulJointHip(torch.autograd.Function):
+    """Joint-layout variant: wv = [.., 2H] (gate | up halves from ONE
+    packed up-projection GEMM); backward produces the joint dwv buffer the
+    packed GEMM backward consumes directly (no torch.cat)."""
+
+    @staticmethod
+    def forward(ctx, wv):
+        wv = wv.contiguous()
+        ctx.save_for_backward(wv)
+        return hip_ext().silu_mul_joint_fwd(wv)
+
+    @staticmethod
+    def backward(ctx, dout):
+        (wv,) = ctx.saved_tensors
+        return hip_ext().silu_mul_joint_bwd(dout.contiguous(), wv)
+
+
+def silu_mul_joint(wv: torch.Tensor) -> torch.Tensor:
+    """out = silu(wv[.., :H]) * wv[.., H:] for a packed [.., 2H] tensor."""
+    if use_hip(wv):
+        return _SiluMulJointHip.apply(wv)
+    h = wv.shape[-1] // 2
+    return silu_mul(wv[..., :h].contiguous(), wv[..., h:].contiguous())

@@ -248,6 +248,27 @@ def _shard_linear_(lin: nn.Linear, tp_rank: int, tp_size: int, dim: int) -> None
 
 
 @torch.no_grad()
+def _shard_packed_rows_(lin, tp_rank: int, tp_size: int,
+                        block_sizes: tuple) -> None:
+    """Shard a row-packed weight [sum(blocks), h] where each logical block
+    (e.g. SwiGLU gate|up) must be sharded separately and re-packed."""
+    w = lin.weight
+    rows, base = [], 0
+    brows = []
+    for bs in block_sizes:
+        per = bs // tp_size
+        rows.append(w[base + tp_rank * per: base + (tp_rank + 1) * per])
+        brows.append((base, per))
+        base += bs
+    lin.weight = nn.Parameter(torch.cat(rows, dim=0).clone())
+    if lin.bias is not None:
+        bs_ = [lin.bias[b + tp_rank * per: b + (tp_rank + 1) * per]
+               for b, per in brows]
+        lin.bias = nn.Parameter(torch.cat(bs_, dim=0).clone())
+    lin.out_features = sum(b // tp_size for b in block_sizes)
+
+
+@torch.no_grad()
 def _shard_fused_qkv_(lin, tp_rank: int, tp_size: int, Cq: int, Ckv: int):
     """Shard the joint QKV weight [Cq+2*Ckv, h] per head group: each rank
     keeps its q rows, k rows and v rows re-packed as a contiguous
@@ -334,12 +355,15 @@ class _TPSwiGLUForward:
         self.sp = sp
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
-        from modalities_amd.ops import silu_mul
+        from modalities_amd.ops import silu_mul, silu_mul_joint
         if self.sp:
             x = gather_seq(x, self.group, dim=1)
         else:
             x = copy_to_tp(x, self.group)
-        y = self.mlp.W_2(silu_mul(self.mlp.W(x), self.mlp.V(x)))
+        if getattr(self.mlp, "packed", False):
+            y = self.mlp.W_2(silu_mul_joint(self.mlp.Wv(x)))
+        else:
+            y = self.mlp.W_2(silu_mul(self.mlp.W(x), self.mlp.V(x)))
         if self.sp:
             return scatter_seq(y, self.group, dim=1)
         return reduce_from_tp(y, self.group)
@@ -413,8 +437,15 @@ def get_gpt2_tensor_parallelized_model(model, device_mesh=None, group=None,
             if mlp.hidden_dim % tp_size:
                 raise ValueError(f"SwiGLU hidden_dim {mlp.hidden_dim} not "
                                  f"divisible by tp={tp_size}")
-            _shard_linear_(mlp.W, tp_rank, tp_size, dim=0)
-            _shard_linear_(mlp.V, tp_rank, tp_size, dim=0)
+            if getattr(mlp, "packed", False):
+                # joint [2H, h] weight: shard gate and up halves separately
+                # and re-pack (the joint GEMM stays single per rank)
+                _shard_packed_rows_(mlp.Wv, tp_rank, tp_size,
+                                    (mlp.hidden_dim, mlp.hidden_dim))
+            else:
+                _shard_linear_(mlp.W, tp_rank, tp_size, dim=0)
+                _shard_linear_(mlp.V, tp_rank, tp_size, dim=0)
+            mlp.hidden_dim //= tp_size
             _shard_linear_(mlp.W_2, tp_rank, tp_size, dim=1)
             block.mlp.forward = _TPSwiGLUForward(mlp, group, sequence_parallel)
         else:  # GELU MLP

@@ -467,3 +467,19 @@ def test_fused_qkv_rope_attention_matches_unfused():
 
     torch.testing.assert_close(y_f, y_u, rtol=0, atol=0)
     torch.testing.assert_close(a.grad, b.grad, rtol=0, atol=0)
+
+
+def test_silu_mul_joint_gpu():
+    from modalities_amd.ops.swiglu import silu_mul_joint
+    torch.manual_seed(12)
+    wv = bf(torch.randn(4, 8, 256, device=DEV)).requires_grad_(True)
+    y = silu_mul_joint(wv)
+    g, u = wv[..., :128].float(), wv[..., 128:].float()
+    ref = torch.nn.functional.silu(g) * u
+    torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
+    do = bf(torch.randn_like(y))
+    y.backward(do)
+    wv2 = wv.detach().float().requires_grad_(True)
+    g2, u2 = wv2[..., :128], wv2[..., 128:]
+    (torch.nn.functional.silu(g2) * u2).backward(do.float())
+    torch.testing.assert_close(wv.grad.float(), wv2.grad, rtol=3e-2, atol=3e-2)

@@ -173,3 +173,31 @@ def test_fused_qkv_matches_unfused():
     unfused.eval(), fused.eval()
     o1, o2 = run_fwd(unfused), run_fwd(fused)
     torch.testing.assert_close(o1, o2, rtol=1e-4, atol=1e-5)
+
+
+def test_packed_swiglu_matches_unpacked():
+    """packed=True (one joint W|V GEMM + joint silu-mul) must match the
+    reference separate-W/V layout bitwise given the same weights."""
+    import torch
+
+    from modalities_amd.models.model import SwiGLU
+    torch.manual_seed(3)
+    ref = SwiGLU(32, 128, packed=False)
+    packed = SwiGLU(32, 128, packed=True)
+    with torch.no_grad():
+        packed.Wv.weight[:packed.hidden_dim].copy_(ref.W.weight)
+        packed.Wv.weight[packed.hidden_dim:].copy_(ref.V.weight)
+        packed.W_2.weight.copy_(ref.W_2.weight)
+    x = torch.randn(2, 5, 32, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    y1, y2 = ref(x), packed(x2)
+    torch.testing.assert_close(y1, y2)
+    y1.sum().backward()
+    y2.sum().backward()
+    torch.testing.assert_close(x.grad, x2.grad)
+    torch.testing.assert_close(packed.Wv.weight.grad[:packed.hidden_dim],
+                               ref.W.weight.grad)
+    torch.testing.assert_close(packed.Wv.weight.grad[packed.hidden_dim:],
+                               ref.V.weight.grad)
+    torch.testing.assert_close(packed.W_weight, ref.W.weight)
+    torch.testing.assert_close(packed.V_weight, ref.V.weight)

@@ -58,7 +58,7 @@ def _tp_worker(rank, world, sequence_parallel):
     shapes = {
         "q_attn": tuple(model.blocks[0].attn.q_attn.weight.shape),
         "c_proj": tuple(model.blocks[0].attn.c_proj.weight.shape),
-        "W": tuple(model.blocks[0].mlp.W.weight.shape),
+        "W": tuple(model.blocks[0].mlp.W_weight.shape),
         "W_2": tuple(model.blocks[0].mlp.W_2.weight.shape),
     }
     return (out.detach().numpy(), loss.item(), shapes, g_q.numpy(), g_wte.numpy())
