@@ -201,3 +201,25 @@ def test_packed_swiglu_matches_unpacked():
                                ref.V.weight.grad)
     torch.testing.assert_close(packed.W_weight, ref.W.weight)
     torch.testing.assert_close(packed.V_weight, ref.V.weight)
+
+
+def test_attention_dropout_falls_back_to_sdpa():
+    """dropout > 0 with the HIP flash impl routes the attention op to the
+    SDPA path instead of erroring (reference flash paths support dropout)."""
+    import warnings
+
+    import torch
+
+    from modalities_amd.models.gpt2 import (AttentionImplementation, GPT2LLM,
+                                            GPT2LLMConfig)
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        model = GPT2LLM(GPT2LLMConfig(
+            vocab_size=64, n_layer=1, n_head_q=2, n_head_kv=2, n_embd=32,
+            ffn_hidden=128, sequence_length=16, dropout=0.1))
+    assert any("SDPA" in str(x.message) for x in w)
+    assert model.blocks[0].attn.attention_impl == \
+        AttentionImplementation.PYTORCH_FLASH
+    model.eval()
+    out = model({"input_ids": torch.randint(0, 64, (2, 16))})
+    assert out["logits"].shape == (2, 16, 64)
