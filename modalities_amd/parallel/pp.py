@@ -490,3 +490,60 @@ def get_pipeline_schedule(variant: str, **kwargs) -> PipelineSchedule:
     except KeyError:
         raise ValueError(f"Unknown PP schedule {variant!r}; "
                          f"have {sorted(SCHEDULES)}") from None
+
+
+# ---------------------------------------------------------------------------
+# config-driven construction (reference PipelineFactory +
+# ComponentSelectorFromPipeline, pipeline_parallelism.py:75-129)
+# ---------------------------------------------------------------------------
+
+def get_staged_pipeline_schedule(model, device_mesh, variant: str = "1f1b",
+                                 n_microbatches: int = 1,
+                                 input_weight: float = 1.0,
+                                 output_weight: float = 1.0,
+                                 use_fqn_split: bool = False,
+                                 num_chunks: int = 1, device=None):
+    """Registry factory: split `model` over the mesh's PP dimension and
+    return the schedule (the stage is reachable as `.stage` /
+    `.stages` for the optimizer/engine wiring; see
+    get_stage_from_schedule)."""
+    from modalities_amd.parallel.mesh import ParallelismDegrees
+    pp = device_mesh.dims[ParallelismDegrees.PP]
+    if pp.size == 1:
+        raise ValueError("staged pipeline requested but pp degree is 1")
+    if variant.lower() == "interleaved":
+        if num_chunks < 2:
+            raise ValueError("interleaved schedule needs num_chunks >= 2")
+        stages_all = split_model_into_stages(model, pp.size * num_chunks,
+                                             input_weight, output_weight)
+        mine = [stages_all[i]
+                for i in interleaved_stage_ids(pp.rank, pp.size, num_chunks)]
+        return ScheduleInterleaved1F1B(mine, pp_rank=pp.rank, pp_size=pp.size,
+                                       n_microbatches=n_microbatches,
+                                       group=pp.group, device=device)
+    if use_fqn_split:
+        from modalities_amd.parallel.pp_split import \
+            split_model_into_stages_by_fqn
+        stages = split_model_into_stages_by_fqn(model, pp.size)
+    else:
+        stages = split_model_into_stages(model, pp.size, input_weight,
+                                         output_weight)
+    return get_pipeline_schedule(
+        variant, stage=stages[pp.rank], stage_idx=pp.rank,
+        num_stages=pp.size, n_microbatches=n_microbatches, group=pp.group,
+        device=device)
+
+
+def get_stage_from_schedule(pp_schedule):
+    """Component selector: the local stage module (for optimizer / engine
+    wiring in YAML; reference ComponentSelectorFromPipeline)."""
+    if hasattr(pp_schedule, "stages"):
+        import torch.nn as nn
+
+        class _Chunks(nn.Module):
+            def __init__(self, stages):
+                super().__init__()
+                self.chunks = nn.ModuleList(stages)
+
+        return _Chunks(pp_schedule.stages)
+    return pp_schedule.stage

@@ -45,6 +45,9 @@ from modalities_amd.optimizers.lr_schedulers import (DummyLRScheduler,
 from modalities_amd.optimizers.optimizer_factory import get_adam_w
 from modalities_amd.parallel.mesh import get_device_mesh
 from modalities_amd.parallel.cp import get_gpt2_context_parallel_model
+from modalities_amd.parallel.pp import (
+    get_staged_pipeline_schedule as _pp_get_staged_pipeline_schedule,
+    get_stage_from_schedule as _pp_get_stage_from_schedule)
 from modalities_amd.parallel.tp import get_gpt2_tensor_parallelized_model
 from modalities_amd.registry.registry import ComponentEntity, Registry
 from modalities_amd.tokenization.tokenizer_wrapper import (CharTokenizer,
@@ -204,6 +207,10 @@ COMPONENTS: list[ComponentEntity] = [
                     get_gpt2_tensor_parallelized_model, None),
     ComponentEntity("context_parallelized_model", "gpt2_cp",
                     get_gpt2_context_parallel_model, None),
+    ComponentEntity("pp_schedule", "staged",
+                    _pp_get_staged_pipeline_schedule, None),
+    ComponentEntity("pipelined_model", "selector",
+                    _pp_get_stage_from_schedule, None),
     # training aux
     ComponentEntity("gradient_clipper", "fsdp2", GradientClipper, None),
     ComponentEntity("gradient_clipper", "default", GradientClipper, None),
