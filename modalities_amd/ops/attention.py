@@ -112,7 +112,9 @@ class _FusedQKVRopeAttn(torch.autograd.Function):
         qkv = qkv.contiguous()
         q = ext.rope_fwd_slice(qkv, cos, sin, 0, Hq, D, False)
         k = ext.rope_fwd_slice(qkv, cos, sin, C, Hkv, D, False)
-        v = qkv[..., C + KV:].contiguous().view(B, T, Hkv, D)
+        # V stays a row-strided view into the joint activation: the v2
+        # kernels read it in place (no per-layer 2*B*T*KV copy)
+        v = qkv[..., C + KV:].view(B, T, Hkv, D)
         o, lse = ext.attn_fwd(q, k, v, True, 0)
         ctx.save_for_backward(q, k, v, o, lse, cos, sin)
         ctx.dims = (Hq, Hkv, D)

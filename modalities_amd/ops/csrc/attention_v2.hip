@@ -76,8 +76,12 @@ struct FwdV2 {
     shortx8 kk[NCH], vv[NCH];
   };
 
+  // vp = V row pitch (elements between consecutive (b,t) rows): Hkv*D for
+  // a standalone tensor, Cq+2*Ckv when V is read in place from the joint
+  // QKV activation (kills the per-layer V-slice copy).
   static __device__ void stage_load(Stage& r, const unsigned short* k,
                                     const unsigned short* v, long kv_base,
+                                    long v_base, long vp,
                                     int kv0, int Tkv, int Hkv) {
     const int tid = threadIdx.x;
 #pragma unroll
@@ -89,7 +93,7 @@ struct FwdV2 {
         r.kk[it] = *reinterpret_cast<const shortx8*>(
             k + kv_base + (long)(kv0 + r_) * Hkv * D + c);
         r.vv[it] = *reinterpret_cast<const shortx8*>(
-            v + kv_base + (long)(kv0 + r_) * Hkv * D + c);
+            v + v_base + (long)(kv0 + r_) * vp + c);
       } else {
 #pragma unroll
         for (int m = 0; m < 8; ++m) { r.kk[it][m] = 0; r.vv[it][m] = 0; }
@@ -112,7 +116,7 @@ struct FwdV2 {
 
   static __device__ void run(const unsigned short* __restrict__ q,
                              const unsigned short* __restrict__ k,
-                             const unsigned short* __restrict__ v,
+                             const unsigned short* __restrict__ v, long vp,
                              unsigned short* __restrict__ o,
                              float* __restrict__ lse, int B, int Tq, int Tkv,
                              int q_off, int Hq, int Hkv, float scale,
@@ -130,6 +134,7 @@ struct FwdV2 {
 
     const long q_base = (((long)b * Tq) * Hq + h) * D;
     const long kv_base = (((long)b * Tkv) * Hkv + hkv) * D;
+    const long v_base = ((long)b * Tkv) * vp + (long)hkv * D;
     const int qg = qblk0 + wid * QBLK + ln31;  // this lane's LOCAL q row
     const int qgl = qg + q_off;                // global vs keys
     const float scale2 = scale * 1.44269504f;  // exp2 fold
@@ -170,10 +175,10 @@ struct FwdV2 {
     const int q_lo_wave = qblk0 + wid * QBLK + q_off;
 
     Stage st;
-    stage_load(st, k, v, kv_base, 0, Tkv, Hkv);
+    stage_load(st, k, v, kv_base, v_base, vp, 0, Tkv, Hkv);
     stage_write(st, sm);
     __syncthreads();
-    if (n_tiles > 1) stage_load(st, k, v, kv_base, KVBLK, Tkv, Hkv);
+    if (n_tiles > 1) stage_load(st, k, v, kv_base, v_base, vp, KVBLK, Tkv, Hkv);
 
     // tile-invariant transpose-read base (single LDS buffer)
     const int lam = ln31 & 15;
@@ -295,7 +300,8 @@ struct FwdV2 {
       if (tile + 1 < n_tiles) {
         stage_write(st, sm);  // tile+1 regs -> LDS
         if (tile + 2 < n_tiles)
-          stage_load(st, k, v, kv_base, (tile + 2) * KVBLK, Tkv, Hkv);
+          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * KVBLK,
+                     Tkv, Hkv);
         __syncthreads();      // tile+1 ready
       }
     }
@@ -355,6 +361,7 @@ struct DQV2 {
 
   static __device__ void stage_load(Stage& r, const unsigned short* k,
                                     const unsigned short* v, long kv_base,
+                                    long v_base, long vp,
                                     int kv0, int Tkv, int Hkv) {
     const int tid = threadIdx.x;
 #pragma unroll
@@ -366,7 +373,7 @@ struct DQV2 {
         r.kk[it] = *reinterpret_cast<const shortx8*>(
             k + kv_base + (long)(kv0 + r_) * Hkv * D + c);
         r.vv[it] = *reinterpret_cast<const shortx8*>(
-            v + kv_base + (long)(kv0 + r_) * Hkv * D + c);
+            v + v_base + (long)(kv0 + r_) * vp + c);
       } else {
 #pragma unroll
         for (int m = 0; m < 8; ++m) { r.kk[it][m] = 0; r.vv[it][m] = 0; }
@@ -389,7 +396,8 @@ struct DQV2 {
   }
 
   static __device__ void run(const unsigned short* q, const unsigned short* k,
-                             const unsigned short* v, const unsigned short* dout,
+                             const unsigned short* v, long vp,
+                             const unsigned short* dout,
                              const float* lse, const float* delta,
                              unsigned short* dq, int B, int Tq, int Tkv,
                              int q_off, int Hq, int Hkv, float scale,
@@ -407,6 +415,7 @@ struct DQV2 {
 
     const long q_base = (((long)b * Tq) * Hq + h) * D;
     const long kv_base = (((long)b * Tkv) * Hkv + hkv) * D;
+    const long v_base = ((long)b * Tkv) * vp + (long)hkv * D;
 
     if constexpr (D != DS) {
       for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16; i += NT)
@@ -447,10 +456,10 @@ struct DQV2 {
     const int q_lo_wave = qblk0 + wid * QBLK + q_off;
 
     Stage st;
-    stage_load(st, k, v, kv_base, 0, Tkv, Hkv);
+    stage_load(st, k, v, kv_base, v_base, vp, 0, Tkv, Hkv);
     stage_write(st, sm);
     __syncthreads();
-    if (n_tiles > 1) stage_load(st, k, v, kv_base, KVBLK, Tkv, Hkv);
+    if (n_tiles > 1) stage_load(st, k, v, kv_base, v_base, vp, KVBLK, Tkv, Hkv);
 
     const int lam = ln31 & 15;
     const unsigned trb = lds_addr(sm->kt)
@@ -534,7 +543,8 @@ struct DQV2 {
       if (tile + 1 < n_tiles) {
         stage_write(st, sm);
         if (tile + 2 < n_tiles)
-          stage_load(st, k, v, kv_base, (tile + 2) * KVBLK, Tkv, Hkv);
+          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * KVBLK,
+                     Tkv, Hkv);
         __syncthreads();
       }
     }
@@ -657,7 +667,8 @@ struct DKDVV2 {
   }
 
   static __device__ void run(const unsigned short* q, const unsigned short* k,
-                             const unsigned short* v, const unsigned short* dout,
+                             const unsigned short* v, long vp,
+                             const unsigned short* dout,
                              const float* lse, const float* delta,
                              unsigned short* dk, unsigned short* dv,
                              int B, int Tq, int Tkv, int q_off, int Hq,
@@ -677,6 +688,7 @@ struct DKDVV2 {
     const float scale2k = scale * 1.44269504f;
 
     const long kv_base = (((long)b * Tkv) * Hkv + hkv) * D;
+    const long v_base = ((long)b * Tkv) * vp + (long)hkv * D;
 
     if constexpr (D != DS) {
       for (int i = threadIdx.x; i < (DS / 16 - D / 16) * QBLK * 16; i += NTV) {
@@ -688,7 +700,7 @@ struct DKDVV2 {
     bf16x8 kfrag[NDSTEP], vfrag[DO_DK ? NDSTEP : 1];
     {
       const unsigned short* kr = k + kv_base + (long)kvg * Hkv * D;
-      const unsigned short* vr = v + kv_base + (long)kvg * Hkv * D;
+      const unsigned short* vr = v + v_base + (long)kvg * vp;
 #pragma unroll
       for (int s = 0; s < NDSTEP; ++s) {
         if (kvg < Tkv) {
@@ -907,20 +919,21 @@ struct DKDVV2 {
 template <int D>
 __global__ __launch_bounds__(NT, 2) void fwd2_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
-    unsigned short* o, float* lse, int B, int Tq, int Tkv, int q_off, int Hq,
-    int Hkv, float scale) {
+    long vp, unsigned short* o, float* lse, int B, int Tq, int Tkv, int q_off,
+    int Hq, int Hkv, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  FwdV2<D>::run(q, k, v, o, lse, B, Tq, Tkv, q_off, Hq, Hkv, scale, smem_raw);
+  FwdV2<D>::run(q, k, v, vp, o, lse, B, Tq, Tkv, q_off, Hq, Hkv, scale,
+                smem_raw);
 }
 
 template <int D>
 __global__ __launch_bounds__(NT, 2) void dq2_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
-    const unsigned short* dout, const float* lse, const float* delta,
+    long vp, const unsigned short* dout, const float* lse, const float* delta,
     unsigned short* dq, int B, int Tq, int Tkv, int q_off, int Hq, int Hkv,
     float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  DQV2<D>::run(q, k, v, dout, lse, delta, dq, B, Tq, Tkv, q_off, Hq, Hkv,
+  DQV2<D>::run(q, k, v, vp, dout, lse, delta, dq, B, Tq, Tkv, q_off, Hq, Hkv,
                scale, smem_raw);
 }
 
@@ -928,12 +941,12 @@ template <int D, int MODE>
 __global__ __launch_bounds__((DKDVV2<D, MODE>::NTV), (MODE == 1 ? 3 : 2))
 void dkdv2_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
-    const unsigned short* dout, const float* lse, const float* delta,
+    long vp, const unsigned short* dout, const float* lse, const float* delta,
     unsigned short* dk, unsigned short* dv, int B, int Tq, int Tkv, int q_off,
     int Hq, int Hkv, float scale, long dvp, long dvc) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  DKDVV2<D, MODE>::run(q, k, v, dout, lse, delta, dk, dv, B, Tq, Tkv, q_off,
-                       Hq, Hkv, scale, dvp, dvc, smem_raw);
+  DKDVV2<D, MODE>::run(q, k, v, vp, dout, lse, delta, dk, dv, B, Tq, Tkv,
+                       q_off, Hq, Hkv, scale, dvp, dvc, smem_raw);
 }
 
 // delta preprocess: delta[b,h,t] = rowsum(dO * O) (local copy — device
@@ -974,9 +987,15 @@ std::vector<torch::Tensor> attn_fwd2(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(causal, "attn_fwd2: only causal attention is implemented");
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 4
               && q.is_contiguous(), "q must be contiguous bf16 [B,T,Hq,D]");
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
   const int B = q.size(0), Tq = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Tkv = k.size(1), Hkv = k.size(2);
+  // V may be a row-strided view into the joint QKV activation (the fused
+  // path reads it in place): strides must be (Tkv*vp, vp, D, 1).
+  TORCH_CHECK(k.is_contiguous());
+  TORCH_CHECK(v.stride(3) == 1 && v.stride(2) == D
+              && v.stride(0) == (long)Tkv * v.stride(1),
+              "v must be contiguous or a [B,T,Hkv,D] view with row stride");
+  const long vp = v.stride(1);
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
   TORCH_CHECK(D == 64 || D == 80 || D == 128,
               "attn_fwd2: head_dim must be 64, 80 or 128");
@@ -990,7 +1009,7 @@ std::vector<torch::Tensor> attn_fwd2(torch::Tensor q, torch::Tensor k,
     hipLaunchKernelGGL(kfn, grid, dim3(attn2::NT), smem, stream,
                        (const unsigned short*)q.data_ptr(),
                        (const unsigned short*)k.data_ptr(),
-                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)v.data_ptr(), vp,
                        (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
                        B, Tq, Tkv, (int)q_offset, Hq, Hkv, scale);
   };
@@ -1009,6 +1028,10 @@ static std::vector<torch::Tensor> attn_bwd2_impl(
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.is_contiguous());
   const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Tkv = k.size(1), Hkv = k.size(2);
+  TORCH_CHECK(v.stride(3) == 1 && v.stride(2) == D
+              && v.stride(0) == (long)Tkv * v.stride(1),
+              "v must be contiguous or a [B,T,Hkv,D] view with row stride");
+  const long vstride = v.stride(1);
   const int q_off = (int)q_offset;
   const float scale = 1.0f / sqrtf((float)D);
   TORCH_CHECK(D == 64 || D == 80 || D == 128,
@@ -1052,7 +1075,7 @@ static std::vector<torch::Tensor> attn_bwd2_impl(
     hipLaunchKernelGGL(dqk, dim3(n_qblk, Hq, B), dim3(attn2::NT), smem_dq,
                        stream, (const unsigned short*)q.data_ptr(),
                        (const unsigned short*)k.data_ptr(),
-                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)v.data_ptr(), vstride,
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (unsigned short*)dq.data_ptr(), B, T, Tkv, q_off, Hq,
@@ -1065,7 +1088,7 @@ static std::vector<torch::Tensor> attn_bwd2_impl(
     hipLaunchKernelGGL(kfn, dim3(n_kvblk, Hkv, B), dim3(nthreads), smem,
                        st, (const unsigned short*)q.data_ptr(),
                        (const unsigned short*)k.data_ptr(),
-                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)v.data_ptr(), vstride,
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (unsigned short*)dk.data_ptr(),
@@ -1087,10 +1110,27 @@ static std::vector<torch::Tensor> attn_bwd2_impl(
     launch_kv(attn2::dkdv2_kernel<128, 2>, DK::WG_KV, DK::NTV, DK::SMEM_BYTES,
               side2);
   } else if (D == 80) {
-    using F = attn2::DKDVV2<80, 0>;
+    // A/B: fused dkdv (1 kernel, 8 waves, 254 regs) vs split dv/dk
+    // (register-light 4-wave kernels, +25% MFMA) — MA_DKDV80_SPLIT=0/1
+    static const bool split80 = [] {
+      const char* e = getenv("MA_DKDV80_SPLIT");
+      return e ? atoi(e) != 0 : false;
+    }();
     launch_dq(attn2::dq2_kernel<80>, sizeof(typename attn2::DQV2<80>::Smem));
-    launch_kv(attn2::dkdv2_kernel<80, 0>, F::WG_KV, F::NTV, F::SMEM_BYTES,
-              side);
+    if (split80) {
+      using DV = attn2::DKDVV2<80, 1>;
+      using DK = attn2::DKDVV2<80, 2>;
+      hipStreamWaitEvent(side2, ev_fork, 0);
+      used_side2 = true;
+      launch_kv(attn2::dkdv2_kernel<80, 1>, DV::WG_KV, DV::NTV,
+                DV::SMEM_BYTES, side);
+      launch_kv(attn2::dkdv2_kernel<80, 2>, DK::WG_KV, DK::NTV,
+                DK::SMEM_BYTES, side2);
+    } else {
+      using F = attn2::DKDVV2<80, 0>;
+      launch_kv(attn2::dkdv2_kernel<80, 0>, F::WG_KV, F::NTV, F::SMEM_BYTES,
+                side);
+    }
   } else {
     using F = attn2::DKDVV2<64, 0>;
     launch_dq(attn2::dq2_kernel<64>, sizeof(typename attn2::DQV2<64>::Smem));
@@ -1110,7 +1150,8 @@ std::vector<torch::Tensor> attn_bwd2(torch::Tensor dout, torch::Tensor q,
                                      torch::Tensor k, torch::Tensor v,
                                      torch::Tensor o, torch::Tensor lse,
                                      bool causal, long q_offset) {
-  auto dv = torch::empty_like(v);
+  // dv is always a fresh contiguous tensor (v itself may be a strided view)
+  auto dv = torch::empty(v.sizes(), q.options());
   const long Hkv = v.size(2), D = v.size(3);
   auto r = attn_bwd2_impl(dout, q, k, v, o, lse, causal, q_offset, dv,
                           Hkv * D, 0);
