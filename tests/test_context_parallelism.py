@@ -290,3 +290,28 @@ def test_cp2_dp2_matches_single_process(variant):
     for r in range(4):
         assert results[r] == pytest.approx(ref_losses, rel=2e-4), \
             (results[r], ref_losses)
+
+
+def _ring_bf16_dtype_worker(rank, world):
+    """ranks that merge >=2 ring partials must still return
+    the INPUT dtype (the fp32 logaddexp weights used to silently promote
+    the merged output, breaking the following c_proj GEMM on bf16)."""
+    import torch.distributed as dist
+
+    from modalities_amd.parallel.cp import cp_attention_ring
+    torch.manual_seed(0)
+    B, Tl, H, D = 1, 16, 2, 16
+    q = torch.randn(B, Tl, H, D).bfloat16()
+    k = torch.randn(B, Tl, H, D).bfloat16()
+    v = torch.randn(B, Tl, H, D).bfloat16()
+    y = cp_attention_ring(q, k, v, dist.group.WORLD, rank, world)
+    c_proj = torch.nn.Linear(H * D, H * D, bias=False).bfloat16()
+    out = c_proj(y.reshape(B, Tl, H * D))  # would raise on dtype mismatch
+    return str(y.dtype), str(out.dtype)
+
+
+def test_ring_bf16_output_dtype():
+    results = run_distributed(_ring_bf16_dtype_worker, world_size=2,
+                              port=find_free_port())
+    for r in range(2):
+        assert results[r] == ("torch.bfloat16", "torch.bfloat16")
