@@ -65,11 +65,16 @@ struct FwdV2 {
   static constexpr int NDSTEP = G::NDSTEP, NDBLK = G::NDBLK;
   static constexpr int WG_Q = QBLK * NW;
 
+  // NSUB kv sub-tiles of KVBLK rows staged per barrier pair (amortizes
+  // the two per-stage __syncthreads + load issue over 2x the MFMA work)
+  static constexpr int NSUB = 2;
+  static constexpr int TKV = KVBLK * NSUB;
+  static constexpr int PANEL = (DS / 16) * KVBLK * 16;  // vt elems per sub
   struct Smem {
-    unsigned short k[KVBLK * DP];
-    unsigned short vt[(DS / 16) * KVBLK * 16];
+    unsigned short k[NSUB * KVBLK * DP];
+    unsigned short vt[NSUB * PANEL];
   };
-  static constexpr int NCHS = KVBLK * D / 8;       // 16B chunks per tensor
+  static constexpr int NCHS = TKV * D / 8;         // 16B chunks per tensor
   static constexpr int NCH = (NCHS + NT - 1) / NT; // chunks per thread
 
   struct Stage {
@@ -108,9 +113,12 @@ struct FwdV2 {
       const int i = tid + it * NT;
       if (NCHS % NT != 0 && i >= NCHS) continue;
       const int r_ = (i * 8) / D, c = (i * 8) % D;
-      *reinterpret_cast<shortx8*>(&sm->k[r_ * DP + c]) = r.kk[it];
+      const int sub = r_ / KVBLK, rin = r_ % KVBLK;
       *reinterpret_cast<shortx8*>(
-          &sm->vt[(c >> 4) * (KVBLK * 16) + r_ * 16 + (c & 15)]) = r.vv[it];
+          &sm->k[sub * (KVBLK * DP) + rin * DP + c]) = r.kk[it];
+      *reinterpret_cast<shortx8*>(
+          &sm->vt[sub * PANEL + (c >> 4) * (KVBLK * 16) + rin * 16
+                  + (c & 15)]) = r.vv[it];
     }
   }
 
@@ -142,8 +150,11 @@ struct FwdV2 {
     // zero the d >= D panels of the V image once (D=80: panel 5 feeds the
     // dblk-2 MFMAs and must read as 0; panels 6,7 cleared too, cheap)
     if constexpr (D != DS) {
-      for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16; i += NT)
-        sm->vt[(D / 16) * KVBLK * 16 + i] = 0;
+#pragma unroll
+      for (int sub = 0; sub < NSUB; ++sub)
+        for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16;
+             i += NT)
+          sm->vt[sub * PANEL + (D / 16) * KVBLK * 16 + i] = 0;
       // first stage_write below is followed by __syncthreads
     }
 
@@ -169,7 +180,7 @@ struct FwdV2 {
       for (int r = 0; r < 16; ++r) acc_o[dblk][r] = 0.f;
 
     const int q_hi_wg = min(qblk0 + WG_Q - 1, Tq - 1) + q_off;
-    const int n_tiles = (min(q_hi_wg, Tkv - 1)) / KVBLK + 1;
+    const int n_tiles = (min(q_hi_wg, Tkv - 1)) / TKV + 1;
     // this wave's own causal bound (skip fully-masked tiles' compute)
     const int q_hi_wave = min(qblk0 + wid * QBLK + QBLK - 1, Tq - 1) + q_off;
     const int q_lo_wave = qblk0 + wid * QBLK + q_off;
@@ -178,7 +189,7 @@ struct FwdV2 {
     stage_load(st, k, v, kv_base, v_base, vp, 0, Tkv, Hkv);
     stage_write(st, sm);
     __syncthreads();
-    if (n_tiles > 1) stage_load(st, k, v, kv_base, v_base, vp, KVBLK, Tkv, Hkv);
+    if (n_tiles > 1) stage_load(st, k, v, kv_base, v_base, vp, TKV, Tkv, Hkv);
 
     // tile-invariant transpose-read base (single LDS buffer)
     const int lam = ln31 & 15;
@@ -187,10 +198,13 @@ struct FwdV2 {
         + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 
     for (int tile = 0; tile < n_tiles; ++tile) {
-      const int kv0 = tile * KVBLK;
-      const bool wave_active = (kv0 <= q_hi_wave);
+      const int kv0t = tile * TKV;
+      const bool stage_active = (kv0t <= q_hi_wave);
 
-      if (wave_active) {
+      if (stage_active) attnc::static_for<NSUB>([&](auto sub_) {
+        constexpr int sub = decltype(sub_)::value;
+        const int kv0 = kv0t + KVBLK * sub;
+        if (kv0 > q_hi_wave) return;  // sub fully masked for this wave
         // ---- S^T = K Q^T ------------------------------------------------
         floatx16 s0, s1;
 #pragma unroll
@@ -199,8 +213,10 @@ struct FwdV2 {
         for (int s = 0; s < NDSTEP; ++s) {
           const int col = hi * 8 + 16 * s;
           const int r0 = ln31, r1 = ln31 + 32;
-          bf16x8 ka = *reinterpret_cast<const bf16x8*>(&sm->k[r0 * DP + col]);
-          bf16x8 kb = *reinterpret_cast<const bf16x8*>(&sm->k[r1 * DP + col]);
+          bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+              &sm->k[sub * (KVBLK * DP) + r0 * DP + col]);
+          bf16x8 kb = *reinterpret_cast<const bf16x8*>(
+              &sm->k[sub * (KVBLK * DP) + r1 * DP + col]);
           s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[s], s0, 0, 0, 0);
           s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb, qfrag[s], s1, 0, 0, 0);
         }
@@ -279,7 +295,8 @@ struct FwdV2 {
           uintx2 rv[2 * (KVBLK / 16)];
           attnc::static_for<KVBLK / 16>([&](auto s_) {
             constexpr int s = decltype(s_)::value;
-            constexpr unsigned a = dblk * 2 * (KVBLK * 16) * 2 + s * 512;
+            constexpr unsigned a = sub * (PANEL * 2)
+                + dblk * 2 * (KVBLK * 16) * 2 + s * 512;
             rv[2 * s] = attnc::tr_read_b64_off<a>(trb);
             rv[2 * s + 1] = attnc::tr_read_b64_off<a + 128>(trb);
           });
@@ -294,13 +311,13 @@ struct FwdV2 {
                 va, pfrag[s], acc_o[dblk], 0, 0, 0);
           }
         });
-      }  // wave_active
+      });  // sub / stage_active
 
       __syncthreads();  // all waves done reading tile `tile`
       if (tile + 1 < n_tiles) {
         stage_write(st, sm);  // tile+1 regs -> LDS
         if (tile + 2 < n_tiles)
-          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * KVBLK,
+          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * TKV,
                      Tkv, Hkv);
         __syncthreads();      // tile+1 ready
       }
@@ -347,12 +364,16 @@ struct DQV2 {
   static constexpr int NDSTEP = G::NDSTEP, NDBLK = G::NDBLK;
   static constexpr int WG_Q = QBLK * NW;
 
+  // NSUB kv sub-tiles of KVBLK rows staged per barrier pair (see DKDVV2)
+  static constexpr int NSUB = 2;
+  static constexpr int TKV = KVBLK * NSUB;
+  static constexpr int PANEL = (DS / 16) * KVBLK * 16;  // kt elems per sub
   struct Smem {
-    unsigned short k[KVBLK * DP];
-    unsigned short v[KVBLK * DP];
-    unsigned short kt[(DS / 16) * KVBLK * 16];
+    unsigned short k[NSUB * KVBLK * DP];
+    unsigned short v[NSUB * KVBLK * DP];
+    unsigned short kt[NSUB * PANEL];
   };
-  static constexpr int NCHS = KVBLK * D / 8;
+  static constexpr int NCHS = TKV * D / 8;
   static constexpr int NCH = (NCHS + NT - 1) / NT;
 
   struct Stage {
@@ -388,10 +409,14 @@ struct DQV2 {
       const int i = tid + it * NT;
       if (NCHS % NT != 0 && i >= NCHS) continue;
       const int r_ = (i * 8) / D, c = (i * 8) % D;
-      *reinterpret_cast<shortx8*>(&sm->k[r_ * DP + c]) = r.kk[it];
-      *reinterpret_cast<shortx8*>(&sm->v[r_ * DP + c]) = r.vv[it];
+      const int sub = r_ / KVBLK, rin = r_ % KVBLK;
       *reinterpret_cast<shortx8*>(
-          &sm->kt[(c >> 4) * (KVBLK * 16) + r_ * 16 + (c & 15)]) = r.kk[it];
+          &sm->k[sub * (KVBLK * DP) + rin * DP + c]) = r.kk[it];
+      *reinterpret_cast<shortx8*>(
+          &sm->v[sub * (KVBLK * DP) + rin * DP + c]) = r.vv[it];
+      *reinterpret_cast<shortx8*>(
+          &sm->kt[sub * PANEL + (c >> 4) * (KVBLK * 16) + rin * 16
+                  + (c & 15)]) = r.kk[it];
     }
   }
 
@@ -418,8 +443,11 @@ struct DQV2 {
     const long v_base = ((long)b * Tkv) * vp + (long)hkv * D;
 
     if constexpr (D != DS) {
-      for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16; i += NT)
-        sm->kt[(D / 16) * KVBLK * 16 + i] = 0;
+#pragma unroll
+      for (int sub = 0; sub < NSUB; ++sub)
+        for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16;
+             i += NT)
+          sm->kt[sub * PANEL + (D / 16) * KVBLK * 16 + i] = 0;
     }
 
     bf16x8 qfrag[NDSTEP], dofrag[NDSTEP];
@@ -451,7 +479,7 @@ struct DQV2 {
       for (int r = 0; r < 16; ++r) acc_dq[dblk][r] = 0.f;
 
     const int q_hi_wg = min(qblk0 + WG_Q - 1, Tq - 1) + q_off;
-    const int n_tiles = min(q_hi_wg, Tkv - 1) / KVBLK + 1;
+    const int n_tiles = min(q_hi_wg, Tkv - 1) / TKV + 1;
     const int q_hi_wave = min(qblk0 + wid * QBLK + QBLK - 1, Tq - 1) + q_off;
     const int q_lo_wave = qblk0 + wid * QBLK + q_off;
 
@@ -459,7 +487,7 @@ struct DQV2 {
     stage_load(st, k, v, kv_base, v_base, vp, 0, Tkv, Hkv);
     stage_write(st, sm);
     __syncthreads();
-    if (n_tiles > 1) stage_load(st, k, v, kv_base, v_base, vp, KVBLK, Tkv, Hkv);
+    if (n_tiles > 1) stage_load(st, k, v, kv_base, v_base, vp, TKV, Tkv, Hkv);
 
     const int lam = ln31 & 15;
     const unsigned trb = lds_addr(sm->kt)
@@ -467,10 +495,13 @@ struct DQV2 {
         + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 
     for (int tile = 0; tile < n_tiles; ++tile) {
-      const int kv0 = tile * KVBLK;
-      const bool wave_active = (kv0 <= q_hi_wave);
+      const int kv0t = tile * TKV;
+      const bool stage_active = (kv0t <= q_hi_wave);
 
-      if (wave_active) {
+      if (stage_active) attnc::static_for<NSUB>([&](auto sub_) {
+        constexpr int sub = decltype(sub_)::value;
+        const int kv0 = kv0t + KVBLK * sub;
+        if (kv0 > q_hi_wave) return;  // sub fully masked for this wave
         attnc::static_for<2>([&](auto half_) {
           constexpr int half = decltype(half_)::value;
           const int kv0h = kv0 + 32 * half;
@@ -482,8 +513,10 @@ struct DQV2 {
           for (int s = 0; s < NDSTEP; ++s) {
             const int col = hi * 8 + 16 * s;
             const int r_ = 32 * half + ln31;
-            bf16x8 ka = *reinterpret_cast<const bf16x8*>(&sm->k[r_ * DP + col]);
-            bf16x8 va = *reinterpret_cast<const bf16x8*>(&sm->v[r_ * DP + col]);
+            bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+                &sm->k[sub * (KVBLK * DP) + r_ * DP + col]);
+            bf16x8 va = *reinterpret_cast<const bf16x8*>(
+                &sm->v[sub * (KVBLK * DP) + r_ * DP + col]);
             s_h = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[s], s_h,
                                                           0, 0, 0);
             dp_h = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[s], dp_h,
@@ -520,8 +553,8 @@ struct DQV2 {
             uintx2 rk[4];
             attnc::static_for<2>([&](auto s2_) {
               constexpr int s2 = decltype(s2_)::value;
-              constexpr unsigned a = dblk * 2 * (KVBLK * 16) * 2
-                  + (2 * half + s2) * 512;
+              constexpr unsigned a = sub * (PANEL * 2)
+                  + dblk * 2 * (KVBLK * 16) * 2 + (2 * half + s2) * 512;
               rk[2 * s2] = attnc::tr_read_b64_off<a>(trb);
               rk[2 * s2 + 1] = attnc::tr_read_b64_off<a + 128>(trb);
             });
@@ -537,13 +570,13 @@ struct DQV2 {
             }
           });
         });  // half
-      }  // wave_active
+      });  // sub / stage_active
 
       __syncthreads();
       if (tile + 1 < n_tiles) {
         stage_write(st, sm);
         if (tile + 2 < n_tiles)
-          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * KVBLK,
+          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * TKV,
                      Tkv, Hkv);
         __syncthreads();
       }
@@ -597,18 +630,27 @@ struct DKDVV2 {
   static constexpr int NTV = NWV * 64;
   static constexpr int WG_KV = QBLK * NWV;
 
-  // manual LDS carve (only the images this mode reads)
+  // Two 32-row q sub-tiles are staged per barrier pair (NSUB): amortizes
+  // the per-stage __syncthreads pair + global-load issue over 2x the MFMA
+  // work (the per-tile fixed costs dominated at 32-row tiles). The fused
+  // D=80 kernel sits exactly at the 256-register line — doubling its
+  // staging registers would spill into the hot loop, so it stays at 1.
+  static constexpr int NSUB = (MODE == 0 && D_ == 80) ? 1 : 2;
+  static constexpr int TQ = QBLK * NSUB;               // q rows per stage
+  // manual LDS carve (only the images this mode reads); each image is
+  // NSUB consecutive per-sub blocks
   static constexpr int DP = G::DP;                     // padded row pitch
-  static constexpr int PANEL = (DS / 16) * QBLK * 16;  // elements
+  static constexpr int PANEL = (DS / 16) * QBLK * 16;  // elements per sub
   static constexpr int O_QROW = 0;
-  static constexpr int O_DOROW = O_QROW + QBLK * DP;
-  static constexpr int O_QT = O_DOROW + (DO_DK ? QBLK * DP : 0);
-  static constexpr int O_DOT = O_QT + (DO_DK ? PANEL : 0);
-  static constexpr int O_END = O_DOT + (DO_DV ? PANEL : 0);
+  static constexpr int O_DOROW = O_QROW + NSUB * QBLK * DP;
+  static constexpr int O_QT = O_DOROW + (DO_DK ? NSUB * QBLK * DP : 0);
+  static constexpr int O_DOT = O_QT + (DO_DK ? NSUB * PANEL : 0);
+  static constexpr int O_END = O_DOT + (DO_DV ? NSUB * PANEL : 0);
   static constexpr int O_STATS = (O_END * 2 + 15) / 16 * 16;  // bytes, 16-al
-  static constexpr size_t SMEM_BYTES = O_STATS + (DO_DK ? 64 : 32) * 4;
+  static constexpr size_t SMEM_BYTES = O_STATS
+      + (size_t)(DO_DK ? 2 : 1) * NSUB * 32 * 4;
 
-  static constexpr int NCHS = QBLK * D / 8;
+  static constexpr int NCHS = TQ * D / 8;
   static constexpr int NCH = (NCHS + NTV - 1) / NTV;
 
   struct Stage {
@@ -636,10 +678,10 @@ struct DKDVV2 {
         for (int m = 0; m < 8; ++m) { r.qq[it][m] = 0; r.dd[it][m] = 0; }
       }
     }
-    if (tid < 32) {
+    if (tid < TQ) {
       r.stat = (q0 + tid < Tq) ? lse_h[q0 + tid] * 1.44269504f : 0.f;
-    } else if (DO_DK && tid < 64) {
-      r.stat = (q0 + tid - 32 < Tq) ? delta_h[q0 + tid - 32] : 0.f;
+    } else if (DO_DK && tid < 2 * TQ) {
+      r.stat = (q0 + tid - TQ < Tq) ? delta_h[q0 + tid - TQ] : 0.f;
     }
   }
 
@@ -652,18 +694,22 @@ struct DKDVV2 {
       const int i = tid + it * NTV;
       if (NCHS % NTV != 0 && i >= NCHS) continue;
       const int row = (i * 8) / D, c = (i * 8) % D;
-      *reinterpret_cast<shortx8*>(&base[O_QROW + row * DP + c]) = r.qq[it];
-      const int pan = (c >> 4) * (QBLK * 16) + row * 16 + (c & 15);
+      const int sub = row / QBLK, rin = row % QBLK;
+      *reinterpret_cast<shortx8*>(
+          &base[O_QROW + sub * (QBLK * DP) + rin * DP + c]) = r.qq[it];
+      const int pan = sub * PANEL + (c >> 4) * (QBLK * 16) + rin * 16
+          + (c & 15);
       if constexpr (DO_DK) {
-        *reinterpret_cast<shortx8*>(&base[O_DOROW + row * DP + c]) = r.dd[it];
+        *reinterpret_cast<shortx8*>(
+            &base[O_DOROW + sub * (QBLK * DP) + rin * DP + c]) = r.dd[it];
         *reinterpret_cast<shortx8*>(&base[O_QT + pan]) = r.qq[it];
       }
       if constexpr (DO_DV) {
         *reinterpret_cast<shortx8*>(&base[O_DOT + pan]) = r.dd[it];
       }
     }
-    if (tid < 32) stats[tid] = r.stat;
-    else if (DO_DK && tid < 64) stats[tid] = r.stat;
+    if (tid < TQ) stats[tid] = r.stat;
+    else if (DO_DK && tid < 2 * TQ) stats[tid] = r.stat;
   }
 
   static __device__ void run(const unsigned short* q, const unsigned short* k,
@@ -676,7 +722,7 @@ struct DKDVV2 {
                              char* smem_raw) {
     unsigned short* lds = reinterpret_cast<unsigned short*>(smem_raw);
     float* lse_s = reinterpret_cast<float*>(smem_raw + O_STATS);
-    float* delta_s = lse_s + 32;
+    float* delta_s = lse_s + TQ;
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
     const int hi = lane >> 5, ln31 = lane & 31;
@@ -691,10 +737,14 @@ struct DKDVV2 {
     const long v_base = ((long)b * Tkv) * vp + (long)hkv * D;
 
     if constexpr (D != DS) {
-      for (int i = threadIdx.x; i < (DS / 16 - D / 16) * QBLK * 16; i += NTV) {
-        if constexpr (DO_DK) lds[O_QT + (D / 16) * QBLK * 16 + i] = 0;
-        if constexpr (DO_DV) lds[O_DOT + (D / 16) * QBLK * 16 + i] = 0;
-      }
+#pragma unroll
+      for (int sub = 0; sub < NSUB; ++sub)
+        for (int i = threadIdx.x; i < (DS / 16 - D / 16) * QBLK * 16;
+             i += NTV) {
+          const int off = sub * PANEL + (D / 16) * QBLK * 16 + i;
+          if constexpr (DO_DK) lds[O_QT + off] = 0;
+          if constexpr (DO_DV) lds[O_DOT + off] = 0;
+        }
     }
 
     bf16x8 kfrag[NDSTEP], vfrag[DO_DK ? NDSTEP : 1];
@@ -726,13 +776,13 @@ struct DKDVV2 {
         if constexpr (DO_DV) acc_dv[dblk][r] = 0.f;
       }
 
-    // flattened (GQA q-head) x (q tile) iteration space with incremental
-    // cursors (div/mod-free; see v1)
-    const int first_qtile = max(0, kvblk0 - q_off) / QBLK;
-    const int n_qtiles = (Tq + QBLK - 1) / QBLK;
+    // flattened (GQA q-head) x (q stage of TQ rows) iteration space with
+    // incremental cursors (div/mod-free; see v1)
+    const int first_qtile = max(0, kvblk0 - q_off) / TQ;
+    const int n_qtiles = (Tq + TQ - 1) / TQ;
     const int tiles_per_head = n_qtiles - first_qtile;
     const int n_iter = rep * tiles_per_head;
-    const int q0_first = first_qtile * QBLK;
+    const int q0_first = first_qtile * TQ;
 
     struct Cursor {
       int q0;
@@ -747,8 +797,8 @@ struct DKDVV2 {
       c.delta_h = delta + (((long)b * Hq) + hq) * Tq;
     };
     auto cursor_next = [&](Cursor& c) {
-      c.q0 += QBLK;
-      if (c.q0 >= n_qtiles * QBLK) {
+      c.q0 += TQ;
+      if (c.q0 >= n_qtiles * TQ) {
         c.q0 = q0_first;
         c.q_base += (long)D;
         c.lse_h += Tq;
@@ -775,12 +825,17 @@ struct DKDVV2 {
     const unsigned dot_trb = lds_addr(lds + O_DOT) + tr_lane;
 
     for (int t = 0; t < n_iter; ++t) {
-      const int q0 = cc.q0;
+      const int q0t = cc.q0;
       cursor_next(cc);
-      // a tile strictly below this wave's kv rows is fully masked for it
-      const bool wave_active = (q0 + QBLK - 1 + q_off >= wv_kv0);
+      // a stage strictly below this wave's kv rows is fully masked for it
+      const bool stage_active = (q0t + TQ - 1 + q_off >= wv_kv0);
 
-      if (wave_active) {
+      if (stage_active) attnc::static_for<NSUB>([&](auto sub_) {
+        constexpr int sub = decltype(sub_)::value;
+        const int q0 = q0t + sub * QBLK;
+        if (q0 + QBLK - 1 + q_off < wv_kv0) return;  // sub fully masked
+        const float* lse_sub = lse_s + sub * 32;
+        const float* delta_sub = delta_s + sub * 32;
         // S[q][kv], dP[q][kv] (col = kv = ln31, rows q = crow)
         floatx16 s_acc, dp_acc;
 #pragma unroll
@@ -789,12 +844,12 @@ struct DKDVV2 {
         for (int s = 0; s < NDSTEP; ++s) {
           const int col = hi * 8 + 16 * s;
           bf16x8 qa = *reinterpret_cast<const bf16x8*>(
-              &lds[O_QROW + ln31 * DP + col]);
+              &lds[O_QROW + sub * (QBLK * DP) + ln31 * DP + col]);
           s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[s], s_acc,
                                                           0, 0, 0);
           if constexpr (DO_DK) {
             bf16x8 da = *reinterpret_cast<const bf16x8*>(
-                &lds[O_DOROW + ln31 * DP + col]);
+                &lds[O_DOROW + sub * (QBLK * DP) + ln31 * DP + col]);
             dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vfrag[s],
                                                              dp_acc, 0, 0, 0);
           }
@@ -804,15 +859,15 @@ struct DKDVV2 {
         // s_acc becomes P; dp_acc becomes dS' when dK is computed)
         if (q0 + q_off >= wv_kv0 + QBLK && q0 + QBLK <= Tq
             && wv_kv0 + QBLK <= Tkv) {
-          // interior: every q row of this tile covers every kv row here
+          // interior: every q row of this sub covers every kv row here
 #pragma unroll
           for (int r = 0; r < 16; ++r) {
-            const float lq = lse_s[crow(r, hi)];
+            const float lq = lse_sub[crow(r, hi)];
             const float pv = __builtin_amdgcn_exp2f(
                 fmaf(s_acc[r], scale2k, -lq));
             s_acc[r] = pv;
             if constexpr (DO_DK) {
-              const float dl = delta_s[crow(r, hi)];
+              const float dl = delta_sub[crow(r, hi)];
               dp_acc[r] = scale * pv * (dp_acc[r] - dl);
             }
           }
@@ -821,12 +876,12 @@ struct DKDVV2 {
           for (int r = 0; r < 16; ++r) {
             const int qg = q0 + crow(r, hi);  // local
             const bool ok = (qg < Tq) && (qg + q_off >= kvg) && (kvg < Tkv);
-            const float lq = lse_s[crow(r, hi)];
+            const float lq = lse_sub[crow(r, hi)];
             const float pv = ok
                 ? __builtin_amdgcn_exp2f(fmaf(s_acc[r], scale2k, -lq)) : 0.f;
             s_acc[r] = pv;
             if constexpr (DO_DK) {
-              const float dl = delta_s[crow(r, hi)];
+              const float dl = delta_sub[crow(r, hi)];
               dp_acc[r] = scale * pv * (dp_acc[r] - dl);
             }
           }
@@ -836,13 +891,14 @@ struct DKDVV2 {
         if constexpr (DO_DK) attnc::c16_to_frags(dp_acc, dsfrag);
 
         // dV += P^T dO ; dK += dS'^T Q  (B-operands via tr reads; base
-        // VGPR + offset immediates)
+        // VGPR + offset immediates, sub block folded into the immediate)
         attnc::static_for<NDBLK>([&](auto dblk_) {
           constexpr int dblk = decltype(dblk_)::value;
           uintx2 rd[4], rq[4];
           attnc::static_for<2>([&](auto s_) {
             constexpr int s = decltype(s_)::value;
-            constexpr unsigned a = dblk * 2 * (QBLK * 16) * 2 + s * 512;
+            constexpr unsigned a = sub * (PANEL * 2)
+                + dblk * 2 * (QBLK * 16) * 2 + s * 512;
             if constexpr (DO_DV) {
               rd[2 * s] = attnc::tr_read_b64_off<a>(dot_trb);
               rd[2 * s + 1] = attnc::tr_read_b64_off<a + 128>(dot_trb);
@@ -872,7 +928,7 @@ struct DKDVV2 {
             }
           }
         });
-      }  // wave_active
+      });  // sub / stage_active
 
       __syncthreads();
       if (t + 1 < n_iter) {
@@ -938,7 +994,8 @@ __global__ __launch_bounds__(NT, 2) void dq2_kernel(
 }
 
 template <int D, int MODE>
-__global__ __launch_bounds__((DKDVV2<D, MODE>::NTV), (MODE == 1 ? 3 : 2))
+__global__ __launch_bounds__((DKDVV2<D, MODE>::NTV),
+                             ((MODE == 1 && D == 80) ? 3 : 2))
 void dkdv2_kernel(
     const unsigned short* q, const unsigned short* k, const unsigned short* v,
     long vp, const unsigned short* dout, const float* lse, const float* delta,

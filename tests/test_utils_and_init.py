@@ -156,3 +156,51 @@ def test_hf_conversion_logit_equality():
         ffn_hidden=256, sequence_length=64, activation_type="swiglu"))
     hf = convert_model_checkpoint(model)
     check_converted_model(hf, model, num_testruns=2, vocab_size=128, seq_len=32)
+
+
+# ---------------------------------------------------------------------------
+# Init distribution statistics (reference tests/test_initialization_fsdpx.py:
+# after composed init, each parameter group's empirical std must match its
+# configured target).
+
+def test_composed_init_distribution_statistics():
+    import math
+
+    import torch
+
+    from modalities_amd.models.gpt2 import GPT2LLM, GPT2LLMConfig
+    from modalities_amd.nn.model_initialization import (
+        get_composed_model_initializer)
+
+    torch.manual_seed(0)
+    h, L = 256, 4
+    cfg = GPT2LLMConfig(vocab_size=2048, n_layer=L, n_head_q=4, n_head_kv=4,
+                        n_embd=h, ffn_hidden=4 * h, sequence_length=32,
+                        dropout=0.0)
+    model = GPT2LLM(cfg)
+    init = get_composed_model_initializer(
+        model_type="gpt2", weight_init_type="scaled", mean=0.0, std="auto",
+        num_layers=L, hidden_dim=h)
+    init.initialize_in_place(model)
+
+    auto_std = math.sqrt(2 / (5 * h))
+    proj_std = auto_std / math.sqrt(2 * L)
+
+    def emp_std(t):
+        return t.float().std().item()
+
+    # embeddings at the plain/auto std
+    assert abs(emp_std(model.wte.weight) - auto_std) / auto_std < 0.05
+    # in-projections (qkv / mlp up) at the plain std
+    qkv_w = model.blocks[0].attn.qkv_attn.weight if cfg.fused_qkv \
+        else model.blocks[0].attn.q_attn.weight
+    assert abs(emp_std(qkv_w) - auto_std) / auto_std < 0.05
+    up = model.blocks[0].mlp.W_weight
+    assert abs(emp_std(up) - auto_std) / auto_std < 0.05
+    # OUT-projections scaled down by sqrt(2L)
+    assert abs(emp_std(model.blocks[0].attn.c_proj.weight) - proj_std) \
+        / proj_std < 0.05
+    assert abs(emp_std(model.blocks[0].mlp.W_2.weight) - proj_std) \
+        / proj_std < 0.05
+    # norms untouched (ones)
+    assert torch.all(model.blocks[0].attention_norm.weight == 1.0)
