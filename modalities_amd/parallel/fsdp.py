@@ -488,7 +488,12 @@ class XGMIShardedModel(nn.Module):
 
     @torch.no_grad()
     def clip_grad_norm_(self, max_norm: Optional[float],
-                        norm_type: float = 2.0) -> torch.Tensor:
+                        norm_type: float = 2.0,
+                        pp_group=None) -> torch.Tensor:
+        """Global-norm clip over the DP-sharded grads; with pipeline
+        parallelism pass the PP group so the norm combines over stages
+        (reference fsdp_gradient_clipper.py:166-169) — each rank then clips
+        by the MODEL-global norm, and the published value is global."""
         from modalities_amd.ops.adamw import multi_tensor_l2norm, multi_tensor_scale_
         fresh = [u.grad_shard for u in self.units if u.grad_fresh]
         local = multi_tensor_l2norm(fresh) ** 2
@@ -496,6 +501,8 @@ class XGMIShardedModel(nn.Module):
             local = local.to(self.device)
         if self.world > 1:
             dist.all_reduce(local, group=self.group)
+        if pp_group is not None:
+            dist.all_reduce(local, group=pp_group)
         total = local.sqrt()
         if max_norm is not None and max_norm > 0 and fresh:
             clip = (max_norm / (total + 1e-6)).clamp(max=1.0)

@@ -471,3 +471,41 @@ def test_pp2_fqn_stages_match_single_process():
     results = run_distributed(_pp_fqn_worker, world_size=2,
                               port=find_free_port())
     assert results[1] == pytest.approx(ref_losses, rel=1e-5)
+
+
+def _pp_gradnorm_worker(rank, world):
+    import torch.distributed as dist
+
+    from modalities_amd.loss_functions import CLMCrossEntropyLoss
+    from modalities_amd.parallel.pp import get_pipeline_schedule
+    from modalities_amd.training.gradient_clipping import GradientClipper
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    stages = split_model_into_stages(model, world)
+    sched = get_pipeline_schedule(
+        "gpipe", stage=stages[rank], stage_idx=rank, num_stages=world,
+        n_microbatches=2, group=dist.group.WORLD)
+    x, y = make_batch()
+    sched.step(x, y, CLMCrossEntropyLoss("target_ids", "logits"))
+    clipper = GradientClipper(max_norm=1.0, pp_group=dist.group.WORLD)
+    return clipper(stages[rank]).item()
+
+
+def test_pp_grad_norm_is_model_global():
+    """The published grad norm under PP must be the MODEL-global norm
+    (identical on every stage; reference fsdp_gradient_clipper.py:166-169),
+    not the stage-local norm."""
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    x, y = make_batch()
+    for mb_x, mb_y in zip(x.chunk(2), y.chunk(2)):
+        out = model({"input_ids": mb_x})["logits"]
+        loss = torch.nn.functional.cross_entropy(
+            out.reshape(-1, VOCAB).float(), mb_y.reshape(-1))
+        (loss / 2).backward()
+    ref = torch.sqrt(sum(p.grad.float().pow(2).sum()
+                         for p in model.parameters())).item()
+    results = run_distributed(_pp_gradnorm_worker, world_size=2,
+                              port=find_free_port())
+    assert results[0] == pytest.approx(results[1], rel=1e-6)
+    assert results[0] == pytest.approx(ref, rel=1e-4)
