@@ -509,3 +509,41 @@ def test_pp_grad_norm_is_model_global():
                               port=find_free_port())
     assert results[0] == pytest.approx(results[1], rel=1e-6)
     assert results[0] == pytest.approx(ref, rel=1e-4)
+
+
+def _evaluator_pp_worker(rank, world):
+    import torch.distributed as dist
+
+    from modalities_amd.batch import DatasetBatch
+    from modalities_amd.loss_functions import CLMCrossEntropyLoss
+    from modalities_amd.parallel.pp import get_pipeline_schedule
+    from modalities_amd.training.evaluator import Evaluator
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    stages = split_model_into_stages(model, world)
+    sched = get_pipeline_schedule(
+        "1f1b", stage=stages[rank], stage_idx=rank, num_stages=world,
+        n_microbatches=N_MB, group=dist.group.WORLD)
+    ev = Evaluator(progress_publisher=_NullPub(), evaluation_result_publisher=_NullPub(),
+                   pp_schedule=sched)
+    x, y = make_batch()
+    batch = DatasetBatch(samples={"input_ids": x}, targets={"target_ids": y})
+    loss = ev.evaluate_batch(batch, stages[rank],
+                             CLMCrossEntropyLoss("target_ids", "logits"))
+    return loss.item()
+
+
+class _NullPub:
+    def publish_message(self, *a, **k):
+        pass
+
+
+def test_evaluator_dispatches_pp_schedule():
+    """Evaluator.evaluate_batch runs the forward-only PP schedule and every
+    rank sees the broadcast mean loss (reference evaluator.py:88-180)."""
+    ref_losses, _ = reference_loss_and_grads()
+    ref_mean = sum(ref_losses) / N_MB
+    results = run_distributed(_evaluator_pp_worker, world_size=2,
+                              port=find_free_port())
+    for r in range(2):
+        assert results[r] == pytest.approx(ref_mean, rel=1e-5)
