@@ -635,7 +635,12 @@ struct DKDVV2 {
   // work (the per-tile fixed costs dominated at 32-row tiles). The fused
   // D=80 kernel sits exactly at the 256-register line — doubling its
   // staging registers would spill into the hot loop, so it stays at 1.
+  // The fused D=80 kernel sits exactly at the 256-register line — the
+  // NSUB staging registers would spill into the hot loop (184 B scratch
+  // measured even with halved tr batches), so it stays at 1 sub-tile;
+  // the D=80 split kernels (MA_DKDV80_SPLIT=1) run NSUB=2.
   static constexpr int NSUB = (MODE == 0 && D_ == 80) ? 1 : 2;
+  static constexpr bool TRHALF = false;
   static constexpr int TQ = QBLK * NSUB;               // q rows per stage
   // manual LDS carve (only the images this mode reads); each image is
   // NSUB consecutive per-sub blocks
@@ -894,38 +899,52 @@ struct DKDVV2 {
         // VGPR + offset immediates, sub block folded into the immediate)
         attnc::static_for<NDBLK>([&](auto dblk_) {
           constexpr int dblk = decltype(dblk_)::value;
-          uintx2 rd[4], rq[4];
-          attnc::static_for<2>([&](auto s_) {
+          auto step = [&](auto s_, uintx2* rd, uintx2* rq) {
             constexpr int s = decltype(s_)::value;
             constexpr unsigned a = sub * (PANEL * 2)
                 + dblk * 2 * (QBLK * 16) * 2 + s * 512;
             if constexpr (DO_DV) {
-              rd[2 * s] = attnc::tr_read_b64_off<a>(dot_trb);
-              rd[2 * s + 1] = attnc::tr_read_b64_off<a + 128>(dot_trb);
+              rd[0] = attnc::tr_read_b64_off<a>(dot_trb);
+              rd[1] = attnc::tr_read_b64_off<a + 128>(dot_trb);
             }
             if constexpr (DO_DK) {
-              rq[2 * s] = attnc::tr_read_b64_off<a>(qt_trb);
-              rq[2 * s + 1] = attnc::tr_read_b64_off<a + 128>(qt_trb);
+              rq[0] = attnc::tr_read_b64_off<a>(qt_trb);
+              rq[1] = attnc::tr_read_b64_off<a + 128>(qt_trb);
             }
-          });
-          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-          __builtin_amdgcn_sched_barrier(0);
-#pragma unroll
-          for (int s = 0; s < 2; ++s) {
+          };
+          auto consume = [&](int s, uintx2* rd, uintx2* rq) {
             if constexpr (DO_DV) {
-              unsigned wd[4] = {rd[2 * s][0], rd[2 * s][1],
-                                rd[2 * s + 1][0], rd[2 * s + 1][1]};
+              unsigned wd[4] = {rd[0][0], rd[0][1], rd[1][0], rd[1][1]};
               bf16x8 dob = *reinterpret_cast<bf16x8*>(wd);
               acc_dv[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                   pfrag[s], dob, acc_dv[dblk], 0, 0, 0);
             }
             if constexpr (DO_DK) {
-              unsigned wq[4] = {rq[2 * s][0], rq[2 * s][1],
-                                rq[2 * s + 1][0], rq[2 * s + 1][1]};
+              unsigned wq[4] = {rq[0][0], rq[0][1], rq[1][0], rq[1][1]};
               bf16x8 qb = *reinterpret_cast<bf16x8*>(wq);
               acc_dk[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                   dsfrag[s], qb, acc_dk[dblk], 0, 0, 0);
             }
+          };
+          if constexpr (TRHALF) {
+            // one k-step per drain (half the in-flight read registers)
+            uintx2 rd[2], rq[2];
+            attnc::static_for<2>([&](auto s_) {
+              step(s_, rd, rq);
+              asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+              __builtin_amdgcn_sched_barrier(0);
+              consume(decltype(s_)::value, rd, rq);
+            });
+          } else {
+            uintx2 rd[4], rq[4];
+            attnc::static_for<2>([&](auto s_) {
+              constexpr int s = decltype(s_)::value;
+              step(s_, rd + 2 * s, rq + 2 * s);
+            });
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+            for (int s = 0; s < 2; ++s) consume(s, rd + 2 * s, rq + 2 * s);
           }
         });
       });  // sub / stage_active
