@@ -635,10 +635,12 @@ struct DKDVV2 {
   // work (the per-tile fixed costs dominated at 32-row tiles). The fused
   // D=80 kernel sits exactly at the 256-register line — doubling its
   // staging registers would spill into the hot loop, so it stays at 1.
-  // The fused D=80 kernel sits exactly at the 256-register line — the
-  // NSUB staging registers would spill into the hot loop (184 B scratch
-  // measured even with halved tr batches), so it stays at 1 sub-tile;
-  // the D=80 split kernels (MA_DKDV80_SPLIT=1) run NSUB=2.
+  // The fused D=80 kernel sits exactly at the 256-register line: the
+  // NSUB staging registers spill into the hot loop (184 B/lane measured,
+  // even with halved tr batches + compressed cursors), so it stays at
+  // one sub-tile. The A/B against the register-clean split dV/dK pair at
+  // NSUB=2 still favors the fused kernel (1.95 vs 2.01 ms/layer,
+  // profiles/r2_bench_2p7b_1gpu.md).
   static constexpr int NSUB = (MODE == 0 && D_ == 80) ? 1 : 2;
   static constexpr bool TRHALF = false;
   static constexpr int TQ = QBLK * NSUB;               // q rows per stage
@@ -789,38 +791,41 @@ struct DKDVV2 {
     const int n_iter = rep * tiles_per_head;
     const int q0_first = first_qtile * TQ;
 
+    // Cursors are two small ints (q0, GQA head index); the pointer bases
+    // are recomputed at each stage_load — a handful of VALU once per
+    // stage instead of 6 live registers per cursor (this kernel sits at
+    // the 256-register line).
     struct Cursor {
-      int q0;
-      long q_base;
-      const float *lse_h, *delta_h;
+      int q0, hq;
     };
     auto cursor_init = [&](Cursor& c) {
-      const int hq = hkv * rep;
       c.q0 = q0_first;
-      c.q_base = (((long)b * Tq) * Hq + hq) * D;
-      c.lse_h = lse + (((long)b * Hq) + hq) * Tq;
-      c.delta_h = delta + (((long)b * Hq) + hq) * Tq;
+      c.hq = hkv * rep;
     };
     auto cursor_next = [&](Cursor& c) {
       c.q0 += TQ;
       if (c.q0 >= n_qtiles * TQ) {
         c.q0 = q0_first;
-        c.q_base += (long)D;
-        c.lse_h += Tq;
-        c.delta_h += Tq;
+        c.hq += 1;
       }
+    };
+    auto do_stage_load = [&](Stage& st_, const Cursor& c) {
+      const long qb = (((long)b * Tq) * Hq + c.hq) * D;
+      const float* lh = lse + (((long)b * Hq) + c.hq) * Tq;
+      const float* dh = delta + (((long)b * Hq) + c.hq) * Tq;
+      stage_load(st_, q, dout, qb, lh, dh, c.q0, Tq, Hq);
     };
 
     Stage st;
     Cursor cc, cp;
     cursor_init(cc);
     cursor_init(cp);
-    stage_load(st, q, dout, cc.q_base, cc.lse_h, cc.delta_h, cc.q0, Tq, Hq);
+    do_stage_load(st, cc);
     stage_write(st, smem_raw);
     __syncthreads();
     if (n_iter > 1) {
       cursor_next(cp);
-      stage_load(st, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0, Tq, Hq);
+      do_stage_load(st, cp);
     }
 
     const int lam = ln31 & 15;
@@ -954,8 +959,7 @@ struct DKDVV2 {
         stage_write(st, smem_raw);
         if (t + 2 < n_iter) {
           cursor_next(cp);
-          stage_load(st, q, dout, cp.q_base, cp.lse_h, cp.delta_h, cp.q0,
-                     Tq, Hq);
+          do_stage_load(st, cp);
         }
         __syncthreads();
       }
