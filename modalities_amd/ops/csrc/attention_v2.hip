@@ -74,6 +74,9 @@ struct FwdV2 {
     unsigned short k[NSUB * KVBLK * DP];
     unsigned short vt[NSUB * PANEL];
   };
+  // double-buffered: single-barrier T14 loop (see DKDVV2)
+  static constexpr int BUF_BYTES = (int)((sizeof(Smem) + 15) / 16 * 16);
+  static constexpr size_t SMEM_BYTES = (size_t)BUF_BYTES * 2;
   static constexpr int NCHS = TKV * D / 8;         // 16B chunks per tensor
   static constexpr int NCH = (NCHS + NT - 1) / NT; // chunks per thread
 
@@ -129,7 +132,6 @@ struct FwdV2 {
                              float* __restrict__ lse, int B, int Tq, int Tkv,
                              int q_off, int Hq, int Hkv, float scale,
                              char* smem_raw) {
-    Smem* sm = reinterpret_cast<Smem*>(smem_raw);
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
     const int hi = lane >> 5, ln31 = lane & 31;
@@ -151,10 +153,14 @@ struct FwdV2 {
     // dblk-2 MFMAs and must read as 0; panels 6,7 cleared too, cheap)
     if constexpr (D != DS) {
 #pragma unroll
-      for (int sub = 0; sub < NSUB; ++sub)
-        for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16;
-             i += NT)
-          sm->vt[sub * PANEL + (D / 16) * KVBLK * 16 + i] = 0;
+      for (int bufi = 0; bufi < 2; ++bufi) {
+        Smem* sz = reinterpret_cast<Smem*>(smem_raw + bufi * BUF_BYTES);
+#pragma unroll
+        for (int sub = 0; sub < NSUB; ++sub)
+          for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16;
+               i += NT)
+            sz->vt[sub * PANEL + (D / 16) * KVBLK * 16 + i] = 0;
+      }
       // first stage_write below is followed by __syncthreads
     }
 
@@ -187,19 +193,31 @@ struct FwdV2 {
 
     Stage st;
     stage_load(st, k, v, kv_base, v_base, vp, 0, Tkv, Hkv);
-    stage_write(st, sm);
+    stage_write(st, reinterpret_cast<Smem*>(smem_raw));
     __syncthreads();
     if (n_tiles > 1) stage_load(st, k, v, kv_base, v_base, vp, TKV, Tkv, Hkv);
 
-    // tile-invariant transpose-read base (single LDS buffer)
     const int lam = ln31 & 15;
-    const unsigned trb = lds_addr(sm->vt)
+    const unsigned trb0 = lds_addr(
+        reinterpret_cast<Smem*>(smem_raw)->vt)
         + (unsigned)(ln31 >> 4) * (KVBLK * 16 * 2) + (unsigned)hi * 256
         + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 
     for (int tile = 0; tile < n_tiles; ++tile) {
       const int kv0t = tile * TKV;
       const bool stage_active = (kv0t <= q_hi_wave);
+      const unsigned boff = (unsigned)(tile & 1) * BUF_BYTES;
+      Smem* const sm = reinterpret_cast<Smem*>(smem_raw + boff);
+      const unsigned trb = trb0 + boff;
+      // single-barrier T14: stage tile+1 into the other buffer and issue
+      // tile+2's loads BEFORE this tile's compute
+      if (tile + 1 < n_tiles) {
+        stage_write(st, reinterpret_cast<Smem*>(
+            smem_raw + (unsigned)(~tile & 1) * BUF_BYTES));
+        if (tile + 2 < n_tiles)
+          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * TKV,
+                     Tkv, Hkv);
+      }
 
       if (stage_active) attnc::static_for<NSUB>([&](auto sub_) {
         constexpr int sub = decltype(sub_)::value;
@@ -313,14 +331,7 @@ struct FwdV2 {
         });
       });  // sub / stage_active
 
-      __syncthreads();  // all waves done reading tile `tile`
-      if (tile + 1 < n_tiles) {
-        stage_write(st, sm);  // tile+1 regs -> LDS
-        if (tile + 2 < n_tiles)
-          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * TKV,
-                     Tkv, Hkv);
-        __syncthreads();      // tile+1 ready
-      }
+      __syncthreads();  // tile reads done AND tile+1 writes landed
     }
 
     // ---- epilogue -------------------------------------------------------
@@ -654,8 +665,14 @@ struct DKDVV2 {
   static constexpr int O_DOT = O_QT + (DO_DK ? NSUB * PANEL : 0);
   static constexpr int O_END = O_DOT + (DO_DV ? NSUB * PANEL : 0);
   static constexpr int O_STATS = (O_END * 2 + 15) / 16 * 16;  // bytes, 16-al
-  static constexpr size_t SMEM_BYTES = O_STATS
-      + (size_t)(DO_DK ? 2 : 1) * NSUB * 32 * 4;
+  // Double-buffered stages run the canonical single-barrier T14 loop
+  // (write tile t+1 into the OTHER buffer + issue t+2 loads BEFORE the
+  // compute of t; one __syncthreads per stage). The dK-only kernel runs
+  // 2 workgroups/CU and cannot fit two buffers in its LDS share.
+  static constexpr bool DBUF = (MODE != 2);
+  static constexpr int BUF_BYTES =
+      (O_STATS + (DO_DK ? 2 : 1) * NSUB * 32 * 4 + 15) / 16 * 16;
+  static constexpr size_t SMEM_BYTES = (size_t)BUF_BYTES * (DBUF ? 2 : 1);
 
   static constexpr int NCHS = TQ * D / 8;
   static constexpr int NCH = (NCHS + NTV - 1) / NTV;
@@ -745,13 +762,18 @@ struct DKDVV2 {
 
     if constexpr (D != DS) {
 #pragma unroll
-      for (int sub = 0; sub < NSUB; ++sub)
-        for (int i = threadIdx.x; i < (DS / 16 - D / 16) * QBLK * 16;
-             i += NTV) {
-          const int off = sub * PANEL + (D / 16) * QBLK * 16 + i;
-          if constexpr (DO_DK) lds[O_QT + off] = 0;
-          if constexpr (DO_DV) lds[O_DOT + off] = 0;
-        }
+      for (int bufi = 0; bufi < (DBUF ? 2 : 1); ++bufi) {
+        unsigned short* lz = reinterpret_cast<unsigned short*>(
+            smem_raw + bufi * BUF_BYTES);
+#pragma unroll
+        for (int sub = 0; sub < NSUB; ++sub)
+          for (int i = threadIdx.x; i < (DS / 16 - D / 16) * QBLK * 16;
+               i += NTV) {
+            const int off = sub * PANEL + (D / 16) * QBLK * 16 + i;
+            if constexpr (DO_DK) lz[O_QT + off] = 0;
+            if constexpr (DO_DV) lz[O_DOT + off] = 0;
+          }
+      }
     }
 
     bf16x8 kfrag[NDSTEP], vfrag[DO_DK ? NDSTEP : 1];
@@ -831,12 +853,31 @@ struct DKDVV2 {
     const int lam = ln31 & 15;
     const unsigned tr_lane = (unsigned)(ln31 >> 4) * (QBLK * 16 * 2)
         + (unsigned)hi * 256 + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
-    const unsigned qt_trb = lds_addr(lds + O_QT) + tr_lane;
-    const unsigned dot_trb = lds_addr(lds + O_DOT) + tr_lane;
+    const unsigned qt_trb0 = lds_addr(lds + O_QT) + tr_lane;
+    const unsigned dot_trb0 = lds_addr(lds + O_DOT) + tr_lane;
 
     for (int t = 0; t < n_iter; ++t) {
       const int q0t = cc.q0;
       cursor_next(cc);
+      const unsigned boff = DBUF ? (unsigned)(t & 1) * BUF_BYTES : 0u;
+      unsigned short* const ldsb =
+          reinterpret_cast<unsigned short*>(smem_raw + boff);
+      const float* const lse_b =
+          reinterpret_cast<const float*>(smem_raw + boff + O_STATS);
+      const float* const delta_b = lse_b + TQ;
+      const unsigned qt_trb = qt_trb0 + boff;
+      const unsigned dot_trb = dot_trb0 + boff;
+      if constexpr (DBUF) {
+        // single-barrier T14: stage t+1 into the other buffer and issue
+        // t+2's loads BEFORE this tile's compute
+        if (t + 1 < n_iter) {
+          stage_write(st, smem_raw + (unsigned)(~t & 1) * BUF_BYTES);
+          if (t + 2 < n_iter) {
+            cursor_next(cp);
+            do_stage_load(st, cp);
+          }
+        }
+      }
       // a stage strictly below this wave's kv rows is fully masked for it
       const bool stage_active = (q0t + TQ - 1 + q_off >= wv_kv0);
 
@@ -844,8 +885,8 @@ struct DKDVV2 {
         constexpr int sub = decltype(sub_)::value;
         const int q0 = q0t + sub * QBLK;
         if (q0 + QBLK - 1 + q_off < wv_kv0) return;  // sub fully masked
-        const float* lse_sub = lse_s + sub * 32;
-        const float* delta_sub = delta_s + sub * 32;
+        const float* lse_sub = lse_b + sub * 32;
+        const float* delta_sub = delta_b + sub * 32;
         // S[q][kv], dP[q][kv] (col = kv = ln31, rows q = crow)
         floatx16 s_acc, dp_acc;
 #pragma unroll
@@ -854,12 +895,12 @@ struct DKDVV2 {
         for (int s = 0; s < NDSTEP; ++s) {
           const int col = hi * 8 + 16 * s;
           bf16x8 qa = *reinterpret_cast<const bf16x8*>(
-              &lds[O_QROW + sub * (QBLK * DP) + ln31 * DP + col]);
+              &ldsb[O_QROW + sub * (QBLK * DP) + ln31 * DP + col]);
           s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[s], s_acc,
                                                           0, 0, 0);
           if constexpr (DO_DK) {
             bf16x8 da = *reinterpret_cast<const bf16x8*>(
-                &lds[O_DOROW + sub * (QBLK * DP) + ln31 * DP + col]);
+                &ldsb[O_DOROW + sub * (QBLK * DP) + ln31 * DP + col]);
             dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vfrag[s],
                                                              dp_acc, 0, 0, 0);
           }
@@ -955,13 +996,15 @@ struct DKDVV2 {
       });  // sub / stage_active
 
       __syncthreads();
-      if (t + 1 < n_iter) {
-        stage_write(st, smem_raw);
-        if (t + 2 < n_iter) {
-          cursor_next(cp);
-          do_stage_load(st, cp);
+      if constexpr (!DBUF) {
+        if (t + 1 < n_iter) {
+          stage_write(st, smem_raw);
+          if (t + 2 < n_iter) {
+            cursor_next(cp);
+            do_stage_load(st, cp);
+          }
+          __syncthreads();
         }
-        __syncthreads();
       }
     }
 
@@ -1093,9 +1136,9 @@ std::vector<torch::Tensor> attn_fwd2(torch::Tensor q, torch::Tensor k,
                        (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),
                        B, Tq, Tkv, (int)q_offset, Hq, Hkv, scale);
   };
-  if (D == 128) launch(attn2::fwd2_kernel<128>, sizeof(typename attn2::FwdV2<128>::Smem));
-  else if (D == 80) launch(attn2::fwd2_kernel<80>, sizeof(typename attn2::FwdV2<80>::Smem));
-  else launch(attn2::fwd2_kernel<64>, sizeof(typename attn2::FwdV2<64>::Smem));
+  if (D == 128) launch(attn2::fwd2_kernel<128>, attn2::FwdV2<128>::SMEM_BYTES);
+  else if (D == 80) launch(attn2::fwd2_kernel<80>, attn2::FwdV2<80>::SMEM_BYTES);
+  else launch(attn2::fwd2_kernel<64>, attn2::FwdV2<64>::SMEM_BYTES);
   HIP_CHECK_KERNEL();
   return {o, lse};
 }
