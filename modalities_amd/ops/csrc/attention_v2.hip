@@ -375,15 +375,20 @@ struct DQV2 {
   static constexpr int NDSTEP = G::NDSTEP, NDBLK = G::NDBLK;
   static constexpr int WG_Q = QBLK * NW;
 
-  // NSUB kv sub-tiles of KVBLK rows staged per barrier pair (see DKDVV2)
+  // NSUB kv sub-tiles of KVBLK rows staged per barrier pair (see DKDVV2).
+  // K lives ONLY as the [D/16][KVBLK][16] panel image: the S^T A-operand
+  // reads 8 consecutive d of one kv row, which is contiguous within a
+  // panel (<=2-way bank conflict) — dropping the row-major K image makes
+  // room for double buffering (single-barrier T14 loop).
   static constexpr int NSUB = 2;
   static constexpr int TKV = KVBLK * NSUB;
   static constexpr int PANEL = (DS / 16) * KVBLK * 16;  // kt elems per sub
   struct Smem {
-    unsigned short k[NSUB * KVBLK * DP];
     unsigned short v[NSUB * KVBLK * DP];
     unsigned short kt[NSUB * PANEL];
   };
+  static constexpr int BUF_BYTES = (int)((sizeof(Smem) + 15) / 16 * 16);
+  static constexpr size_t SMEM_BYTES = (size_t)BUF_BYTES * 2;
   static constexpr int NCHS = TKV * D / 8;
   static constexpr int NCH = (NCHS + NT - 1) / NT;
 
@@ -422,8 +427,6 @@ struct DQV2 {
       const int r_ = (i * 8) / D, c = (i * 8) % D;
       const int sub = r_ / KVBLK, rin = r_ % KVBLK;
       *reinterpret_cast<shortx8*>(
-          &sm->k[sub * (KVBLK * DP) + rin * DP + c]) = r.kk[it];
-      *reinterpret_cast<shortx8*>(
           &sm->v[sub * (KVBLK * DP) + rin * DP + c]) = r.vv[it];
       *reinterpret_cast<shortx8*>(
           &sm->kt[sub * PANEL + (c >> 4) * (KVBLK * 16) + rin * 16
@@ -438,7 +441,6 @@ struct DQV2 {
                              unsigned short* dq, int B, int Tq, int Tkv,
                              int q_off, int Hq, int Hkv, float scale,
                              char* smem_raw) {
-    Smem* sm = reinterpret_cast<Smem*>(smem_raw);
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
     const int hi = lane >> 5, ln31 = lane & 31;
@@ -455,10 +457,14 @@ struct DQV2 {
 
     if constexpr (D != DS) {
 #pragma unroll
-      for (int sub = 0; sub < NSUB; ++sub)
-        for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16;
-             i += NT)
-          sm->kt[sub * PANEL + (D / 16) * KVBLK * 16 + i] = 0;
+      for (int bufi = 0; bufi < 2; ++bufi) {
+        Smem* sz = reinterpret_cast<Smem*>(smem_raw + bufi * BUF_BYTES);
+#pragma unroll
+        for (int sub = 0; sub < NSUB; ++sub)
+          for (int i = threadIdx.x; i < (DS / 16 - D / 16) * KVBLK * 16;
+               i += NT)
+            sz->kt[sub * PANEL + (D / 16) * KVBLK * 16 + i] = 0;
+      }
     }
 
     bf16x8 qfrag[NDSTEP], dofrag[NDSTEP];
@@ -496,18 +502,29 @@ struct DQV2 {
 
     Stage st;
     stage_load(st, k, v, kv_base, v_base, vp, 0, Tkv, Hkv);
-    stage_write(st, sm);
+    stage_write(st, reinterpret_cast<Smem*>(smem_raw));
     __syncthreads();
     if (n_tiles > 1) stage_load(st, k, v, kv_base, v_base, vp, TKV, Tkv, Hkv);
 
     const int lam = ln31 & 15;
-    const unsigned trb = lds_addr(sm->kt)
+    const unsigned trb0 = lds_addr(reinterpret_cast<Smem*>(smem_raw)->kt)
         + (unsigned)(ln31 >> 4) * (KVBLK * 16 * 2) + (unsigned)hi * 256
         + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 
     for (int tile = 0; tile < n_tiles; ++tile) {
       const int kv0t = tile * TKV;
       const bool stage_active = (kv0t <= q_hi_wave);
+      const unsigned boff = (unsigned)(tile & 1) * BUF_BYTES;
+      Smem* const sm = reinterpret_cast<Smem*>(smem_raw + boff);
+      const unsigned trb = trb0 + boff;
+      // single-barrier T14 (see FwdV2)
+      if (tile + 1 < n_tiles) {
+        stage_write(st, reinterpret_cast<Smem*>(
+            smem_raw + (unsigned)(~tile & 1) * BUF_BYTES));
+        if (tile + 2 < n_tiles)
+          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * TKV,
+                     Tkv, Hkv);
+      }
 
       if (stage_active) attnc::static_for<NSUB>([&](auto sub_) {
         constexpr int sub = decltype(sub_)::value;
@@ -524,8 +541,10 @@ struct DQV2 {
           for (int s = 0; s < NDSTEP; ++s) {
             const int col = hi * 8 + 16 * s;
             const int r_ = 32 * half + ln31;
+            // K row-read from the panel image: 8 consecutive d of row r_
+            // are contiguous within panel s (<=2-way bank conflict)
             bf16x8 ka = *reinterpret_cast<const bf16x8*>(
-                &sm->k[sub * (KVBLK * DP) + r_ * DP + col]);
+                &sm->kt[sub * PANEL + s * (KVBLK * 16) + r_ * 16 + hi * 8]);
             bf16x8 va = *reinterpret_cast<const bf16x8*>(
                 &sm->v[sub * (KVBLK * DP) + r_ * DP + col]);
             s_h = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[s], s_h,
@@ -583,14 +602,7 @@ struct DQV2 {
         });  // half
       });  // sub / stage_active
 
-      __syncthreads();
-      if (tile + 1 < n_tiles) {
-        stage_write(st, sm);
-        if (tile + 2 < n_tiles)
-          stage_load(st, k, v, kv_base, v_base, vp, (tile + 2) * TKV,
-                     Tkv, Hkv);
-        __syncthreads();
-      }
+      __syncthreads();  // tile reads done AND tile+1 writes landed
     }
 
     if (qg < Tq) {
@@ -1225,7 +1237,7 @@ static std::vector<torch::Tensor> attn_bwd2_impl(
     // dV-only / dK-only kernels concurrently instead
     using DV = attn2::DKDVV2<128, 1>;
     using DK = attn2::DKDVV2<128, 2>;
-    launch_dq(attn2::dq2_kernel<128>, sizeof(typename attn2::DQV2<128>::Smem));
+    launch_dq(attn2::dq2_kernel<128>, attn2::DQV2<128>::SMEM_BYTES);
     hipStreamWaitEvent(side2, ev_fork, 0);
     used_side2 = true;
     launch_kv(attn2::dkdv2_kernel<128, 1>, DV::WG_KV, DV::NTV, DV::SMEM_BYTES,
@@ -1239,7 +1251,7 @@ static std::vector<torch::Tensor> attn_bwd2_impl(
       const char* e = getenv("MA_DKDV80_SPLIT");
       return e ? atoi(e) != 0 : false;
     }();
-    launch_dq(attn2::dq2_kernel<80>, sizeof(typename attn2::DQV2<80>::Smem));
+    launch_dq(attn2::dq2_kernel<80>, attn2::DQV2<80>::SMEM_BYTES);
     if (split80) {
       using DV = attn2::DKDVV2<80, 1>;
       using DK = attn2::DKDVV2<80, 2>;
@@ -1256,7 +1268,7 @@ static std::vector<torch::Tensor> attn_bwd2_impl(
     }
   } else {
     using F = attn2::DKDVV2<64, 0>;
-    launch_dq(attn2::dq2_kernel<64>, sizeof(typename attn2::DQV2<64>::Smem));
+    launch_dq(attn2::dq2_kernel<64>, attn2::DQV2<64>::SMEM_BYTES);
     launch_kv(attn2::dkdv2_kernel<64, 0>, F::WG_KV, F::NTV, F::SMEM_BYTES,
               side);
   }
