@@ -165,6 +165,7 @@ def test_attn_fwd(B, T, Hq, Hkv, D, impl):
     (2, 256, 4, 2, 64),
     (2, 256, 4, 2, 80),
     (1, 300, 4, 4, 80),
+    (2, 192, 8, 2, 80),    # GQA at the reference head_dim
     (1, 1024, 2, 2, 128),
 ])
 def test_attn_bwd(B, T, Hq, Hkv, D, impl):
@@ -350,14 +351,15 @@ def test_hipgraph_step_matches_eager():
 
 
 @pytest.mark.gpu
+@pytest.mark.parametrize("D", [128, 80])
 @pytest.mark.parametrize("Tq,Tkv,off", [(128, 512, 0), (128, 512, 128),
                                         (128, 512, 384), (100, 300, 100)])
-def test_attn_offset_causal_fwd_bwd(Tq, Tkv, off):
+def test_attn_offset_causal_fwd_bwd(Tq, Tkv, off, D):
     """Context-parallel shape: local q chunk at global key offset `off`
     against full keys (offset-causal). fwd + bwd vs fp32 reference."""
     from modalities_amd.ops.attention import _attention_ref
     torch.manual_seed(11)
-    B, Hq, Hkv, D = 2, 4, 2, 128
+    B, Hq, Hkv = 2, 4, 2
     q = bf(torch.randn(B, Tq, Hq, D, device=DEV))
     k = bf(torch.randn(B, Tkv, Hkv, D, device=DEV))
     v = bf(torch.randn(B, Tkv, Hkv, D, device=DEV))
