@@ -116,8 +116,8 @@ def main():
     if args.seq_len is None:
         args.seq_len = cfg.sequence_length
     cfg.sequence_length = args.seq_len
-    if args.tp == 1:
-        cfg.fused_qkv = True  # one [h, h+2kv] GEMM per block (MI355X-first)
+    cfg.fused_qkv = True  # one [h, h+2kv] GEMM per block (MI355X-first;
+    # TP shards the joint weight per head group since r2)
     torch.manual_seed(1234)  # identical init on all ranks
     model = GPT2LLM(cfg)
     num_params = sum(p_.numel() for p_ in model.parameters())

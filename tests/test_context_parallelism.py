@@ -315,3 +315,66 @@ def test_ring_bf16_output_dtype():
                               port=find_free_port())
     for r in range(2):
         assert results[r] == ("torch.bfloat16", "torch.bfloat16")
+
+
+def _cp_trainer_worker(rank, world):
+    """ADVICE r1 #3 end-to-end: the TRAINER slices targets for CP and sums
+    partial grads over the cp group — a config-driven CP run must match
+    single-process training."""
+    import torch.distributed as dist
+
+    from modalities_amd.batch import DatasetBatch
+    from modalities_amd.loss_functions import CLMCrossEntropyLoss
+    from modalities_amd.parallel.cp import get_gpt2_context_parallel_model
+    from modalities_amd.training.trainer import Trainer
+
+    torch.manual_seed(0)
+    model = GPT2LLM(tiny_cfg())
+    model = get_gpt2_context_parallel_model(model, group=dist.group.WORLD,
+                                            cp_rank=rank, cp_size=world)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3, betas=(0.9, 0.95),
+                            eps=1e-8, weight_decay=0.0)
+    trainer = Trainer(
+        global_rank=rank, progress_publisher=_SilentPub(),
+        evaluation_result_publisher=_SilentPub(), gradient_acc_steps=1,
+        global_num_tokens_per_train_step=1, num_seen_train_steps=0,
+        global_num_seen_tokens=0, num_target_steps=2, num_target_tokens=2,
+        training_log_interval_in_steps=100)
+    losses = []
+    for i in range(2):
+        x, y = make_batch(700 + i)
+        batch = DatasetBatch(samples={"input_ids": x},
+                             targets={"target_ids": y})
+        _, loss, _ = trainer._train_batch(
+            batch, model, opt, None,
+            CLMCrossEntropyLoss("target_ids", "logits"), i)
+        g = loss.detach().clone()
+        dist.all_reduce(g)
+        losses.append(g.item() / world)  # mean of per-rank local means
+    return losses
+
+
+class _SilentPub:
+    def publish_message(self, *a, **k):
+        pass
+
+
+def test_cp_trainer_wiring_matches_single_process():
+    torch.manual_seed(0)
+    ref_model = GPT2LLM(tiny_cfg())
+    ref_opt = torch.optim.AdamW(ref_model.parameters(), lr=1e-3,
+                                betas=(0.9, 0.95), eps=1e-8, weight_decay=0.0)
+    ref = []
+    for i in range(2):
+        x, y = make_batch(700 + i)
+        out = ref_model({"input_ids": x})["logits"]
+        loss = torch.nn.functional.cross_entropy(
+            out.reshape(-1, VOCAB).float(), y.reshape(-1))
+        loss.backward()
+        ref_opt.step()
+        ref_opt.zero_grad()
+        ref.append(loss.item())
+    results = run_distributed(_cp_trainer_worker, world_size=2,
+                              port=find_free_port())
+    for r in range(2):
+        assert results[r] == pytest.approx(ref, rel=2e-4), (results[r], ref)
