@@ -203,12 +203,6 @@ struct FwdV2 {
         + (unsigned)(ln31 >> 4) * (KVBLK * 16 * 2) + (unsigned)hi * 256
         + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 
-    // T5 static form: the second-dispatched half of an 8-wave workgroup
-    // loses VALU arbitration on every segment; one static priority raise
-    // (wave-uniform via readfirstlane: s_setprio ignores EXEC) removes
-    // its start-of-segment penalty (guide §5.5 T5).
-    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= NT / 2)
-      __builtin_amdgcn_s_setprio(1);
 
     for (int tile = 0; tile < n_tiles; ++tile) {
       const int kv0t = tile * TKV;
@@ -518,8 +512,6 @@ struct DQV2 {
         + (unsigned)(ln31 >> 4) * (KVBLK * 16 * 2) + (unsigned)hi * 256
         + ((lam >> 2) * 16 + 4 * (lam & 3)) * 2;
 
-    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= NT / 2)
-      __builtin_amdgcn_s_setprio(1);  // T5 static form (see FwdV2)
 
     for (int tile = 0; tile < n_tiles; ++tile) {
       const int kv0t = tile * TKV;
@@ -878,10 +870,6 @@ struct DKDVV2 {
     const unsigned qt_trb0 = lds_addr(lds + O_QT) + tr_lane;
     const unsigned dot_trb0 = lds_addr(lds + O_DOT) + tr_lane;
 
-    if constexpr (MODE == 0) {  // 8-wave WG: T5 static form (see FwdV2)
-      if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= NTV / 2)
-        __builtin_amdgcn_s_setprio(1);
-    }
 
     for (int t = 0; t < n_iter; ++t) {
       const int q0t = cc.q0;
