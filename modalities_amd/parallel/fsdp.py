@@ -165,6 +165,14 @@ class FlatParamUnit:
         if streams.on_gpu and self._gather_event is not None:
             torch.cuda.current_stream(self.device).wait_event(self._gather_event)
             self._gather_event = None
+            # full_buf was allocated on the gather stream but is consumed
+            # (and eventually freed) on the compute stream: tell the caching
+            # allocator so it cannot hand the pages to another stream while
+            # compute still reads them (classic cross-stream FSDP hazard;
+            # only reachable at world > 1 on GPU).
+            if self.full_buf is not None and self.full_buf.is_cuda:
+                self.full_buf.record_stream(
+                    torch.cuda.current_stream(self.device))
         self._point_params_to(self.full_buf)
 
     @torch.no_grad()
@@ -217,6 +225,12 @@ class FlatParamUnit:
             if streams.on_gpu:
                 self._reduce_event = torch.cuda.Event()
                 self._reduce_event.record(streams.reduce)
+                # grad_full was allocated on the compute stream (grad views)
+                # but the collective reads it on the reduce stream; dropping
+                # the reference below frees it against the ALLOCATION stream
+                # — record the reduce stream so the allocator delays reuse
+                # until the collective completed (world > 1 GPU only).
+                gf.record_stream(streams.reduce)
         for _, _, p in self.param_infos:
             p.grad = None
         self.grad_full = None
