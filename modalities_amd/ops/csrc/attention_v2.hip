@@ -26,6 +26,9 @@
 // (src/modalities/models/gpt2/gpt2_model.py:595-658).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPCachingAllocator.h>
+#include <c10/hip/HIPStream.h>
+
 #include "attn_common.h"
 
 namespace attn2 {
@@ -1280,6 +1283,23 @@ static std::vector<torch::Tensor> attn_bwd2_impl(
   if (used_side2) {
     hipEventRecord(ev_join2, side2);
     hipStreamWaitEvent(stream, ev_join2, 0);
+  }
+  // dk/dv (and the delta reads) are touched by kernels on the side
+  // streams but were allocated on the caller's stream: record the side
+  // streams so the caching allocator cannot hand their pages to another
+  // stream before the side work completes (join orders only `stream`).
+  auto rec = [&](const torch::Tensor& t, hipStream_t st) {
+    c10::hip::HIPCachingAllocator::recordStream(
+        t.storage().data_ptr(),
+        c10::hip::getStreamFromExternal(st, q.get_device()));
+  };
+  rec(dk, side);
+  rec(dv, side);
+  rec(delta, side);
+  if (used_side2) {
+    rec(dk, side2);
+    rec(dv, side2);
+    rec(delta, side2);
   }
   return {dq, dk, dv};
 }
