@@ -69,6 +69,9 @@ def main():
     p.add_argument("--reshard", action="store_true")
     p.add_argument("--ac", action="store_true",
                    help="full activation checkpointing")
+    p.add_argument("--ac-variant", type=str, default="full",
+                   choices=["full", "selective_layer", "selective_op"],
+                   help="checkpointing variant when --ac is set")
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree")
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph step capture")
@@ -136,8 +139,14 @@ def main():
     if args.ac:
         from modalities_amd.training.activation_checkpointing import (
             ActivationCheckpointingVariant, apply_activation_checkpointing_)
-        apply_activation_checkpointing_(
-            model, ActivationCheckpointingVariant.FULL_ACTIVATION_CHECKPOINTING)
+        variant = {
+            "full": ActivationCheckpointingVariant.FULL_ACTIVATION_CHECKPOINTING,
+            "selective_layer":
+                ActivationCheckpointingVariant.SELECTIVE_LAYER_ACTIVATION_CHECKPOINTING,
+            "selective_op":
+                ActivationCheckpointingVariant.SELECTIVE_OP_ACTIVATION_CHECKPOINTING,
+        }[args.ac_variant]
+        apply_activation_checkpointing_(model, variant)
 
     if mesh is not None:
         from modalities_amd.models.model_factory import ModelFactory
