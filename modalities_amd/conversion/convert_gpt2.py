@@ -121,4 +121,11 @@ def convert_gpt2_to_hf(config_path: Path, output_dir: Path,
     output_dir = Path(output_dir)
     output_dir.mkdir(parents=True, exist_ok=True)
     hf.save_pretrained(output_dir)
+
+    # ship a loadable tokenizer with the export when the config names an
+    # SP tokenizer (reference convert_gpt2.py + conversion_tokenizer.py)
+    tok = config_dict.get("tokenizer")
+    if tok and tok.get("variant_key") == "pretrained_sp_tokenizer":
+        from modalities_amd.conversion.convert_tokenizer import convert_tokenizer
+        convert_tokenizer(tok["config"]["tokenizer_model_file"], str(output_dir))
     return hf

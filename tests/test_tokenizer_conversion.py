@@ -1,0 +1,40 @@
+"""SentencePiece → HF tokenizer conversion (reference
+conversion/gpt2/conversion_tokenizer.py behavior): the saved HF tokenizer
+must encode identically to the raw SentencePiece processor."""
+
+import pytest
+
+sentencepiece = pytest.importorskip("sentencepiece")
+transformers = pytest.importorskip("transformers")
+
+
+@pytest.fixture(scope="module")
+def sp_model(tmp_path_factory):
+    root = tmp_path_factory.mktemp("sp")
+    corpus = root / "corpus.txt"
+    words = ["alpha", "beta", "gamma", "delta", "omega", "train", "model", "token"]
+    import random
+    rng = random.Random(3)
+    corpus.write_text("\n".join(" ".join(rng.choices(words, k=12))
+                                for _ in range(400)))
+    sentencepiece.SentencePieceTrainer.train(
+        input=str(corpus), model_prefix=str(root / "tiny"),
+        vocab_size=64, model_type="bpe", minloglevel=2)
+    return root / "tiny.model"
+
+
+def test_sp_to_hf_conversion_roundtrip(sp_model, tmp_path):
+    from modalities_amd.conversion.convert_tokenizer import convert_tokenizer
+    from modalities_amd.tokenization.tokenizer_wrapper import PreTrainedSPTokenizer
+
+    bos, eos, pad, unk = convert_tokenizer(str(sp_model), str(tmp_path / "hf_tok"))
+    assert bos == 1 and eos == 2 and pad == -1 and unk == 0  # SP BPE defaults
+
+    hf = transformers.LlamaTokenizer.from_pretrained(str(tmp_path / "hf_tok"))
+    sp = PreTrainedSPTokenizer(str(sp_model))
+    for text in ["alpha beta gamma train model",
+                 "token omega delta", "alphabeta  gamma"]:
+        assert hf.encode(text) == sp.tokenize(text), text
+        assert hf.decode(hf.encode(text)) == sp.decode(sp.tokenize(text))
+    # no HF-side bos/eos insertion
+    assert 1 not in hf.encode("alpha") and 2 not in hf.encode("alpha")
