@@ -33,7 +33,7 @@ def test_sp_to_hf_conversion_roundtrip(sp_model, tmp_path):
     hf = transformers.LlamaTokenizer.from_pretrained(str(tmp_path / "hf_tok"))
     sp = PreTrainedSPTokenizer(str(sp_model))
     for text in ["alpha beta gamma train model",
-                 "token omega delta", "alphabeta  gamma"]:
+                 "token omega delta", "alphabeta gamma"]:
         assert hf.encode(text) == sp.tokenize(text), text
         assert hf.decode(hf.encode(text)) == sp.decode(sp.tokenize(text))
     # no HF-side bos/eos insertion
