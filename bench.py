@@ -146,7 +146,9 @@ def main():
             "selective_op":
                 ActivationCheckpointingVariant.SELECTIVE_OP_ACTIVATION_CHECKPOINTING,
         }[args.ac_variant]
-        apply_activation_checkpointing_(model, variant)
+        # selective_layer means every-2nd layer (k=1 would equal FULL)
+        k = 2 if args.ac_variant == "selective_layer" else 1
+        apply_activation_checkpointing_(model, variant, every_k_layers=k)
 
     if mesh is not None:
         from modalities_amd.models.model_factory import ModelFactory
