@@ -223,3 +223,26 @@ def test_attention_dropout_falls_back_to_sdpa():
     model.eval()
     out = model({"input_ids": torch.randint(0, 64, (2, 16))})
     assert out["logits"].shape == (2, 16, 64)
+
+
+def test_torch_compile_blockwise_matches_eager():
+    """get_compiled_model (reference tests/test_torch_compile.py behavior):
+    per-block compile keeps outputs equal to eager and blocks compiled."""
+    from modalities_amd.models.model_factory import ModelFactory
+
+    torch.manual_seed(0)
+    model = GPT2LLM(cfg())
+    g = torch.Generator().manual_seed(4)
+    ids = torch.randint(0, VOCAB, (2, 16), generator=g)
+    with torch.no_grad():
+        ref = model({"input_ids": ids})["logits"].clone()
+
+    compiled = ModelFactory.get_compiled_model(model, block_names=["GPT2Block"])
+    from torch._dynamo import OptimizedModule  # noqa: PLC0415
+    n_compiled = sum(isinstance(m, OptimizedModule) or
+                     getattr(m, "_compiled_call_impl", None) is not None
+                     for m in compiled.modules())
+    assert n_compiled >= 2, "expected every block compiled"
+    with torch.no_grad():
+        out = compiled({"input_ids": ids})["logits"]
+    torch.testing.assert_close(out, ref, rtol=2e-4, atol=2e-4)
