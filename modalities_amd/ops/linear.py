@@ -24,6 +24,17 @@ _side_stream: Optional[torch.cuda.Stream] = None
 _TWO_STREAM_MAX_WEIGHT = int(os.environ.get(
     "MODALITIES_AMD_TWO_STREAM_MAX_WEIGHT", 64 * 1024 * 1024))
 
+# Same reasoning for the activation row count (tokens per micro-batch), but
+# with a harder failure mode: above ~8k rows hipBLASLt selects persistent /
+# Stream-K kernels for BOTH backward GEMMs, and two device-filling kernels
+# with intra-kernel global synchronization co-running on concurrent HIP
+# streams can wedge the device outright (spinning workgroups hold CUs the
+# other kernel's unlaunched workgroups need — observed as a hard hang at
+# micro-batch >= 3 x 4096 tokens on MI355X, see tools/debug_b4f.py).
+# Shapes above this row bound run the standard sequential backward.
+_TWO_STREAM_MAX_ROWS = int(os.environ.get(
+    "MODALITIES_AMD_TWO_STREAM_MAX_ROWS", 8192))
+
 
 def _get_side_stream() -> torch.cuda.Stream:
     global _side_stream
@@ -44,7 +55,7 @@ class _TwoStreamLinearFn(torch.autograd.Function):
         x, weight = ctx.saved_tensors
         dy2 = dy.reshape(-1, dy.shape[-1])
         x2 = x.reshape(-1, x.shape[-1])
-        if not dy.is_cuda:
+        if not dy.is_cuda or dy2.shape[0] > _TWO_STREAM_MAX_ROWS:
             dw = dy2.t() @ x2
             db = dy2.sum(0) if ctx.has_bias else None
             dx = (dy2 @ weight).view_as(x)

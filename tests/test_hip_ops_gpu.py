@@ -519,3 +519,33 @@ def test_flash_attention_custom_op_and_selective_ac_save():
     out = model({"input_ids": ids})["logits"]
     out.float().mean().backward()
     assert model.wte.weight.grad is not None
+
+
+@pytest.mark.gpu
+def test_two_stream_linear_row_bound_matches_plain():
+    """Above _TWO_STREAM_MAX_ROWS the linear backward must take the plain
+    sequential path (two chip-filling hipBLASLt kernels co-running on
+    concurrent streams can wedge the device — the mbs>=3 hang), and its
+    gradients must match nn.Linear on both sides of the bound."""
+    import torch.nn as nn
+
+    from modalities_amd.ops import linear as tsl
+
+    torch.manual_seed(0)
+    for rows in (4096, tsl._TWO_STREAM_MAX_ROWS + 64):
+        ref = nn.Linear(256, 512, bias=True, device=DEV, dtype=torch.bfloat16)
+        two = tsl.TwoStreamLinear(256, 512, bias=True, device=DEV,
+                                  dtype=torch.bfloat16)
+        with torch.no_grad():
+            two.weight.copy_(ref.weight)
+            two.bias.copy_(ref.bias)
+        x1 = torch.randn(rows, 256, device=DEV, dtype=torch.bfloat16,
+                         requires_grad=True)
+        x2 = x1.detach().clone().requires_grad_(True)
+        ref(x1).float().square().mean().backward()
+        two(x2).float().square().mean().backward()
+        torch.cuda.synchronize()
+        torch.testing.assert_close(x1.grad, x2.grad, rtol=0, atol=0)
+        torch.testing.assert_close(ref.weight.grad, two.weight.grad,
+                                   rtol=0, atol=0)
+        torch.testing.assert_close(ref.bias.grad, two.bias.grad, rtol=0, atol=0)
