@@ -1,0 +1,96 @@
+"""Functional coverage for text generation (reference
+inference/text/inference_component.py behavior) and the communication
+self-test (utils/communication_test.py) on gloo world 2."""
+
+import torch
+
+from modalities_amd.inference.text_generation import TextInferenceComponent
+from modalities_amd.models.gpt2 import GPT2LLM, GPT2LLMConfig
+from modalities_amd.tokenization.tokenizer_wrapper import CharTokenizer
+
+from tests.utils_dist import run_distributed
+
+
+def _tiny_model(vocab=260, seq=48):
+    torch.manual_seed(7)
+    return GPT2LLM(GPT2LLMConfig(
+        vocab_size=vocab, n_layer=2, n_head_q=4, n_head_kv=4, n_embd=64,
+        ffn_hidden=128, sequence_length=seq, seed=5, dropout=0.0))
+
+
+class CountingModel:
+    def __init__(self, inner):
+        self.inner = inner
+        self.calls = 0
+
+    def eval(self):
+        self.inner.eval()
+
+    def __call__(self, inputs):
+        self.calls += 1
+        return self.inner(inputs)
+
+
+def test_greedy_generation_is_deterministic_and_bounded():
+    tok = CharTokenizer()
+    counting = CountingModel(_tiny_model())
+    gen = TextInferenceComponent(counting, tok, prompt_template="{text}",
+                                 sequence_length=24, temperature=0.0)
+    out1 = gen.generate_tokens("hello")
+    calls1 = counting.calls
+    out2 = gen.generate_tokens("hello")
+    assert out1 == out2  # greedy = deterministic
+    # prompt is 5 tokens; at most seq_len - prompt forward steps per call
+    assert calls1 <= 24 - 5
+
+
+def test_generation_stops_at_eod():
+    tok = CharTokenizer()
+    model = _tiny_model()
+
+    class EodAfterTwo:
+        """Wrap the model to force the eod token at the 3rd step."""
+
+        def __init__(self, inner):
+            self.inner = inner
+            self.calls = 0
+
+        def eval(self):
+            self.inner.eval()
+
+        def __call__(self, inputs):
+            self.calls += 1
+            out = self.inner(inputs)
+            if self.calls >= 3:
+                out["logits"] = out["logits"].clone()
+                out["logits"][:, -1, :] = -1e9
+                out["logits"][:, -1, tok.get_token_id("<eod>")] = 1e9
+            return out
+
+    gen = TextInferenceComponent(EodAfterTwo(model), tok,
+                                 prompt_template="{text}",
+                                 sequence_length=40, temperature=0.0)
+    out = gen.generate_tokens("ab")
+    assert gen.model.calls == 3  # stopped right at the forced eod step
+    assert "<eod>" not in out
+
+
+def test_temperature_sampling_runs():
+    tok = CharTokenizer()
+    model = _tiny_model()
+    torch.manual_seed(11)
+    gen = TextInferenceComponent(model, tok, prompt_template="{text}",
+                                 sequence_length=16, temperature=0.8)
+    out = gen.generate_tokens("xy")
+    assert isinstance(out, str)
+
+
+def _comm_worker(rank, world):
+    from modalities_amd.utils.communication_test import run_communication_test
+    run_communication_test(device=torch.device("cpu"))
+    return "ok"
+
+
+def test_communication_self_test_world2():
+    results = run_distributed(_comm_worker, world_size=2, port=29461)
+    assert results == {0: "ok", 1: "ok"}
