@@ -48,3 +48,35 @@ def test_config_driven_tp2_trains(tmp_path):
     assert losses and all(np.isfinite(v) for v in losses), losses
     # checkpoints written
     assert (tmp_path / "ckpt" / "tp2_cfg" / "last_checkpoint_info.json").exists()
+
+
+def _rank_main_pp(rank, world, cfg_path):
+    from modalities_amd.main import Main
+    main_obj = Main(Path(cfg_path), experiment_id="pp2_cfg")
+    components = main_obj.build_components()
+    assert components.pp_schedule is not None
+    main_obj.run(components)
+    return "ok"
+
+
+def test_config_driven_pp2_trains(tmp_path):
+    """YAML-driven PP: device_mesh(pp=2) -> staged schedule ->
+    pipelined_model selector -> Main.run at world 2 on gloo."""
+    rng = np.random.default_rng(9)
+    from modalities_amd.dataloader.packed_data import write_pbin
+    docs = [rng.integers(0, 256, size=200, dtype=np.uint8) for _ in range(8)]
+    pbin = tmp_path / "data.pbin"
+    write_pbin(pbin, docs, token_size_in_bytes=1)
+
+    template = Path(__file__).parent / "configs" / "config_tiny_e2e_pp2.yaml"
+    text = template.read_text()
+    text = text.replace("DATASET_PATH_PLACEHOLDER", str(pbin))
+    text = text.replace("CHECKPOINT_DIR_PLACEHOLDER", str(tmp_path / "ckpt"))
+    text = text.replace("RESULTS_PATH_PLACEHOLDER",
+                        str(tmp_path / "evaluation_results.jsonl"))
+    cfg = tmp_path / "config.yaml"
+    cfg.write_text(text)
+
+    results = run_distributed(_rank_main_pp, world_size=2, port=29473,
+                              args=(str(cfg),), timeout_s=420)
+    assert results == {0: "ok", 1: "ok"}
