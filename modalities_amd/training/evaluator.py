@@ -35,7 +35,19 @@ class Evaluator:
             losses: list = []
             self.pp_schedule.eval_step(inputs, targets, loss_fun, losses)
             return self.pp_schedule.broadcast_mean_loss(losses)
-        result_batch = model_predict_batch(model, batch.to(self.device))
+        batch = batch.to(self.device)
+        cp_info = Trainer._cp_info_of(model)
+        if cp_info is not None:
+            # CP-patched model returns seq-sharded logits [B, T/cp, V]:
+            # slice the targets identically (same as Trainer.train_step)
+            from modalities_amd.batch import DatasetBatch
+            from modalities_amd.parallel.cp import slice_targets_for_cp
+            group, cp_rank, cp_size = cp_info
+            batch = DatasetBatch(
+                samples=batch.samples,
+                targets={k: slice_targets_for_cp(v, cp_rank, cp_size)
+                         for k, v in batch.targets.items()})
+        result_batch = model_predict_batch(model, batch)
         return loss_fun(result_batch)
 
     def evaluate(self, model, data_loaders: list, loss_fun: Loss,
