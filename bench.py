@@ -286,7 +286,7 @@ def main():
             "dtype": "bf16" if on_gpu else "fp32",
             "data": "synthetic",
             "mfu": mfu,
-            "config": {"model": args.model, "global_batch": B * world,
+            "config": {"model": args.model, "global_batch": B * dp_world,
                        "micro_batch": B, "seq_len": T,
                        "parallelism": (f"dp{dp_world}_tp{args.tp}" if args.tp > 1
                                        else f"dp{world}_fullshard")
