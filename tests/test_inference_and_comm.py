@@ -94,3 +94,58 @@ def _comm_worker(rank, world):
 def test_communication_self_test_world2():
     results = run_distributed(_comm_worker, world_size=2, port=29461)
     assert results == {0: "ok", 1: "ok"}
+
+
+def test_generate_text_config_path(tmp_path):
+    """The `generate_text` CLI's component path: config -> model+tokenizer
+    -> TextInferenceComponent (non-interactive; we call generate_tokens
+    instead of the stdin loop)."""
+    cfg_text = """\
+settings:
+  referencing_keys:
+    sample_key: input_ids
+    prediction_key: logits
+  device: cpu
+  sequence_length: 24
+
+model:
+  component_key: model
+  variant_key: gpt2
+  config:
+    sample_key: input_ids
+    prediction_key: logits
+    vocab_size: 260
+    n_layer: 2
+    n_head_q: 4
+    n_head_kv: 4
+    n_embd: 64
+    ffn_hidden: 128
+    sequence_length: 24
+    seed: 3
+
+tokenizer:
+  component_key: tokenizer
+  variant_key: char
+  config: {}
+"""
+    cfg = tmp_path / "gen.yaml"
+    cfg.write_text(cfg_text)
+
+    import torch
+
+    from modalities_amd.config.component_factory import ComponentFactory
+    from modalities_amd.config.instantiation_models import \
+        TextGenerationInstantiationModel
+    from modalities_amd.config.yaml_loader import load_app_config_dict
+    from modalities_amd.registry.components import get_default_registry
+
+    config_dict = load_app_config_dict(cfg)
+    factory = ComponentFactory(get_default_registry())
+    components = factory.build_components(config_dict,
+                                          TextGenerationInstantiationModel)
+    comp = TextInferenceComponent(
+        components.model, components.tokenizer, prompt_template="{text}",
+        sequence_length=components.settings.sequence_length, temperature=0.0,
+        device=torch.device(components.settings.device))
+    out = comp.generate_tokens("hi")
+    assert isinstance(out, str)
