@@ -1,0 +1,83 @@
+"""Property-based tests (hypothesis) for the data layer: pbin roundtrip
+across token sizes, continuous-dataset coverage, CharTokenizer roundtrip,
+and sampler skip/resume equivalence (reference test strategy §4: the data
+path must be byte-exact and resumable)."""
+
+import numpy as np
+from hypothesis import given, settings, strategies as st
+
+from modalities_amd.dataloader.packed_data import (EmbeddedStreamData,
+                                                   token_size_for_vocab,
+                                                   write_pbin)
+from modalities_amd.dataloader.samplers import ResumableDistributedSampler
+from modalities_amd.tokenization.tokenizer_wrapper import CharTokenizer
+
+DTYPES = {1: np.uint8, 2: np.uint16, 4: np.uint32}
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    token_size=st.sampled_from([1, 2, 4]),
+    docs=st.lists(st.integers(min_value=1, max_value=40), min_size=1,
+                  max_size=12),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_pbin_roundtrip_exact(tmp_path_factory, token_size, docs, seed):
+    rng = np.random.default_rng(seed)
+    hi = 2 ** (8 * token_size) - 1
+    arrays = [rng.integers(0, hi, size=n).astype(DTYPES[token_size])
+              for n in docs]
+    path = tmp_path_factory.mktemp("pbin") / "d.pbin"
+    write_pbin(path, arrays, token_size_in_bytes=token_size)
+    stream = EmbeddedStreamData(path)
+    assert stream.token_size_in_bytes == token_size
+    assert len(stream.index_base) == len(arrays)
+    assert stream.num_tokens == sum(len(a) for a in arrays)
+    for (off, length), a in zip(stream.index_base, arrays):
+        np.testing.assert_array_equal(stream.tokens(off, length), a)
+
+
+@settings(max_examples=30, deadline=None)
+@given(text=st.text(
+    alphabet=st.characters(min_codepoint=1, max_codepoint=127), max_size=200))
+def test_char_tokenizer_ascii_roundtrip(text):
+    tok = CharTokenizer()
+    ids = tok.tokenize(text)
+    assert tok.decode(ids) == text
+    assert all(0 <= t < tok.vocab_size for t in ids)
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    n=st.integers(4, 200),
+    world=st.sampled_from([1, 2, 4]),
+    skip_mult=st.integers(0, 6),
+    shuffle=st.booleans(),
+    seed=st.integers(0, 1000),
+)
+def test_sampler_skip_equals_tail_of_full_run(n, world, skip_mult, shuffle, seed):
+    """Resume contract: skipping k*world global samples yields exactly the
+    tail of the unskipped enumeration, per rank."""
+    skip = skip_mult * world
+    data = list(range(n))
+    for rank in range(world):
+        full = list(ResumableDistributedSampler(
+            dataset=data, rank=rank, num_replicas=world, shuffle=shuffle,
+            seed=seed, drop_last=True))
+        if skip >= len(full) * world:
+            continue
+        resumed = list(ResumableDistributedSampler(
+            dataset=data, rank=rank, num_replicas=world, shuffle=shuffle,
+            seed=seed, drop_last=True, skip_num_global_samples=skip))
+        assert resumed == full[skip // world:], (rank, skip)
+
+
+@settings(max_examples=20, deadline=None)
+@given(vocab=st.integers(2, 2**32 - 1))
+def test_token_size_for_vocab_bounds(vocab):
+    ts = token_size_for_vocab(vocab)
+    assert ts in (1, 2, 4)
+    assert vocab <= 2 ** (8 * ts)
+    if ts > 1:
+        smaller = {2: 1, 4: 2}[ts]
+        assert vocab > 2 ** (8 * smaller)
