@@ -81,3 +81,50 @@ def test_token_size_for_vocab_bounds(vocab):
     if ts > 1:
         smaller = {2: 1, 4: 2}[ts]
         assert vocab > 2 ** (8 * smaller)
+
+
+@settings(max_examples=50, deadline=None)
+@given(
+    dp=st.sampled_from([1, 2, 4, 8]),
+    mbs=st.integers(1, 8),
+    acc=st.integers(1, 8),
+    seq=st.sampled_from([128, 1024, 4096]),
+    steps=st.integers(1, 10_000),
+)
+def test_number_conversion_steps_tokens_inverse(dp, mbs, acc, seq, steps):
+    """steps -> tokens -> steps must be the identity (warmstart arithmetic
+    cannot drift)."""
+    from modalities_amd.utils.number_conversion import NumberConversion
+    tokens = NumberConversion.get_num_tokens_from_num_steps(
+        num_steps=steps, dp_degree=dp, local_micro_batch_size=mbs,
+        sequence_length=seq, gradient_accumulation_steps=acc)
+    assert tokens == steps * dp * mbs * acc * seq
+    back = NumberConversion.get_num_steps_from_num_tokens(
+        dp_degree=dp, local_micro_batch_size=mbs, global_num_tokens=tokens,
+        sequence_length=seq, gradient_accumulation_steps=acc)
+    assert back == steps
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    seen=st.integers(0, 10**12),
+    target=st.integers(0, 10**12),
+    steps=st.integers(0, 10**6),
+    tsteps=st.integers(0, 10**6),
+    eid=st.text(alphabet=st.characters(whitelist_categories=("Ll", "Nd")),
+                min_size=1, max_size=12),
+)
+def test_checkpoint_path_regex_roundtrip(tmp_path_factory, seen, target, steps,
+                                         tsteps, eid):
+    """The checkpoint folder-name schema must round-trip through the
+    number-conversion regex family."""
+    from pathlib import Path
+
+    from modalities_amd.utils.number_conversion import NumberConversion
+    name = (f"eid_{eid}-seen_steps_{steps}-seen_tokens_{seen}"
+            f"-target_steps_{tsteps}-target_tokens_{target}")
+    p = Path("/ckpt") / name
+    assert NumberConversion.get_num_seen_steps_from_checkpoint_path(p) == steps
+    assert NumberConversion.get_global_num_seen_tokens_from_checkpoint_path(p) == seen
+    assert NumberConversion.get_num_target_steps_from_checkpoint_path(p) == tsteps
+    assert NumberConversion.get_global_num_target_tokens_from_checkpoint_path(p) == target
