@@ -128,3 +128,47 @@ def test_checkpoint_path_regex_roundtrip(tmp_path_factory, seen, target, steps,
     assert NumberConversion.get_global_num_seen_tokens_from_checkpoint_path(p) == seen
     assert NumberConversion.get_num_target_steps_from_checkpoint_path(p) == tsteps
     assert NumberConversion.get_global_num_target_tokens_from_checkpoint_path(p) == target
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    seq=st.lists(st.integers(0, 9), min_size=2, max_size=64),
+    seed=st.integers(0, 999),
+)
+def test_loss_masking_matches_naive_scan(seq, seed):
+    """The vectorized cumsum span masking must equal a naive per-token scan
+    (b=100 opens a span, e=101 closes it; only tokens strictly inside train)."""
+    import torch
+
+    from modalities_amd.batch import DatasetBatch
+    from modalities_amd.dataloader.dataloader import LossMaskingCollateFnWrapper
+
+    rng = np.random.default_rng(seed)
+    tokens = list(seq)
+    # sprinkle well-formed marker pairs
+    for _ in range(rng.integers(0, 3)):
+        i, j = sorted(rng.integers(0, len(tokens) + 1, size=2))
+        tokens = tokens[:i] + [100] + tokens[i:j] + [101] + tokens[j:]
+    t = torch.tensor([tokens])
+
+    def passthrough(_):
+        return DatasetBatch(samples={"input_ids": t.clone()},
+                            targets={"target_ids": t.clone()})
+
+    wrapper = LossMaskingCollateFnWrapper(
+        passthrough, target_keys_to_mask=["target_ids"],
+        loss_ignore_index=-100, b_mask_token_id=100, e_mask_token_id=101)
+    got = wrapper(None).targets["target_ids"][0].tolist()
+
+    # naive scan
+    expected, depth = [], 0
+    for tok in tokens:
+        if tok == 100:
+            expected.append(-100)
+            depth += 1
+        elif tok == 101:
+            expected.append(-100)
+            depth -= 1
+        else:
+            expected.append(tok if depth > 0 else -100)
+    assert got == expected
