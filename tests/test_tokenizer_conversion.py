@@ -38,3 +38,43 @@ def test_sp_to_hf_conversion_roundtrip(sp_model, tmp_path):
         assert hf.decode(hf.encode(text)) == sp.decode(sp.tokenize(text))
     # no HF-side bos/eos insertion
     assert 1 not in hf.encode("alpha") and 2 not in hf.encode("alpha")
+
+
+def test_convert_gpt2_to_hf_ships_sp_tokenizer(sp_model, tmp_path):
+    """convert_gpt2_to_hf writes the converted tokenizer alongside the model
+    export when the config names an SP tokenizer."""
+    cfg_text = f"""\
+model:
+  component_key: model
+  variant_key: gpt2
+  config:
+    sample_key: input_ids
+    prediction_key: logits
+    vocab_size: 64
+    n_layer: 1
+    n_head_q: 2
+    n_head_kv: 2
+    n_embd: 32
+    ffn_hidden: 64
+    sequence_length: 16
+    seed: 2
+
+tokenizer:
+  component_key: tokenizer
+  variant_key: pretrained_sp_tokenizer
+  config:
+    tokenizer_model_file: {sp_model}
+"""
+    cfg = tmp_path / "conv.yaml"
+    cfg.write_text(cfg_text)
+
+    from modalities_amd.conversion.convert_gpt2 import convert_gpt2_to_hf
+    out = tmp_path / "export"
+    convert_gpt2_to_hf(cfg, out, verify=True)
+    # model export present
+    assert (out / "config.json").exists()
+    # tokenizer shipped and loadable
+    hf_tok = transformers.LlamaTokenizer.from_pretrained(str(out))
+    from modalities_amd.tokenization.tokenizer_wrapper import PreTrainedSPTokenizer
+    sp = PreTrainedSPTokenizer(str(sp_model))
+    assert hf_tok.encode("alpha beta") == sp.tokenize("alpha beta")
