@@ -80,13 +80,30 @@ def get_raw_app_state(model, optimizer, lr_scheduler=None) -> AppState:
     return AppState(model, optimizer, lr_scheduler)
 
 
+def _mesh_partition(device_mesh):
+    """(partition, dp_rank, dp_world) for PP/TP composition: each pp/tp
+    coordinate is its own model partition with its own shard layout; dp
+    coords name the shard files inside it (None = flat layout)."""
+    if device_mesh is None:
+        return "", None, None
+    from modalities_amd.parallel.mesh import ParallelismDegrees
+    pp = device_mesh.dims[ParallelismDegrees.PP]
+    tp = device_mesh.dims[ParallelismDegrees.TP]
+    if pp.size == 1 and tp.size == 1:
+        return "", None, None
+    part = ((f"pp{pp.rank}" if pp.size > 1 else "")
+            + (f"tp{tp.rank}" if tp.size > 1 else ""))
+    return part, device_mesh.dp_rank, device_mesh.dp_degree
+
+
 def get_warmstart_app_state(model, optimizer, checkpoint_folder_path: Path,
-                            lr_scheduler=None) -> AppState:
+                            lr_scheduler=None, device_mesh=None) -> AppState:
     import torch.distributed as dist
     app_state = AppState(model, optimizer, lr_scheduler)
     rank = dist.get_rank() if dist.is_initialized() else 0
-    ShardedCheckpointLoading(rank).load_checkpoint_(app_state,
-                                                    Path(checkpoint_folder_path))
+    part, _, _ = _mesh_partition(device_mesh)
+    ShardedCheckpointLoading(rank, partition=part).load_checkpoint_(
+        app_state, Path(checkpoint_folder_path))
     return app_state
 
 
@@ -96,11 +113,15 @@ def get_checkpoint_saving(checkpoint_saving_strategy,
 
 
 def get_sharded_checkpoint_saving_execution(checkpoint_path: Path,
-                                            experiment_id: str
+                                            experiment_id: str,
+                                            device_mesh=None
                                             ) -> ShardedCheckpointSaving:
     import torch.distributed as dist
     rank = dist.get_rank() if dist.is_initialized() else 0
-    return ShardedCheckpointSaving(Path(checkpoint_path), experiment_id, rank)
+    part, dp_rank, dp_world = _mesh_partition(device_mesh)
+    return ShardedCheckpointSaving(Path(checkpoint_path), experiment_id, rank,
+                                   partition=part, dp_rank=dp_rank,
+                                   dp_world=dp_world)
 
 
 def get_resumable_sampler(dataset, epoch: int = 0, shuffle: bool = False,

@@ -80,6 +80,12 @@ def test_config_driven_pp2_trains(tmp_path):
     results = run_distributed(_rank_main_pp, world_size=2, port=29473,
                               args=(str(cfg),), timeout_s=420)
     assert results == {0: "ok", 1: "ok"}
+    # each PP partition saved its own meta + shard namespace
+    ckpt_root = tmp_path / "ckpt" / "pp2_cfg"
+    folders = [p for p in ckpt_root.iterdir() if p.is_dir()]
+    assert folders
+    names = {f.name for f in folders[0].iterdir()}
+    assert "meta_pp0.json" in names and "meta_pp1.json" in names, names
 
 
 def _rank_main_cp(rank, world, cfg_path):
