@@ -182,3 +182,68 @@ def _rank_sampler_identity(rank, world):
 def test_mesh_sampler_tp_peers_share_data():
     results = run_distributed(_rank_sampler_identity, world_size=4, port=29491)
     assert all(v == "ok" for v in results.values())
+
+
+def _rank_warmstart_pp(rank, world, cfg_path, ws_cfg_path):
+    import yaml
+
+    from modalities_amd.main import Main
+
+    # run A: train 8 steps with PP2, checkpointing every 4
+    main_a = Main(Path(cfg_path), experiment_id="ppA")
+    main_a.run(main_a.build_components())
+
+    # find the step-4 checkpoint
+    root = Path(cfg_path).parent / "ckpt" / "ppA"
+    step4 = [p for p in root.iterdir() if "seen_steps_4" in p.name][0]
+
+    # run B: warmstart from it through the config path
+    def ws_resolver(key):
+        return {"checkpoint_folder_path": str(step4)}[key]
+
+    main_b = Main(Path(ws_cfg_path), experiment_id="ppB",
+                  additional_resolver_funs={"warmstart_env": ws_resolver})
+    main_b.run(main_b.build_components())
+    return "ok"
+
+
+def test_config_driven_pp2_warmstart(tmp_path):
+    """Config-path warmstart under PP2: each partition loads its own
+    meta/shard namespace via the mesh-aware app_state factory."""
+    import yaml
+
+    rng = np.random.default_rng(31)
+    from modalities_amd.dataloader.packed_data import write_pbin
+    docs = [rng.integers(0, 256, size=200, dtype=np.uint8) for _ in range(8)]
+    pbin = tmp_path / "data.pbin"
+    write_pbin(pbin, docs, token_size_in_bytes=1)
+
+    template = Path(__file__).parent / "configs" / "config_tiny_e2e_pp2.yaml"
+    text = template.read_text()
+    text = text.replace("DATASET_PATH_PLACEHOLDER", str(pbin))
+    text = text.replace("CHECKPOINT_DIR_PLACEHOLDER", str(tmp_path / "ckpt"))
+    text = text.replace("RESULTS_PATH_PLACEHOLDER",
+                        str(tmp_path / "a_results.jsonl"))
+    cfg = tmp_path / "config.yaml"
+    cfg.write_text(text)
+
+    # warmstart config: swap app_state for the mesh-aware sharded_warmstart
+    cfg_dict = yaml.safe_load(text.replace(str(tmp_path / "a_results.jsonl"),
+                                           str(tmp_path / "b_results.jsonl")))
+    cfg_dict["app_state"] = {
+        "component_key": "app_state", "variant_key": "sharded_warmstart",
+        "config": {
+            "model": {"instance_key": "wrapped_model", "pass_type": "BY_REFERENCE"},
+            "optimizer": {"instance_key": "optimizer", "pass_type": "BY_REFERENCE"},
+            "lr_scheduler": {"instance_key": "scheduler", "pass_type": "BY_REFERENCE"},
+            "checkpoint_folder_path": "${warmstart_env:checkpoint_folder_path}",
+            "device_mesh": {"instance_key": "device_mesh",
+                            "pass_type": "BY_REFERENCE"},
+        },
+    }
+    ws_cfg = tmp_path / "ws.yaml"
+    ws_cfg.write_text(yaml.safe_dump(cfg_dict, sort_keys=False))
+
+    results = run_distributed(_rank_warmstart_pp, world_size=2, port=29497,
+                              args=(str(cfg), str(ws_cfg)), timeout_s=420)
+    assert results == {0: "ok", 1: "ok"}
