@@ -247,3 +247,32 @@ def test_config_driven_pp2_warmstart(tmp_path):
     results = run_distributed(_rank_warmstart_pp, world_size=2, port=29497,
                               args=(str(cfg), str(ws_cfg)), timeout_s=420)
     assert results == {0: "ok", 1: "ok"}
+
+
+def _rank_main_hsdp(rank, world, cfg_path):
+    from modalities_amd.main import Main
+    main_obj = Main(Path(cfg_path), experiment_id="hsdp_cfg")
+    main_obj.run(main_obj.build_components())
+    return "ok"
+
+
+def test_config_driven_hsdp_world4(tmp_path):
+    """World-4 HSDP from YAML: replicate2 x shard2 mesh degrees."""
+    rng = np.random.default_rng(41)
+    from modalities_amd.dataloader.packed_data import write_pbin
+    docs = [rng.integers(0, 256, size=600, dtype=np.uint8) for _ in range(10)]
+    pbin = tmp_path / "data.pbin"
+    write_pbin(pbin, docs, token_size_in_bytes=1)
+
+    template = Path(__file__).parent / "configs" / "config_tiny_e2e_hsdp.yaml"
+    text = template.read_text()
+    text = text.replace("DATASET_PATH_PLACEHOLDER", str(pbin))
+    text = text.replace("CHECKPOINT_DIR_PLACEHOLDER", str(tmp_path / "ckpt"))
+    text = text.replace("RESULTS_PATH_PLACEHOLDER",
+                        str(tmp_path / "evaluation_results.jsonl"))
+    cfg = tmp_path / "config.yaml"
+    cfg.write_text(text)
+
+    results = run_distributed(_rank_main_hsdp, world_size=4, port=29501,
+                              args=(str(cfg),), timeout_s=420)
+    assert all(v == "ok" for v in results.values()) and len(results) == 4
