@@ -58,13 +58,17 @@ class ShardedCheckpointSaving:
 
     def __init__(self, checkpoint_path: Path, experiment_id: str,
                  global_rank: int, partition: str = "",
-                 dp_rank: int = None, dp_world: int = None):
+                 dp_rank: int = None, dp_world: int = None,
+                 write_enabled: bool = True):
         self.checkpoint_path = Path(checkpoint_path)
         self.experiment_id = experiment_id
         self.global_rank = global_rank
         self.partition = partition
         self.dp_rank = dp_rank
         self.dp_world = dp_world
+        # replicate/CP peers hold identical parameters: only their rank-0
+        # peer writes shards/meta (all ranks still hit the closing barrier)
+        self.write_enabled = write_enabled
 
     def _folder(self, progress: TrainingProgress) -> Path:
         return (self.checkpoint_path / self.experiment_id
@@ -80,10 +84,12 @@ class ShardedCheckpointSaving:
         world = self.dp_world if self.dp_world is not None else             (dist.get_world_size() if is_dist() else 1)
         rank = self.dp_rank if self.dp_rank is not None else self.global_rank
 
-        shards = {k: v.detach().to("cpu") for k, v in app_state.shard_state().items()}
-        torch.save(shards, folder / f"shards{self._suffix()}_rank_{rank}.pt")
+        if self.write_enabled:
+            shards = {k: v.detach().to("cpu")
+                      for k, v in app_state.shard_state().items()}
+            torch.save(shards, folder / f"shards{self._suffix()}_rank_{rank}.pt")
 
-        if rank == 0:
+        if rank == 0 and self.write_enabled:
             meta = {
                 "world_size": world,
                 "shard_layout": app_state.shard_layout(),

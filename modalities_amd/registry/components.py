@@ -81,19 +81,25 @@ def get_raw_app_state(model, optimizer, lr_scheduler=None) -> AppState:
 
 
 def _mesh_partition(device_mesh):
-    """(partition, dp_rank, dp_world) for PP/TP composition: each pp/tp
-    coordinate is its own model partition with its own shard layout; dp
-    coords name the shard files inside it (None = flat layout)."""
+    """(partition, shard_rank, shard_world, write_enabled) for saving under
+    parallelism composition. Each pp/tp coordinate is its own model
+    partition with its own shard layout; the persisted layout inside a
+    partition is the DP-SHARD group's (replicate/CP peers hold identical
+    parameters, so only their rank-0 peer writes)."""
     if device_mesh is None:
-        return "", None, None
+        return "", None, None, True
     from modalities_amd.parallel.mesh import ParallelismDegrees
     pp = device_mesh.dims[ParallelismDegrees.PP]
     tp = device_mesh.dims[ParallelismDegrees.TP]
-    if pp.size == 1 and tp.size == 1:
-        return "", None, None
+    rep = device_mesh.dims[ParallelismDegrees.DP_REPLICATE]
+    cp = device_mesh.dims[ParallelismDegrees.CP]
+    shard = device_mesh.dims[ParallelismDegrees.DP_SHARD]
+    write = rep.rank == 0 and cp.rank == 0
+    if pp.size == 1 and tp.size == 1 and rep.size == 1 and cp.size == 1:
+        return "", None, None, True
     part = ((f"pp{pp.rank}" if pp.size > 1 else "")
             + (f"tp{tp.rank}" if tp.size > 1 else ""))
-    return part, device_mesh.dp_rank, device_mesh.dp_degree
+    return part, shard.rank, shard.size, write
 
 
 def get_warmstart_app_state(model, optimizer, checkpoint_folder_path: Path,
@@ -101,7 +107,7 @@ def get_warmstart_app_state(model, optimizer, checkpoint_folder_path: Path,
     import torch.distributed as dist
     app_state = AppState(model, optimizer, lr_scheduler)
     rank = dist.get_rank() if dist.is_initialized() else 0
-    part, _, _ = _mesh_partition(device_mesh)
+    part, _, _, _ = _mesh_partition(device_mesh)
     ShardedCheckpointLoading(rank, partition=part).load_checkpoint_(
         app_state, Path(checkpoint_folder_path))
     return app_state
@@ -118,10 +124,10 @@ def get_sharded_checkpoint_saving_execution(checkpoint_path: Path,
                                             ) -> ShardedCheckpointSaving:
     import torch.distributed as dist
     rank = dist.get_rank() if dist.is_initialized() else 0
-    part, dp_rank, dp_world = _mesh_partition(device_mesh)
+    part, dp_rank, dp_world, write = _mesh_partition(device_mesh)
     return ShardedCheckpointSaving(Path(checkpoint_path), experiment_id, rank,
                                    partition=part, dp_rank=dp_rank,
-                                   dp_world=dp_world)
+                                   dp_world=dp_world, write_enabled=write)
 
 
 def get_resumable_sampler(dataset, epoch: int = 0, shuffle: bool = False,
