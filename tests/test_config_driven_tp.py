@@ -276,3 +276,11 @@ def test_config_driven_hsdp_world4(tmp_path):
     results = run_distributed(_rank_main_hsdp, world_size=4, port=29501,
                               args=(str(cfg),), timeout_s=420)
     assert all(v == "ok" for v in results.values()) and len(results) == 4
+    # persisted layout = the SHARD group (2 shards), written once: the
+    # replicate peers are write-gated
+    ckpt_root = tmp_path / "ckpt" / "hsdp_cfg"
+    folder = next(p for p in ckpt_root.iterdir() if p.is_dir())
+    names = sorted(f.name for f in folder.iterdir())
+    assert names == ["meta.json", "shards_rank_0.pt", "shards_rank_1.pt"], names
+    meta = json.loads((folder / "meta.json").read_text())
+    assert meta["world_size"] == 2
