@@ -43,8 +43,18 @@ class TextInferenceComponent:
         generated: list[int] = []
         max_new = max(0, self.sequence_length - input_ids.shape[1])
         self.model.eval()
-        for _ in range(max_new):
+
+        # KV-cache incremental decode when the model supports it (GPT2LLM);
+        # fall back to full-context re-forward otherwise.
+        cache = None
+        if hasattr(self.model, "forward_cached") and                 hasattr(self.model, "new_kv_cache"):
+            cache = self.model.new_kv_cache(input_ids.shape[0],
+                                            max_len=self.sequence_length)
+            out = self.model.forward_cached({self.sample_key: input_ids}, cache)
+        else:
             out = self.model({self.sample_key: input_ids})
+
+        for i in range(max_new):
             logits = out[self.prediction_key][:, -1, :].float()
             if self.temperature > 0:
                 probs = torch.softmax(logits / self.temperature, dim=-1)
@@ -58,7 +68,16 @@ class TextInferenceComponent:
             if echo:
                 sys.stdout.write(self.tokenizer.decode([token]))
                 sys.stdout.flush()
-            input_ids = torch.cat([input_ids, next_id], dim=1)
+            if i + 1 >= max_new:
+                break
+            if cache is not None:
+                if cache.pos >= self.sequence_length:
+                    break
+                out = self.model.forward_cached(
+                    {self.sample_key: next_id}, cache)
+            else:
+                input_ids = torch.cat([input_ids, next_id], dim=1)
+                out = self.model({self.sample_key: input_ids})
         if echo:
             sys.stdout.write("\n")
         return self.tokenizer.decode(generated)

@@ -178,6 +178,51 @@ class CausalSelfAttention(nn.Module):
         y = y.reshape(B, T, C)
         return self.resid_dropout(self.c_proj(y))
 
+    def forward_cached(self, x, rope_cos, rope_sin, cache_k, cache_v,
+                       pos: int):
+        """Incremental-decode step: project the T_new tokens at absolute
+        positions [pos, pos+T_new), append K/V into the preallocated cache,
+        attend q against the cache prefix with q_offset=pos (the same
+        offset-causal contract the CP path uses). Inference-only
+        (no grad); MI355X-first addition — the reference re-forwards the
+        full context per generated token."""
+        B, T, C = x.shape
+        kv_dim = self.head_dim * self.n_head_kv
+        if self.fused_qkv:
+            qkv = self.qkv_attn(x)
+            q, k, v = qkv.split([C, kv_dim, kv_dim], dim=-1)
+        else:
+            q, k, v = self.q_attn(x), self.k_attn(x), self.v_attn(x)
+        q = q.view(B, T, self.n_head_q, self.head_dim)
+        k = k.view(B, T, self.n_head_kv, self.head_dim)
+        v = v.view(B, T, self.n_head_kv, self.head_dim)
+        if self.q_norm is not None:
+            q = self.q_norm(q)
+            k = self.k_norm(k)
+        if rope_cos is not None:
+            cos, sin = rope_cos[pos:pos + T], rope_sin[pos:pos + T]
+            q = rope_apply(q, cos, sin)
+            k = rope_apply(k, cos, sin)
+        cache_k[:, pos:pos + T] = k
+        cache_v[:, pos:pos + T] = v
+        k_full = cache_k[:, :pos + T]
+        v_full = cache_v[:, :pos + T]
+        if self.attention_impl == AttentionImplementation.HIP_FLASH and                 use_hip(q):
+            y = flash_attention(q, k_full.contiguous(), v_full.contiguous(),
+                                causal=True, q_offset=pos)
+        else:
+            rep = self.n_head_q // self.n_head_kv
+            qt = q.transpose(1, 2)
+            kt = k_full.transpose(1, 2).repeat_interleave(rep, dim=1)
+            vt = v_full.transpose(1, 2).repeat_interleave(rep, dim=1)
+            S = pos + T
+            mask = (torch.arange(S, device=x.device)[None, :]
+                    <= (pos + torch.arange(T, device=x.device))[:, None])
+            y = torch.nn.functional.scaled_dot_product_attention(
+                qt, kt, vt, attn_mask=mask)
+            y = y.transpose(1, 2)
+        return self.c_proj(y.reshape(B, T, C))
+
     def _attend(self, q, k, v):
         if self.attention_impl == AttentionImplementation.HIP_FLASH:
             return flash_attention(q, k, v, causal=True)
@@ -230,6 +275,12 @@ class GPT2Block(nn.Module):
 
     def forward(self, x, rope_cos, rope_sin):
         x = x + self.attn(self.attention_norm(x), rope_cos, rope_sin)
+        x = x + self.mlp(self.ffn_norm(x))
+        return x
+
+    def forward_cached(self, x, rope_cos, rope_sin, cache_k, cache_v, pos):
+        x = x + self.attn.forward_cached(self.attention_norm(x), rope_cos,
+                                         rope_sin, cache_k, cache_v, pos)
         x = x + self.mlp(self.ffn_norm(x))
         return x
 
@@ -295,3 +346,54 @@ class GPT2LLM(NNModel):
 
     def forward(self, inputs: dict[str, torch.Tensor]) -> dict[str, torch.Tensor]:
         return self.forward_impl(inputs)
+
+    # -- incremental decoding (KV cache) --------------------------------
+    def new_kv_cache(self, batch_size: int, max_len: Optional[int] = None,
+                     device=None, dtype=None) -> "KVCache":
+        cfg = self.config
+        max_len = max_len or cfg.sequence_length
+        p = next(self.parameters())
+        return KVCache(cfg.n_layer, batch_size, max_len,
+                       cfg.n_head_kv, cfg.n_embd // cfg.n_head_q,
+                       device or p.device, dtype or p.dtype)
+
+    @torch.no_grad()
+    def forward_cached(self, inputs: dict[str, torch.Tensor],
+                       cache: "KVCache") -> dict[str, torch.Tensor]:
+        """Forward only the NEW tokens; attends against the cached prefix.
+        Advances cache.pos. Logits are returned for the new tokens only."""
+        input_ids = inputs[self.sample_key]
+        B, T = input_ids.shape
+        pos = cache.pos
+        if pos + T > cache.max_len:
+            raise ValueError(f"KV cache overflow: {pos}+{T} > {cache.max_len}")
+        x = self.wte(input_ids)
+        if self.wpe is not None:
+            p = torch.arange(pos, pos + T, dtype=torch.long,
+                             device=input_ids.device)
+            x = x + self.wpe(p)
+        rope_cos, rope_sin = self._rope(cache.max_len, x.device)
+        for i, block in enumerate(self.blocks):
+            x = block.forward_cached(x, rope_cos, rope_sin,
+                                     cache.k[i], cache.v[i], pos)
+        cache.pos = pos + T
+        x = self.lm_head_norm(x)
+        return {self.prediction_key: self.lm_head(x)}
+
+
+class KVCache:
+    """Preallocated per-layer K/V buffers for incremental decoding."""
+
+    def __init__(self, n_layer: int, batch_size: int, max_len: int,
+                 n_head_kv: int, head_dim: int, device, dtype):
+        self.max_len = max_len
+        self.pos = 0
+        self.k = [torch.zeros(batch_size, max_len, n_head_kv, head_dim,
+                              device=device, dtype=dtype)
+                  for _ in range(n_layer)]
+        self.v = [torch.zeros(batch_size, max_len, n_head_kv, head_dim,
+                              device=device, dtype=dtype)
+                  for _ in range(n_layer)]
+
+    def reset(self):
+        self.pos = 0

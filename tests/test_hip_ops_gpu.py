@@ -549,3 +549,35 @@ def test_two_stream_linear_row_bound_matches_plain():
         torch.testing.assert_close(ref.weight.grad, two.weight.grad,
                                    rtol=0, atol=0)
         torch.testing.assert_close(ref.bias.grad, two.bias.grad, rtol=0, atol=0)
+
+
+@pytest.mark.gpu
+def test_kv_cache_decode_matches_full_reforward_gpu():
+    """KV-cache decode on the HIP flash path (q_offset attention against
+    the cache prefix) must match the full re-forward in bf16."""
+    from modalities_amd.models.gpt2 import GPT2LLM, GPT2LLMConfig
+
+    torch.manual_seed(3)
+    model = GPT2LLM(GPT2LLMConfig(
+        vocab_size=128, n_layer=2, n_head_q=4, n_head_kv=2, n_embd=512,
+        ffn_hidden=1024, sequence_length=256, seed=7, dropout=0.0,
+        fused_qkv=True)).to(DEV).to(torch.bfloat16).eval()
+
+    g = torch.Generator().manual_seed(9)
+    prompt = torch.randint(0, 128, (2, 100), generator=g).to(DEV)
+
+    with torch.no_grad():
+        cache = model.new_kv_cache(2, max_len=160)
+        out_c = model.forward_cached({"input_ids": prompt}, cache)["logits"]
+        ref = model({"input_ids": prompt})["logits"]
+        torch.testing.assert_close(out_c.float(), ref.float(),
+                                   rtol=2e-2, atol=2e-2)
+        ids = prompt
+        for _ in range(8):
+            nxt = ref[:, -1, :].argmax(-1, keepdim=True)
+            ids = torch.cat([ids, nxt], dim=1)
+            ref = model({"input_ids": ids})["logits"]
+            out_c = model.forward_cached({"input_ids": nxt}, cache)["logits"]
+            torch.testing.assert_close(out_c[:, -1].float(),
+                                       ref[:, -1].float(),
+                                       rtol=2e-2, atol=2e-2)

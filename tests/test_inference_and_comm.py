@@ -2,6 +2,7 @@
 inference/text/inference_component.py behavior) and the communication
 self-test (utils/communication_test.py) on gloo world 2."""
 
+import pytest
 import torch
 
 from modalities_amd.inference.text_generation import TextInferenceComponent
@@ -149,3 +150,41 @@ tokenizer:
         device=torch.device(components.settings.device))
     out = comp.generate_tokens("hi")
     assert isinstance(out, str)
+
+
+@pytest.mark.parametrize("kw", [
+    dict(),
+    dict(fused_qkv=True),
+    dict(n_head_kv=2),
+    dict(use_qk_norm=True),
+    dict(poe_type="ABSOLUTE"),
+])
+def test_kv_cache_decode_matches_full_reforward(kw):
+    """forward_cached (prefill + per-token decode) must reproduce the full
+    re-forward logits at every step."""
+    torch.manual_seed(0)
+    cfg = dict(vocab_size=97, n_layer=2, n_head_q=4, n_head_kv=4, n_embd=64,
+               ffn_hidden=128, sequence_length=32, seed=11, dropout=0.0)
+    cfg.update(kw)
+    model = GPT2LLM(GPT2LLMConfig(**cfg))
+    model.eval()
+
+    g = torch.Generator().manual_seed(5)
+    prompt = torch.randint(0, 97, (2, 7), generator=g)
+
+    with torch.no_grad():
+        cache = model.new_kv_cache(2, max_len=16)
+        out_c = model.forward_cached({"input_ids": prompt}, cache)["logits"]
+        ref = model({"input_ids": prompt})["logits"]
+        torch.testing.assert_close(out_c, ref, rtol=1e-4, atol=1e-4)
+
+        ids = prompt
+        for _ in range(6):
+            nxt = ref[:, -1, :].argmax(-1, keepdim=True)
+            ids = torch.cat([ids, nxt], dim=1)
+            ref = model({"input_ids": ids})["logits"]
+            out_c = model.forward_cached({"input_ids": nxt}, cache)["logits"]
+            torch.testing.assert_close(out_c[:, -1], ref[:, -1],
+                                       rtol=1e-4, atol=1e-4)
+        assert cache.pos == ids.shape[1]
+
