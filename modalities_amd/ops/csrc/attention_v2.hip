@@ -1134,8 +1134,10 @@ std::vector<torch::Tensor> attn_fwd2(torch::Tensor q, torch::Tensor k,
   // path reads it in place): strides must be (Tkv*vp, vp, D, 1).
   TORCH_CHECK(k.is_contiguous());
   TORCH_CHECK(v.stride(3) == 1 && v.stride(2) == D
-              && v.stride(0) == (long)Tkv * v.stride(1),
+              && (B == 1 || v.stride(0) == (long)Tkv * v.stride(1)),
               "v must be contiguous or a [B,T,Hkv,D] view with row stride");
+  // (B == 1: the batch-stride term is never used, so a dim-1 slice of a
+  // longer KV-cache buffer is read in place — zero-copy decode)
   const long vp = v.stride(1);
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
   TORCH_CHECK(D == 64 || D == 80 || D == 128,
@@ -1170,8 +1172,10 @@ static std::vector<torch::Tensor> attn_bwd2_impl(
   const int B = q.size(0), T = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Tkv = k.size(1), Hkv = k.size(2);
   TORCH_CHECK(v.stride(3) == 1 && v.stride(2) == D
-              && v.stride(0) == (long)Tkv * v.stride(1),
+              && (B == 1 || v.stride(0) == (long)Tkv * v.stride(1)),
               "v must be contiguous or a [B,T,Hkv,D] view with row stride");
+  // (B == 1: the batch-stride term is never used, so a dim-1 slice of a
+  // longer KV-cache buffer is read in place — zero-copy decode)
   const long vstride = v.stride(1);
   const int q_off = (int)q_offset;
   const float scale = 1.0f / sqrtf((float)D);
