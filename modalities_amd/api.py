@@ -176,6 +176,7 @@ def generate_text(config_path: Path) -> None:
         prompt_template=gen_cfg.get("prompt_template", "{text}"),
         sequence_length=settings.sequence_length,
         temperature=gen_cfg.get("temperature", 1.0),
+        top_k=gen_cfg.get("top_k"), top_p=gen_cfg.get("top_p"),
         eod_token=gen_cfg.get("eod_token", "<eod>"),
         device=torch.device(settings.device),
         sample_key=settings.referencing_keys.get("sample_key", "input_ids"),

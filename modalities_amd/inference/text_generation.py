@@ -18,7 +18,8 @@ class TextInferenceComponent:
     def __init__(self, model, tokenizer: TokenizerWrapper, prompt_template: str,
                  sequence_length: int, temperature: float = 1.0,
                  eod_token: str = "<eod>", device: Optional[torch.device] = None,
-                 sample_key: str = "input_ids", prediction_key: str = "logits"):
+                 sample_key: str = "input_ids", prediction_key: str = "logits",
+                 top_k: Optional[int] = None, top_p: Optional[float] = None):
         self.model = model
         self.tokenizer = tokenizer
         self.prompt_template = prompt_template
@@ -28,6 +29,32 @@ class TextInferenceComponent:
         self.device = device or torch.device("cpu")
         self.sample_key = sample_key
         self.prediction_key = prediction_key
+        self.top_k = top_k
+        self.top_p = top_p
+
+    def _sample(self, logits: torch.Tensor) -> torch.Tensor:
+        """Greedy (temperature==0) or temperature sampling with optional
+        top-k / nucleus (top-p) filtering."""
+        if self.temperature <= 0:
+            return logits.argmax(dim=-1, keepdim=True)
+        logits = logits / self.temperature
+        if self.top_k is not None and self.top_k > 0:
+            kth = torch.topk(logits, min(self.top_k, logits.shape[-1]),
+                             dim=-1).values[..., -1:]
+            logits = logits.masked_fill(logits < kth, float("-inf"))
+        if self.top_p is not None and 0.0 < self.top_p < 1.0:
+            sorted_logits, sorted_idx = torch.sort(logits, descending=True,
+                                                   dim=-1)
+            cum = torch.softmax(sorted_logits, dim=-1).cumsum(-1)
+            # drop tokens whose EXCLUSIVE cumulative prob already exceeds
+            # top_p (the highest-prob token always stays)
+            drop_sorted = torch.zeros_like(cum, dtype=torch.bool)
+            drop_sorted[..., 1:] = cum[..., :-1] > self.top_p
+            drop = torch.zeros_like(drop_sorted).scatter(
+                -1, sorted_idx, drop_sorted)
+            logits = logits.masked_fill(drop, float("-inf"))
+        probs = torch.softmax(logits, dim=-1)
+        return torch.multinomial(probs, num_samples=1)
 
     @torch.no_grad()
     def generate_tokens(self, context: str, echo: bool = False) -> str:
@@ -56,11 +83,7 @@ class TextInferenceComponent:
 
         for i in range(max_new):
             logits = out[self.prediction_key][:, -1, :].float()
-            if self.temperature > 0:
-                probs = torch.softmax(logits / self.temperature, dim=-1)
-                next_id = torch.multinomial(probs, num_samples=1)
-            else:
-                next_id = logits.argmax(dim=-1, keepdim=True)
+            next_id = self._sample(logits)
             token = next_id.item()
             if eod_id is not None and token == eod_id:
                 break

@@ -188,3 +188,28 @@ def test_kv_cache_decode_matches_full_reforward(kw):
                                        rtol=1e-4, atol=1e-4)
         assert cache.pos == ids.shape[1]
 
+
+
+def test_top_k_and_top_p_sampling():
+    """top_k=1 must equal greedy; top_p keeps exactly the nucleus."""
+    tok = CharTokenizer()
+    model = _tiny_model()
+
+    greedy = TextInferenceComponent(model, tok, "{text}", 16, temperature=0.0)
+    topk1 = TextInferenceComponent(model, tok, "{text}", 16, temperature=0.7,
+                                   top_k=1)
+    assert greedy.generate_tokens("ab") == topk1.generate_tokens("ab")
+
+    # nucleus filter keeps the minimal prefix covering top_p
+    gen = TextInferenceComponent(model, tok, "{text}", 16, temperature=1.0,
+                                 top_p=0.5)
+    logits = torch.log(torch.tensor([[0.4, 0.3, 0.2, 0.1]]))
+    torch.manual_seed(0)
+    seen = {gen._sample(logits).item() for _ in range(200)}
+    assert seen == {0, 1}  # 0.4 then 0.4+0.3 > 0.5 -> nucleus = {0, 1}
+
+    # top_k=2 restricts to the two best
+    gen2 = TextInferenceComponent(model, tok, "{text}", 16, temperature=1.0,
+                                  top_k=2)
+    seen2 = {gen2._sample(logits).item() for _ in range(200)}
+    assert seen2 == {0, 1}
