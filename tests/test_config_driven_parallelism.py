@@ -343,3 +343,66 @@ def test_config_driven_hsdp_warmstart(tmp_path):
     results = run_distributed(_rank_warmstart_hsdp, world_size=4, port=29507,
                               args=(str(cfg), str(ws_cfg)), timeout_s=420)
     assert all(v == "ok" for v in results.values()) and len(results) == 4
+
+
+def _rank_warmstart_cp(rank, world, cfg_path, ws_cfg_path):
+    from modalities_amd.main import Main
+
+    main_a = Main(Path(cfg_path), experiment_id="cpA")
+    main_a.run(main_a.build_components())
+    root = Path(cfg_path).parent / "ckpt" / "cpA"
+    step4 = [p for p in root.iterdir() if "seen_steps_4" in p.name][0]
+
+    def ws_resolver(key):
+        return {"checkpoint_folder_path": str(step4)}[key]
+
+    main_b = Main(Path(ws_cfg_path), experiment_id="cpB",
+                  additional_resolver_funs={"warmstart_env": ws_resolver})
+    main_b.run(main_b.build_components())
+    return "ok"
+
+
+def test_config_driven_cp2_warmstart(tmp_path):
+    """Resume a CP run: the cp-rank-0 peer wrote the (world-1) layout; both
+    CP peers reload it identically through the mesh-aware app_state."""
+    import yaml
+
+    rng = np.random.default_rng(61)
+    from modalities_amd.dataloader.packed_data import write_pbin
+    docs = [rng.integers(0, 256, size=200, dtype=np.uint8) for _ in range(8)]
+    pbin = tmp_path / "data.pbin"
+    write_pbin(pbin, docs, token_size_in_bytes=1)
+
+    template = Path(__file__).parent / "configs" / "config_tiny_e2e_cp2.yaml"
+    text = template.read_text()
+    text = text.replace("DATASET_PATH_PLACEHOLDER", str(pbin))
+    text = text.replace("CHECKPOINT_DIR_PLACEHOLDER", str(tmp_path / "ckpt"))
+    text = text.replace("RESULTS_PATH_PLACEHOLDER",
+                        str(tmp_path / "a_results.jsonl"))
+    cfg = tmp_path / "config.yaml"
+    cfg.write_text(text)
+
+    cfg_dict = yaml.safe_load(text.replace(str(tmp_path / "a_results.jsonl"),
+                                           str(tmp_path / "b_results.jsonl")))
+    cfg_dict["app_state"] = {
+        "component_key": "app_state", "variant_key": "sharded_warmstart",
+        "config": {
+            "model": {"instance_key": "wrapped_model", "pass_type": "BY_REFERENCE"},
+            "optimizer": {"instance_key": "optimizer", "pass_type": "BY_REFERENCE"},
+            "lr_scheduler": {"instance_key": "scheduler", "pass_type": "BY_REFERENCE"},
+            "checkpoint_folder_path": "${warmstart_env:checkpoint_folder_path}",
+            "device_mesh": {"instance_key": "device_mesh",
+                            "pass_type": "BY_REFERENCE"},
+        },
+    }
+    ws_cfg = tmp_path / "ws.yaml"
+    ws_cfg.write_text(yaml.safe_dump(cfg_dict, sort_keys=False))
+
+    results = run_distributed(_rank_warmstart_cp, world_size=2, port=29513,
+                              args=(str(cfg), str(ws_cfg)), timeout_s=420)
+    assert results == {0: "ok", 1: "ok"}
+    # only the cp-rank-0 peer wrote: flat namespace, single shard, world 1
+    root = tmp_path / "ckpt" / "cpA"
+    folder = next(p for p in root.iterdir() if p.is_dir())
+    names = sorted(f.name for f in folder.iterdir())
+    assert names == ["meta.json", "shards_rank_0.pt"], names
