@@ -68,6 +68,10 @@ class CheckpointedBlock(nn.Module):
             kw["context_fn"] = self._context_fn
         return checkpoint(self.block, *args, **kwargs, **kw)
 
+    def forward_cached(self, *args, **kwargs):
+        # incremental decoding is inference-only: no recompute needed
+        return self.block.forward_cached(*args, **kwargs)
+
 
 def apply_activation_checkpointing_(model,
                                     variant: ActivationCheckpointingVariant,
