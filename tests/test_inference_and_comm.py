@@ -213,3 +213,16 @@ def test_top_k_and_top_p_sampling():
                                   top_k=2)
     seen2 = {gen2._sample(logits).item() for _ in range(200)}
     assert seen2 == {0, 1}
+
+
+def test_kv_cache_overflow_raises():
+    model = _tiny_model()
+    model.eval()
+    cache = model.new_kv_cache(1, max_len=8)
+    with torch.no_grad():
+        model.forward_cached({"input_ids": torch.randint(0, 97, (1, 6))}, cache)
+        with pytest.raises(ValueError, match="overflow"):
+            model.forward_cached({"input_ids": torch.randint(0, 97, (1, 4))},
+                                 cache)
+    cache.reset()
+    assert cache.pos == 0
