@@ -37,8 +37,13 @@ class ModelFactory:
                                            ActivationCheckpointingVariant,
                                            layers_fqn: str = "blocks",
                                            every_k_layers: int = 1) -> nn.Module:
-        apply_activation_checkpointing_(model, activation_checkpointing_variant,
-                                        layers_fqn, every_k_layers)
+        v = activation_checkpointing_variant
+        if isinstance(v, str):  # YAML gives enum NAME or value; both accepted
+            try:
+                v = ActivationCheckpointingVariant(v.lower())
+            except ValueError:
+                v = ActivationCheckpointingVariant[v.upper()]
+        apply_activation_checkpointing_(model, v, layers_fqn, every_k_layers)
         return model
 
     @staticmethod

@@ -246,3 +246,39 @@ def test_torch_compile_blockwise_matches_eager():
     with torch.no_grad():
         out = compiled({"input_ids": ids})["logits"]
     torch.testing.assert_close(out, ref, rtol=2e-4, atol=2e-4)
+
+
+def test_ac_variant_string_coercion_from_yaml():
+    """YAML gives the AC variant as a string (enum name OR value); the
+    factory must coerce it — a bad string must raise, and selective_layer
+    with k=2 must leave alternate blocks unwrapped."""
+    from modalities_amd.config.component_factory import ComponentFactory
+    from modalities_amd.registry.components import get_default_registry
+
+    def build(variant, k=1):
+        c = {
+            "model": {"component_key": "model", "variant_key": "gpt2",
+                      "config": dict(sample_key="input_ids",
+                                     prediction_key="logits", vocab_size=64,
+                                     n_layer=4, n_head_q=2, n_head_kv=2,
+                                     n_embd=32, ffn_hidden=64,
+                                     sequence_length=16, seed=1)},
+            "acm": {"component_key": "activation_checkpointed_model",
+                    "variant_key": "default",
+                    "config": {
+                        "model": {"instance_key": "model",
+                                  "pass_type": "BY_REFERENCE"},
+                        "activation_checkpointing_variant": variant,
+                        "every_k_layers": k}},
+        }
+        return ComponentFactory(get_default_registry()).build_component_by_key(
+            c, "acm")
+
+    m = build("SELECTIVE_LAYER_ACTIVATION_CHECKPOINTING", k=2)
+    kinds = [type(b).__name__ for b in m.blocks]
+    assert kinds == ["CheckpointedBlock", "GPT2Block",
+                     "CheckpointedBlock", "GPT2Block"], kinds
+    m2 = build("selective_op_activation_checkpointing")
+    assert m2.blocks[0]._context_fn is not None
+    with pytest.raises(KeyError):
+        build("not_a_variant")
