@@ -100,6 +100,16 @@ def CMD_entry_point_generate_text(config_file_path: Path):
     api.generate_text(config_file_path)
 
 
+@main.command(name="serve")
+@click.option("--config_file_path", type=click.Path(exists=True, path_type=Path),
+              required=True)
+@click.option("--host", type=str, default="127.0.0.1")
+@click.option("--port", type=int, default=8000)
+def CMD_entry_point_serve(config_file_path: Path, host: str, port: int):
+    """HTTP inference server (POST /generate) over the KV-cache decoder."""
+    api.serve(config_file_path, host=host, port=port)
+
+
 @main.command(name="convert_pytorch_to_hf_checkpoint")
 @click.option("--config_file_path", type=click.Path(exists=True, path_type=Path),
               required=True)

@@ -154,9 +154,7 @@ def create_filtered_tokenized_dataset(input_data_path: Path,
         Path(input_data_path), Path(output_data_path), filter_routine)
 
 
-def generate_text(config_path: Path) -> None:
-    """Interactive text generation from a config naming model+tokenizer
-    (reference api.py:101-107)."""
+def _build_text_inference_component(config_path: Path):
     from modalities_amd.config.component_factory import ComponentFactory
     from modalities_amd.config.instantiation_models import \
         TextGenerationInstantiationModel
@@ -181,7 +179,19 @@ def generate_text(config_path: Path) -> None:
         device=torch.device(settings.device),
         sample_key=settings.referencing_keys.get("sample_key", "input_ids"),
         prediction_key=settings.referencing_keys.get("prediction_key", "logits"))
-    comp.run()
+    return comp
+
+
+def generate_text(config_path: Path) -> None:
+    """Interactive text generation from a config naming model+tokenizer
+    (reference api.py:101-107)."""
+    _build_text_inference_component(config_path).run()
+
+
+def serve(config_path: Path, host: str = "127.0.0.1", port: int = 8000) -> None:
+    """HTTP inference server (POST /generate) over the KV-cache decoder."""
+    from modalities_amd.inference.server import serve as _serve
+    _serve(_build_text_inference_component(config_path), host=host, port=port)
 
 
 def convert_pytorch_to_hf_checkpoint(config_path: Path, output_hf_checkpoint_dir: Path,
