@@ -10,6 +10,7 @@ Response: {"text": "...", "prompt_tokens": N, "generated_tokens": M,
            "latency_ms": T}
 """
 
+import threading
 import time
 from typing import Optional
 
@@ -33,12 +34,17 @@ def build_app(component: TextInferenceComponent):
     def health():
         return {"status": "ok"}
 
+    # generation mutates component state (sampling overrides, KV cache):
+    # serialize requests — FastAPI runs sync handlers in a thread pool
+    gen_lock = threading.Lock()
+
     @app.post("/generate")
     def generate(req: GenerateRequest):
         text = component.prompt_template.format(text=req.prompt) \
             if "{text}" in component.prompt_template else req.prompt
         n_prompt = len(component.tokenizer.tokenize(text))
         # per-request sampling overrides; sequence_length bounds decoding
+        gen_lock.acquire()
         saved = (component.temperature, component.top_k, component.top_p,
                  component.sequence_length)
         try:
@@ -56,6 +62,7 @@ def build_app(component: TextInferenceComponent):
         finally:
             (component.temperature, component.top_k, component.top_p,
              component.sequence_length) = saved
+            gen_lock.release()
         return {"text": out,
                 "prompt_tokens": n_prompt,
                 "generated_tokens": len(component.tokenizer.tokenize(out)),
