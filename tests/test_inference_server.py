@@ -55,3 +55,45 @@ def test_generate_respects_max_new_tokens(client):
 def test_generate_validation_error(client):
     r = client.post("/generate", json={"prompt": "x", "max_new_tokens": 0})
     assert r.status_code == 422
+
+
+def test_serve_builds_from_config(tmp_path):
+    """The serve CLI's build path: config -> TextInferenceComponent -> app."""
+    cfg_text = """\
+settings:
+  referencing_keys: {sample_key: input_ids, prediction_key: logits}
+  device: cpu
+  sequence_length: 24
+
+model:
+  component_key: model
+  variant_key: gpt2
+  config:
+    sample_key: input_ids
+    prediction_key: logits
+    vocab_size: 260
+    n_layer: 2
+    n_head_q: 4
+    n_head_kv: 4
+    n_embd: 64
+    ffn_hidden: 128
+    sequence_length: 24
+    seed: 3
+
+tokenizer:
+  component_key: tokenizer
+  variant_key: char
+  config: {}
+
+text_inference:
+  prompt_template: "{text}"
+  temperature: 0.0
+"""
+    cfg = tmp_path / "gen.yaml"
+    cfg.write_text(cfg_text)
+
+    from modalities_amd.api import _build_text_inference_component
+    comp = _build_text_inference_component(cfg)
+    c = TestClient(build_app(comp))
+    r = c.post("/generate", json={"prompt": "hey", "max_new_tokens": 4})
+    assert r.status_code == 200 and isinstance(r.json()["text"], str)
