@@ -172,3 +172,36 @@ def test_loss_masking_matches_naive_scan(seq, seed):
         else:
             expected.append(tok if depth > 0 else -100)
     assert got == expected
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    n_docs=st.integers(1, 20),
+    seed=st.integers(0, 500),
+    token_size=st.sampled_from([1, 2]),
+)
+def test_shuffle_preserves_document_multiset(tmp_path_factory, n_docs, seed,
+                                             token_size):
+    """Document-level shuffling must be a permutation: same multiset of
+    documents, deterministic for a fixed seed."""
+    from modalities_amd.preprocessing.shuffle_data import shuffle_tokenized_data
+
+    rng = np.random.default_rng(seed)
+    docs = [rng.integers(0, 2 ** (8 * token_size) - 1,
+                         size=int(rng.integers(1, 30))).astype(DTYPES[token_size])
+            for _ in range(n_docs)]
+    root = tmp_path_factory.mktemp("shuf")
+    src = root / "src.pbin"
+    write_pbin(src, docs, token_size_in_bytes=token_size)
+
+    out1, out2 = root / "a.pbin", root / "b.pbin"
+    shuffle_tokenized_data(src, out1, batch_size=4, seed=seed)
+    shuffle_tokenized_data(src, out2, batch_size=4, seed=seed)
+
+    def read_docs(p):
+        s = EmbeddedStreamData(p)
+        return [tuple(s.tokens(o, l).tolist()) for o, l in s.index_base]
+
+    a, b = read_docs(out1), read_docs(out2)
+    assert a == b  # deterministic
+    assert sorted(a) == sorted(tuple(d.tolist()) for d in docs)  # permutation
