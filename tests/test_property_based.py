@@ -205,3 +205,23 @@ def test_shuffle_preserves_document_multiset(tmp_path_factory, n_docs, seed,
     a, b = read_docs(out1), read_docs(out2)
     assert a == b  # deterministic
     assert sorted(a) == sorted(tuple(d.tolist()) for d in docs)  # permutation
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    T=st.integers(2, 64).filter(lambda t: t % 2 == 0),
+    cp=st.sampled_from([1, 2, 4]),
+    B=st.integers(1, 3),
+)
+def test_cp_target_slices_partition_sequence(T, cp, B):
+    """CP target slices across ranks must tile the sequence exactly."""
+    import torch
+
+    from modalities_amd.parallel.cp import slice_targets_for_cp
+    if T % cp != 0:
+        return
+    t = torch.arange(B * T).view(B, T)
+    parts = [slice_targets_for_cp(t, r, cp) for r in range(cp)]
+    assert all(p.shape == (B, T // cp) for p in parts)
+    recon = torch.cat(parts, dim=1)
+    assert torch.equal(recon, t)
