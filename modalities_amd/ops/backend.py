@@ -1,4 +1,4 @@
-"""HIP extension loading and dispatch policy.
+"""MI355X-native addition (no reference analog): HIP extension loading and dispatch policy.
 
 The extension is built IN-TREE (modalities_amd/ops/_hip_ops.so) by
 ``modalities_amd.ops.build.build_extension()`` (driven from __graft_entry__)

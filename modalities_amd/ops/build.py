@@ -1,4 +1,4 @@
-"""Build the HIP op extension IN-TREE (modalities_amd/ops/_hip_ops.so).
+"""MI355X-native addition (no reference analog): Build the HIP op extension IN-TREE (modalities_amd/ops/_hip_ops.so).
 
 Drives hipcc directly (native HIP source, no hipify) with torch's include/
 lib paths; gfx950 only. The .so lands next to this file so repo snapshots

@@ -1,4 +1,5 @@
-"""TwoStreamLinear: nn.Linear with a two-stream backward.
+"""TwoStreamLinear: nn.Linear with a two-stream backward (MI355X-native
+addition, no reference analog).
 
 Autograd runs a linear's input-gradient and weight-gradient GEMMs
 sequentially on one stream; they are independent, and each alone leaves

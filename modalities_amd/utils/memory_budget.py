@@ -1,4 +1,4 @@
-"""Per-GPU memory budgeting for the XGMI sharding engine (VERDICT r1 #7:
+"""MI355X-native addition (no reference analog): Per-GPU memory budgeting for the XGMI sharding engine (VERDICT r1 #7:
 make the 70B single-node config real with documented memory math).
 
 The engine's steady-state per-GPU footprint for a model of N params on
