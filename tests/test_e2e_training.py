@@ -109,3 +109,12 @@ def test_bench_contract_multirank_cpu(tmp_path):
     assert d["value"] > 0 and d["ms_per_step"] > 0
     assert d["scaling"] == "weak"
     assert d["dtype"] == "fp32"  # CPU fallback dtype (bf16 on GPU)
+    # the full driver contract: every documented field present
+    for field in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                  "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                  "dtype", "data", "config"):
+        assert field in d, field
+    assert d["higher_is_better"] is True and d["data"] == "synthetic"
+    for field in ("model", "global_batch", "micro_batch", "seq_len",
+                  "parallelism", "num_params"):
+        assert field in d["config"], field
