@@ -24,6 +24,11 @@ class ResumableDistributedSampler(Sampler):
         self.skip_num_global_samples = skip_num_global_samples
 
         self.global_num_samples = len(self.dataset)
+        if skip_num_global_samples >= self.global_num_samples > 0:
+            raise ValueError(
+                f"skip_num_global_samples={skip_num_global_samples} >= dataset "
+                f"size {self.global_num_samples}: nothing left to train on "
+                "(warmstart resumed past the end of the data?)")
         self.global_num_samples_effective = self.global_num_samples - skip_num_global_samples
         if drop_last:
             self.num_samples = self.global_num_samples_effective // num_replicas

@@ -189,3 +189,10 @@ def test_cli_error_log_schema(tmp_path, monkeypatch):
         assert key in rec, rec
     assert rec["exception_type"] == "RuntimeError"
     assert "synthetic failure xyz" in rec["message"]
+
+
+def test_sampler_skip_past_end_raises():
+    from modalities_amd.dataloader.samplers import ResumableDistributedSampler
+    with pytest.raises(ValueError, match="nothing left"):
+        ResumableDistributedSampler(dataset=list(range(10)), rank=0,
+                                    num_replicas=2, skip_num_global_samples=10)
